@@ -1,0 +1,110 @@
+"""Learner DP without GPUs: gloo backend, world_size 2,
+on 127.0.0.1. Verifies GradReducer produces the mean gradient and that
+ranks stay in lockstep."""
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from torched_impala_amd.models import MLPPolicy
+from torched_impala_amd.parallel import GradReducer
+
+
+def _worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)  # same init on both ranks
+        model = MLPPolicy((4,), 2)
+        reducer = GradReducer(model, bucket_bytes=1 << 10)  # force >1 bucket
+        torch.manual_seed(100 + rank)  # different data per rank
+        x = torch.randn(3, 2, 4)
+        inputs = dict(frame=x, reward=torch.zeros(3, 2),
+                      done=torch.zeros(3, 2, dtype=torch.bool),
+                      last_action=torch.zeros(3, 2, dtype=torch.long))
+        reducer.prepare()
+        out, _ = model(inputs, ())
+        loss = out["policy_logits"].pow(2).sum() + out["baseline"].pow(2).sum()
+        loss.backward()
+        reducer.finalize()
+        grads = torch.cat([p.grad.flatten() for p in model.parameters()])
+        q.put((rank, loss.item(), grads.numpy().copy()))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_gradreducer_gloo_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29611
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, loss, grads = q.get(timeout=120)
+        results[rank] = (loss, grads)
+    for p in procs:
+        p.join(timeout=30)
+    g0, g1 = results[0][1], results[1][1]
+    # after all-reduce(mean) both ranks hold identical gradients
+    torch.testing.assert_close(torch.from_numpy(g0), torch.from_numpy(g1), rtol=1e-6, atol=1e-7)
+
+
+def _worker_ref(rank, world, port, q):
+    """Compare reduced grads against a single-process sum/2 reference."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+        model = MLPPolicy((4,), 2)
+        reducer = GradReducer(model)
+        x = torch.full((1, 1, 4), float(rank + 1))
+        inputs = dict(frame=x, reward=torch.zeros(1, 1),
+                      done=torch.zeros(1, 1, dtype=torch.bool),
+                      last_action=torch.zeros(1, 1, dtype=torch.long))
+        reducer.prepare()
+        out, _ = model(inputs, ())
+        out["baseline"].sum().backward()
+        reducer.finalize()
+        if rank == 0:
+            q.put(torch.cat([p.grad.flatten() for p in model.parameters()]).numpy().copy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_gradreducer_matches_mean_of_local_grads():
+    # local reference: mean of grads computed with x=1 and x=2
+    torch.manual_seed(0)
+    ref_model = MLPPolicy((4,), 2)
+    ref_grads = []
+    for val in (1.0, 2.0):
+        ref_model.zero_grad()
+        inputs = dict(frame=torch.full((1, 1, 4), val),
+                      reward=torch.zeros(1, 1),
+                      done=torch.zeros(1, 1, dtype=torch.bool),
+                      last_action=torch.zeros(1, 1, dtype=torch.long))
+        out, _ = ref_model(inputs, ())
+        out["baseline"].sum().backward()
+        ref_grads.append(torch.cat(
+            [(p.grad if p.grad is not None else torch.zeros_like(p)).flatten()
+             for p in ref_model.parameters()]))
+    expected = (ref_grads[0] + ref_grads[1]) / 2
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_ref, args=(r, 2, 29612, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=120)
+    for p in procs:
+        p.join(timeout=30)
+    torch.testing.assert_close(torch.from_numpy(got), expected, rtol=1e-6, atol=1e-7)

@@ -1,0 +1,124 @@
+"""Transport correctness: no lost/duplicated slots under
+multi-process contention; tear-freedom via per-cell sequence numbers."""
+import multiprocessing
+import os
+import time
+
+import pytest
+import torch
+
+from torched_impala_amd.transport import SlotQueue, TrajectoryBuffers
+
+
+def test_slotqueue_single_process():
+    q = SlotQueue(8)
+    assert q.pop() == -1
+    for i in range(8):
+        assert q.push(i)
+    assert not q.push(99)  # full
+    got = sorted(q.pop() for _ in range(8))
+    assert got == list(range(8))
+    assert q.pop() == -1
+
+
+def test_slotqueue_pop_wait_timeout_and_abort():
+    q = SlotQueue(4)
+    t0 = time.time()
+    assert q.pop_wait(timeout_ms=100) == -1
+    assert time.time() - t0 >= 0.08
+    q.abort()
+    assert q.pop_wait(timeout_ms=5000) == -2
+
+
+def _producer(q, n_items, offset):
+    for i in range(n_items):
+        v = offset + i
+        while not q.push(v):
+            time.sleep(0.0001)
+
+
+def _consumer(q, out_q, n_total):
+    got = []
+    while len(got) < n_total:
+        v = q.pop_wait(timeout_ms=20000)
+        if v < 0:
+            break
+        got.append(v)
+    out_q.put(got)
+
+
+@pytest.mark.timeout(120)
+def test_slotqueue_mpmc_stress():
+    """2 producers x 2 consumers, 2000 items each: every item delivered
+    exactly once."""
+    ctx = multiprocessing.get_context("spawn")
+    q = SlotQueue(64)
+    n = 2000
+    out = ctx.Queue()
+    producers = [ctx.Process(target=_producer, args=(q, n, k * n))
+                 for k in range(2)]
+    consumers = [ctx.Process(target=_consumer, args=(q, out, n))
+                 for _ in range(2)]
+    for p in producers + consumers:
+        p.start()
+    got = []
+    for _ in range(2):
+        got.extend(out.get(timeout=60))
+    for p in producers + consumers:
+        p.join(timeout=30)
+    assert sorted(got) == list(range(2 * n))
+
+
+def _writer_proc(buffers, free_q, full_q, n_unrolls, writer_id):
+    import torch as th
+    for i in range(n_unrolls):
+        m = free_q.pop_wait(timeout_ms=30000)
+        assert m >= 0
+        slot = buffers.slot(m)
+        # stamp a recognisable pattern: writer_id*1e6 + i in every element
+        val = writer_id * 1000000 + i
+        slot["reward"].fill_(float(val))
+        slot["action"].fill_(val)
+        full_q.push(m)
+
+
+@pytest.mark.timeout(120)
+def test_trajectory_buffers_no_torn_slots():
+    """Multi-process writers + single reader: each drained slot is
+    internally consistent (all elements stamped with one value)."""
+    ctx = multiprocessing.get_context("spawn")
+    buffers = TrajectoryBuffers(unroll_length=4, obs_shape=(2, 4, 4),
+                                obs_dtype=torch.uint8, num_actions=3,
+                                num_slots=8)
+    free_q, full_q = SlotQueue(8), SlotQueue(8)
+    for m in range(8):
+        free_q.push(m)
+    n_unrolls = 200
+    writers = [ctx.Process(target=_writer_proc,
+                           args=(buffers, free_q, full_q, n_unrolls, w))
+               for w in range(3)]
+    for w in writers:
+        w.start()
+    seen = set()
+    for _ in range(3 * n_unrolls):
+        m = full_q.pop_wait(timeout_ms=30000)
+        assert m >= 0
+        slot = buffers.slot(m)
+        r = slot["reward"]
+        v0 = float(r.flatten()[0])
+        assert (r == v0).all(), "torn slot: mixed reward stamps"
+        assert (slot["action"] == int(v0)).all(), "torn slot: action mismatch"
+        seen.add(v0)
+        free_q.push(m)
+    for w in writers:
+        w.join(timeout=30)
+    assert len(seen) == 3 * n_unrolls  # nothing lost or duplicated
+
+
+def test_buffers_nbytes_and_shapes():
+    b = TrajectoryBuffers(unroll_length=20, obs_shape=(4, 84, 84),
+                          obs_dtype=torch.uint8, num_actions=6,
+                          num_slots=64, lstm_hidden=256)
+    assert b.arenas["frame"].shape == (64, 21, 1, 4, 84, 84)
+    assert b.arenas["initial_core_state"].shape == (64, 2, 1, 256)
+    assert b.nbytes() > 64 * 21 * 4 * 84 * 84

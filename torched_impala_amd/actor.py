@@ -1,0 +1,139 @@
+"""CPU actor process (BASELINE.json:5 "CPU actor processes").
+
+Each actor owns ``n = envs_per_actor`` environment instances, a local CPU
+copy of the policy (refreshed from the learner's WeightMirror), and
+writes whole unroll slots into the shared TrajectoryBuffers:
+
+  claim free slot -> (maybe) pull weights -> roll T steps -> publish full.
+
+Slots have shape (T+1, n, ...). Row semantics (TrajectoryBuffers
+docstring): row 0 of each unroll is the carried-over last row of the
+previous unroll, and ``initial_core_state`` is the LSTM state *before*
+the forward that produced that carried row, so the learner can replay
+rows 0..T from it.
+"""
+from __future__ import annotations
+
+import logging
+
+import numpy as np
+import torch
+
+from torched_impala_amd.envs import make_env
+from torched_impala_amd.models import build_model
+
+log = logging.getLogger(__name__)
+
+ROW_KEYS = ("frame", "reward", "done", "action", "behavior_logits",
+            "last_action", "episode_return", "episode_step")
+
+
+class Rollout:
+    """Rollout state over n env copies driven by one policy forward."""
+
+    def __init__(self, flags, actor_index: int, model):
+        n = flags.envs_per_actor
+        self.n = n
+        self.envs = [make_env(flags, seed=flags.seed + 10000 * actor_index + i)
+                     for i in range(n)]
+        self.model = model
+        obs0 = np.stack([e.reset() for e in self.envs])
+        self.frame = torch.from_numpy(obs0)
+        self.reward = torch.zeros(n)
+        self.done = torch.ones(n, dtype=torch.bool)  # first frame: episode start
+        self.last_action = torch.zeros(n, dtype=torch.int64)
+        self.core_state = model.initial_state(n)
+        self.episode_return = torch.zeros(n)
+        self.episode_step = torch.zeros(n, dtype=torch.int32)
+        self.core_before = tuple(t.clone() for t in self.core_state)
+        self.carry = None
+        self._advance()  # produce the first carried row
+
+    @torch.no_grad()
+    def _forward(self):
+        inputs = dict(
+            frame=self.frame.unsqueeze(0),
+            reward=self.reward.unsqueeze(0),
+            done=self.done.unsqueeze(0),
+            last_action=self.last_action.unsqueeze(0),
+        )
+        out, new_core = self.model(inputs, self.core_state)
+        return out, new_core
+
+    def _advance(self):
+        """One timestep: forward -> record row -> step envs.
+        Sets self.carry to the recorded row."""
+        out, new_core = self._forward()
+        action = out["action"][0]
+        self.carry = dict(
+            frame=self.frame, reward=self.reward.clone(),
+            done=self.done.clone(), action=action,
+            behavior_logits=out["policy_logits"][0],
+            last_action=self.last_action,  # the action INPUT to this forward
+            episode_return=self.episode_return.clone(),
+            episode_step=self.episode_step.clone(),
+        )
+        finished = self.done
+        if finished.any():  # reset stats after the done row was recorded
+            self.episode_return = self.episode_return.clone()
+            self.episode_step = self.episode_step.clone()
+            self.episode_return[finished] = 0.0
+            self.episode_step[finished] = 0
+        self.core_state = new_core
+        self.last_action = action
+        # step all envs with the chosen actions (auto-reset on done)
+        obs_list, rew, done = [], [], []
+        act = action.tolist()
+        for i, env in enumerate(self.envs):
+            o, r, d, _ = env.step(act[i])
+            self.episode_return[i] += r
+            self.episode_step[i] += 1
+            if d:
+                o = env.reset()
+            obs_list.append(o)
+            rew.append(r)
+            done.append(d)
+        self.frame = torch.from_numpy(np.stack(obs_list))
+        self.reward = torch.tensor(rew, dtype=torch.float32)
+        self.done = torch.tensor(done, dtype=torch.bool)
+
+    def fill_slot(self, slot, T: int):
+        """Write rows 0..T (shape (T+1, n, ...)) of one slot."""
+        for k in ROW_KEYS:
+            slot[k][0].copy_(self.carry[k])
+        if "initial_core_state" in slot and self.core_before:
+            # (2, n, hidden)
+            slot["initial_core_state"].copy_(torch.stack(self.core_before))
+        for t in range(1, T + 1):
+            if t == T:
+                self.core_before = tuple(x.clone() for x in self.core_state)
+            self._advance()
+            for k in ROW_KEYS:
+                slot[k][t].copy_(self.carry[k])
+
+
+def act(flags, actor_index: int, free_queue, full_queue, buffers, mirror,
+        obs_shape, obs_dtype, num_actions):
+    """Actor process entry point."""
+    try:
+        torch.set_num_threads(1)
+        torch.manual_seed(flags.seed + actor_index)
+        model = build_model(flags, obs_shape, num_actions)
+        model.eval()
+        version = mirror.pull_into(model)
+        roll = Rollout(flags, actor_index, model)
+        T = flags.unroll_length
+        while True:
+            m = free_queue.pop_wait()
+            if m < 0:  # abort/sentinel
+                break
+            version = mirror.pull_into(model, last_seen=version)
+            roll.fill_slot(buffers.slot(m), T)
+            if not full_queue.push(m):
+                log.error("actor %d: full queue overflow", actor_index)
+                break
+    except KeyboardInterrupt:
+        pass
+    except Exception:
+        log.exception("actor %d crashed", actor_index)
+        raise

@@ -1,0 +1,419 @@
+"""The learner: batch assembly, V-trace learn step, and train()/test()
+orchestration.
+
+MI355X specifics:
+* batch drain: full slots are copied shm -> HBM with async copies on a
+  dedicated side HIP stream (arenas hipHostRegister'ed), overlapping the
+  previous step's forward/backward; double-buffered device staging.
+* DP: GradReducer (parallel/ddp.py) all-reduces grad buckets over
+  RCCL/xGMI as backward produces them.
+* losses: ops.fused_vtrace_loss — gfx950 HIP kernels on GPU.
+"""
+from __future__ import annotations
+
+import dataclasses
+import logging
+import os
+import time
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.multiprocessing as mp
+
+from torched_impala_amd import checkpoint as ckpt_mod
+from torched_impala_amd import ops
+from torched_impala_amd.actor import act
+from torched_impala_amd.config import Flags
+from torched_impala_amd.envs import env_spec
+from torched_impala_amd.metrics import MetricsLogger, StageTimers
+from torched_impala_amd.models import build_model
+from torched_impala_amd.parallel import GradReducer, init_distributed
+from torched_impala_amd.transport import SlotQueue, TrajectoryBuffers, WeightMirror
+
+log = logging.getLogger(__name__)
+
+BATCH_KEYS = ("frame", "reward", "done", "action", "behavior_logits",
+              "last_action", "episode_return", "episode_step")
+
+
+def _np_to_torch_dtype(np_dtype):
+    import numpy as np
+    return {np.dtype(np.uint8): torch.uint8,
+            np.dtype(np.float32): torch.float32}[np.dtype(np_dtype)]
+
+
+def create_buffers(flags: Flags, obs_shape, obs_dtype, num_actions
+                   ) -> TrajectoryBuffers:
+    n = flags.envs_per_actor
+    assert flags.batch_size % n == 0, "batch_size must divide envs_per_actor"
+    slots_per_batch = flags.batch_size // n
+    num_slots = flags.num_buffers or max(2 * slots_per_batch,
+                                         slots_per_batch + flags.num_actors)
+    return TrajectoryBuffers(
+        unroll_length=flags.unroll_length,
+        obs_shape=obs_shape,
+        obs_dtype=_np_to_torch_dtype(obs_dtype),
+        num_actions=num_actions,
+        num_slots=num_slots,
+        lstm_hidden=256 if flags.use_lstm else 0,
+        envs_per_slot=n)
+
+
+class Batcher:
+    """Assembles time-major device batches from full ring slots.
+
+    GPU: double-buffered device staging filled by async H2D copies on a
+    side stream; slots are freed once their copy event completes.
+    CPU: plain stack.
+    """
+
+    def __init__(self, flags: Flags, buffers: TrajectoryBuffers,
+                 free_queue: SlotQueue, full_queue: SlotQueue, device):
+        self.flags = flags
+        self.buffers = buffers
+        self.free_queue = free_queue
+        self.full_queue = full_queue
+        self.device = device
+        self.k = flags.batch_size // flags.envs_per_actor
+        self.is_cuda = device.type == "cuda"
+        self._pending: List[Tuple[torch.cuda.Event, List[int]]] = []
+        if self.is_cuda:
+            buffers.pin()
+            self.copy_stream = torch.cuda.Stream()
+            self.staging = []
+            self.consumed_ev = []
+            for _ in range(2):
+                st = {key: torch.empty((self.k, *a.shape[1:]), dtype=a.dtype,
+                                       device=device)
+                      for key, a in buffers.arenas.items()}
+                self.staging.append(st)
+                ev = torch.cuda.Event()
+                ev.record()  # initially free
+                self.consumed_ev.append(ev)
+            self.idx = 0
+
+    def _drain_pending(self, block: bool = False):
+        still = []
+        for ev, slots in self._pending:
+            if block:
+                ev.synchronize()
+            if ev.query():
+                for m in slots:
+                    self.free_queue.push(m)
+            else:
+                still.append((ev, slots))
+        self._pending = still
+
+    def _pop_slots(self) -> List[int]:
+        slots = []
+        while len(slots) < self.k:
+            m = self.full_queue.pop_wait(timeout_ms=60000.0)
+            if m == -2:
+                raise RuntimeError("transport aborted")
+            if m < 0:
+                raise TimeoutError(
+                    "learner starved: no full rollout slot within 60 s "
+                    f"(got {len(slots)}/{self.k}) — actors dead or too slow")
+            slots.append(m)
+        return slots
+
+    def next_batch(self) -> Dict[str, torch.Tensor]:
+        """Returns time-major batch: frame (T+1, B, *obs), ...,
+        initial_core_state (2, B, H) if present."""
+        if not self.is_cuda:
+            return self._next_batch_cpu()
+        self._drain_pending()
+        i = self.idx
+        self.idx = (self.idx + 1) % 2
+        slots = self._pop_slots()
+        st = self.staging[i]
+        with torch.cuda.stream(self.copy_stream):
+            self.copy_stream.wait_event(self.consumed_ev[i])
+            for key, arena in self.buffers.arenas.items():
+                dst = st[key]
+                for j, m in enumerate(slots):
+                    dst[j].copy_(arena[m], non_blocking=True)
+            copy_done = torch.cuda.Event()
+            copy_done.record(self.copy_stream)
+        torch.cuda.current_stream().wait_event(copy_done)
+        self._pending.append((copy_done, slots))
+        batch = self._to_time_major(st)
+        batch["_staging_index"] = i
+        return batch
+
+    def mark_consumed(self, batch: Dict[str, torch.Tensor]):
+        """Record that the compute stream is done reading the staging
+        buffer backing `batch` (call after backward)."""
+        if self.is_cuda:
+            i = batch.pop("_staging_index")
+            self.consumed_ev[i].record()
+
+    def _to_time_major(self, st: Dict[str, torch.Tensor]):
+        out = {}
+        k, n = self.k, self.flags.envs_per_actor
+        for key, v in st.items():
+            if key == "initial_core_state":
+                # (k, 2, n, H) -> (2, k*n, H)
+                out[key] = v.permute(1, 0, 2, 3).reshape(2, k * n, -1)
+            else:
+                # (k, T+1, n, ...) -> (T+1, k*n, ...)
+                out[key] = v.transpose(0, 1).reshape(
+                    v.shape[1], k * n, *v.shape[3:])
+        return out
+
+    def _next_batch_cpu(self):
+        slots = self._pop_slots()
+        st = {key: torch.stack([arena[m] for m in slots])
+              for key, arena in self.buffers.arenas.items()}
+        for m in slots:
+            self.free_queue.push(m)
+        return self._to_time_major(st)
+
+    def shutdown(self):
+        self._drain_pending(block=True)
+
+
+def learn_step(flags: Flags, model, batch: Dict[str, torch.Tensor],
+               optimizer, scheduler, reducer: Optional[GradReducer],
+               timers: Optional[StageTimers] = None) -> Dict[str, float]:
+    """One IMPALA update on one batch."""
+    t = timers or StageTimers()
+    t.start("forward")
+    core_state = ()
+    if "initial_core_state" in batch:
+        ics = batch["initial_core_state"]
+        core_state = (ics[0], ics[1])
+    inputs = dict(frame=batch["frame"], reward=batch["reward"],
+                  done=batch["done"], last_action=batch["last_action"])
+    outputs, _ = model(inputs, core_state)
+
+    logits_all = outputs["policy_logits"]      # (T+1, B, A)
+    baseline_all = outputs["baseline"]         # (T+1, B)
+    bootstrap_value = baseline_all[-1]
+    target_logits = logits_all[:-1]
+    values = baseline_all[:-1]
+    actions = batch["action"][:-1]
+    behavior_logits = batch["behavior_logits"][:-1]
+    rewards = batch["reward"][1:]
+    if flags.reward_clipping == "abs_one":
+        rewards = torch.clamp(rewards, -1.0, 1.0)
+    discounts = (~batch["done"][1:]).float() * flags.discounting
+
+    loss, stats, vs, adv = ops.fused_vtrace_loss(
+        target_logits, values, bootstrap_value, behavior_logits,
+        actions, rewards, discounts,
+        clip_rho=flags.clip_rho_threshold, clip_c=flags.clip_c_threshold,
+        lambda_=flags.lambda_,
+        baseline_cost=flags.baseline_cost, entropy_cost=flags.entropy_cost,
+        force_eager=not flags.use_hip_kernels)
+
+    t.start("backward")
+    optimizer.zero_grad(set_to_none=False)
+    if reducer:
+        reducer.prepare()
+    loss.backward()
+    t.start("allreduce")
+    if reducer:
+        reducer.finalize()
+    t.start("optimizer")
+    torch.nn.utils.clip_grad_norm_(model.parameters(), flags.grad_norm_clipping)
+    optimizer.step()
+    scheduler.step()
+    t.stop()
+    return dict(total_loss=loss.detach(), **stats)
+
+
+def build_optimizer(flags: Flags, model, world_size: int = 1):
+    optimizer = torch.optim.RMSprop(
+        model.parameters(), lr=flags.learning_rate, alpha=flags.alpha,
+        momentum=flags.momentum, eps=flags.epsilon)
+    frames_per_step = flags.unroll_length * flags.batch_size * world_size
+
+    def lr_lambda(step):
+        return max(0.0, 1.0 - step * frames_per_step / flags.total_steps)
+
+    scheduler = torch.optim.lr_scheduler.LambdaLR(optimizer, lr_lambda)
+    return optimizer, scheduler
+
+
+def start_actors(flags: Flags, ctx, buffers, free_queue, full_queue, mirror,
+                 obs_shape, obs_dtype, num_actions) -> List[mp.Process]:
+    procs = []
+    for i in range(flags.num_actors):
+        p = ctx.Process(
+            target=act,
+            args=(flags, i, free_queue, full_queue, buffers, mirror,
+                  obs_shape, obs_dtype, num_actions),
+            daemon=True, name=f"actor-{i}")
+        p.start()
+        procs.append(p)
+    return procs
+
+
+def train(flags: Flags, max_learner_steps: Optional[int] = None,
+          on_step=None) -> Dict[str, float]:
+    """Full actor-learner training. Returns final stats."""
+    device = torch.device(flags.run_device)
+    rank, world = init_distributed(device.type)
+    if device.type == "cuda":
+        device = torch.device("cuda", torch.cuda.current_device())
+    torch.manual_seed(flags.seed + rank)
+
+    obs_shape, obs_dtype, num_actions = env_spec(flags)
+    model = build_model(flags, obs_shape, num_actions).to(device)
+    if world > 1:
+        for p in model.parameters():
+            torch.distributed.broadcast(p.data, src=0)
+
+    cpu_model = build_model(flags, obs_shape, num_actions)
+    cpu_model.load_state_dict({k: v.cpu() for k, v in model.state_dict().items()})
+    mirror = WeightMirror(cpu_model)
+
+    buffers = create_buffers(flags, obs_shape, obs_dtype, num_actions)
+    free_queue = SlotQueue(buffers.num_slots)
+    full_queue = SlotQueue(buffers.num_slots)
+    for m in range(buffers.num_slots):
+        free_queue.push(m)
+
+    ctx = mp.get_context("spawn")
+    actor_args = (flags, ctx, buffers, free_queue, full_queue, mirror,
+                  obs_shape, obs_dtype, num_actions)
+    procs = start_actors(*actor_args)
+
+    optimizer, scheduler = build_optimizer(flags, model, world)
+    reducer = GradReducer(model) if world > 1 else None
+    batcher = Batcher(flags, buffers, free_queue, full_queue, device)
+    metrics = MetricsLogger(flags.metrics_jsonl)
+    timers = StageTimers()
+
+    run_dir = None
+    ckpt_path = None
+    if flags.xpid and rank == 0:
+        run_dir = os.path.join(os.path.expanduser(flags.savedir), flags.xpid)
+        os.makedirs(run_dir, exist_ok=True)
+        ckpt_path = os.path.join(run_dir, "model.tar")
+
+    frames_per_batch = flags.unroll_length * flags.batch_size
+    step_frames = 0
+    learner_steps = 0
+    last_ckpt = time.time()
+    t_start = time.time()
+    last_log = t_start
+    last_log_frames = 0
+    stats: Dict[str, float] = {}
+    try:
+        while step_frames < flags.total_steps:
+            if max_learner_steps is not None and learner_steps >= max_learner_steps:
+                break
+            timers.start("rollout_wait")
+            batch = batcher.next_batch()
+            stats = learn_step(flags, model, batch, optimizer, scheduler,
+                               reducer, timers)
+            batcher.mark_consumed(batch)
+            learner_steps += 1
+            step_frames += frames_per_batch * world
+
+            if learner_steps % flags.weight_sync_interval == 0:
+                timers.start("weight_publish")
+                sd = model.state_dict()
+                cpu_model.load_state_dict(
+                    {k: v.detach().cpu() for k, v in sd.items()})
+                mirror.publish(cpu_model)
+                timers.stop()
+
+            # episode metrics (cheap on CPU; sync-inducing on GPU -> sparse)
+            if device.type != "cuda" or learner_steps % 20 == 0:
+                done = batch["done"][1:]
+                if done.any():
+                    rets = batch["episode_return"][1:][done]
+                    metrics.record_episodes(rets.cpu().tolist())
+
+            # respawn dead actors (SURVEY.md §5c failure detection)
+            for i, p in enumerate(procs):
+                if not p.is_alive():
+                    log.warning("actor %d died (exit %s); respawning",
+                                i, p.exitcode)
+                    procs[i] = ctx.Process(
+                        target=act,
+                        args=(flags, i, free_queue, full_queue, buffers,
+                              mirror, obs_shape, obs_dtype, num_actions),
+                        daemon=True, name=f"actor-{i}")
+                    procs[i].start()
+
+            now = time.time()
+            if rank == 0 and now - last_log > 5.0:
+                fps = (step_frames - last_log_frames) / (now - last_log)
+                mr = metrics.mean_return
+                log.info("step=%d fps=%.0f loss=%.3f return=%s",
+                         step_frames, fps,
+                         float(stats["total_loss"]),
+                         f"{mr:.2f}" if mr is not None else "n/a")
+                metrics.write(step=step_frames, fps=fps,
+                              mean_return=mr,
+                              timers=timers.drain(),
+                              **{k: float(v) for k, v in stats.items()})
+                last_log, last_log_frames = now, step_frames
+
+            if (ckpt_path and not flags.disable_checkpoint
+                    and now - last_ckpt > flags.checkpoint_interval_s):
+                ckpt_mod.save(ckpt_path, model, optimizer, scheduler,
+                              step_frames, flags)
+                last_ckpt = now
+            if on_step is not None:
+                on_step(learner_steps, step_frames, stats)
+    finally:
+        free_queue.abort()
+        full_queue.abort()
+        batcher.shutdown()
+        for p in procs:
+            p.join(timeout=5.0)
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+        metrics.close()
+    if ckpt_path and not flags.disable_checkpoint:
+        ckpt_mod.save(ckpt_path, model, optimizer, scheduler, step_frames, flags)
+    out = {k: float(v) for k, v in stats.items()}
+    out.update(step_frames=step_frames, learner_steps=learner_steps,
+               mean_return=metrics.mean_return or float("nan"),
+               fps=step_frames / max(time.time() - t_start, 1e-9))
+    return out
+
+
+def test(flags: Flags, num_episodes: int = 10, ckpt_path: Optional[str] = None
+         ) -> Dict[str, float]:
+    """Eval entry: load checkpoint, run greedy episodes."""
+    from torched_impala_amd.envs import make_env
+    obs_shape, obs_dtype, num_actions = env_spec(flags)
+    model = build_model(flags, obs_shape, num_actions)
+    model.eval()
+    if ckpt_path is None and flags.xpid:
+        ckpt_path = os.path.join(
+            os.path.expanduser(flags.savedir), flags.xpid, "model.tar")
+    if ckpt_path and os.path.exists(ckpt_path):
+        ckpt_mod.load(ckpt_path, model)
+    env = make_env(flags, seed=flags.seed)
+    returns = []
+    with torch.no_grad():
+        for ep in range(num_episodes):
+            obs = env.reset()
+            core_state = model.initial_state(1)
+            reward, done, last_action, ep_ret = 0.0, True, 0, 0.0
+            while True:
+                inputs = dict(
+                    frame=torch.from_numpy(obs)[None, None],
+                    reward=torch.tensor([[reward]], dtype=torch.float32),
+                    done=torch.tensor([[done]]),
+                    last_action=torch.tensor([[last_action]]))
+                out, core_state = model(inputs, core_state, greedy=True)
+                a = int(out["action"][0, 0])
+                obs, reward, done, _ = env.step(a)
+                ep_ret += reward
+                last_action = a
+                if done:
+                    break
+            returns.append(ep_ret)
+    log.info("test: %d episodes, mean return %.2f", num_episodes,
+             sum(returns) / len(returns))
+    return dict(mean_return=sum(returns) / len(returns),
+                returns=returns)

@@ -1,0 +1,132 @@
+"""Flagship benchmark (the driver contract).
+
+Measures the BASELINE.json:2 metric: env frames/sec at the learner on
+Atari-shaped 84x84x4 synthetic observations, batch 32 unrolls, unroll
+length 20, 32 CPU actor processes per learner GPU, random-init weights
+(BASELINE.json configs #2/#3). Whole system: real actor processes feed
+the shm ring buffer; the learner runs forward/V-trace/loss/backward/
+RMSProp and publishes weights every step.
+
+  python bench.py --gpus N --steps K --warmup W
+
+For N>1 the driver launches it under torch.distributed.run (one rank per
+GPU over RCCL); weak scaling — each rank runs its own 32 actors and
+batch 32. Frames are counted as AGENT steps (no action-repeat); the x4
+Atari convention is value*4 (also reported in config).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--env", default="synthetic-atari")
+    p.add_argument("--batch-size", type=int, default=32)
+    p.add_argument("--unroll-length", type=int, default=20)
+    p.add_argument("--num-actors", type=int, default=32)
+    p.add_argument("--envs-per-actor", type=int, default=1)
+    p.add_argument("--use-lstm", action="store_true", default=False)
+    p.add_argument("--model", default="auto")
+    p.add_argument("--no-hip-kernels", action="store_true")
+    p.add_argument("--device", default="auto")
+    args = p.parse_args()
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from torched_impala_amd.config import Flags
+    from torched_impala_amd import learner
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world != args.gpus and "RANK" in os.environ:
+        world = int(os.environ["WORLD_SIZE"])
+
+    flags = Flags(
+        env=args.env,
+        num_actors=args.num_actors,
+        batch_size=args.batch_size,
+        unroll_length=args.unroll_length,
+        envs_per_actor=args.envs_per_actor,
+        use_lstm=args.use_lstm,
+        model=args.model,
+        total_steps=1 << 60,
+        disable_checkpoint=True,
+        use_hip_kernels=not args.no_hip_kernels,
+        device=args.device,
+        weight_sync_interval=1,
+        seed=1234,
+    )
+
+    W, K = args.warmup, args.steps
+    state = {"t0": None, "t1": None}
+    is_cuda = flags.run_device == "cuda"
+
+    def barrier_sync():
+        if torch.distributed.is_available() and torch.distributed.is_initialized():
+            torch.distributed.barrier()
+        if is_cuda:
+            torch.cuda.synchronize()
+
+    def on_step(learner_steps, step_frames, stats):
+        if learner_steps == W:
+            barrier_sync()
+            state["t0"] = time.perf_counter()
+        elif learner_steps == W + K:
+            barrier_sync()
+            state["t1"] = time.perf_counter()
+
+    learner.train(flags, max_learner_steps=W + K, on_step=on_step)
+
+    elapsed = state["t1"] - state["t0"]
+    rank = int(os.environ.get("RANK", "0"))
+    # max over ranks
+    if torch.distributed.is_available() and torch.distributed.is_initialized():
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if is_cuda else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+    n_gpus = world if is_cuda else args.gpus
+
+    frames_per_batch = flags.unroll_length * flags.batch_size
+    total_frames = K * frames_per_batch * n_gpus
+    value = total_frames / elapsed
+    ms_per_step = elapsed / K * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "env frames/sec (learner) Atari 84x84x4 batch32 unroll20",
+            "value": value,
+            "unit": "frames/s",
+            "n_gpus": n_gpus,
+            "steps": K,
+            "warmup": W,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": ("shallow+lstm" if flags.use_lstm else "shallow"),
+                "global_batch": flags.batch_size * n_gpus,
+                "seq_len": flags.unroll_length,
+                "parallelism": f"dp{n_gpus}",
+                "num_actors_per_gpu": flags.num_actors,
+                "envs_per_actor": flags.envs_per_actor,
+                "frame_convention": "agent_steps",
+                "env_frames_per_sec_x4_convention": value * 4,
+                "obs": "84x84x4 uint8 synthetic",
+            },
+        }), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -93,20 +93,18 @@ class _FusedVtraceLoss(torch.autograd.Function):
                 clip_rho, clip_c, lambda_, baseline_cost, entropy_cost):
         ext = ext_hip()
         T, B, A = target_logits.shape
-        out = ext.vtrace_loss_fwd(
+        logp, target_lp, row_negent, vs, pg_adv = ext.vtrace_loss_fwd(
             target_logits.contiguous(), behavior_logits.contiguous(),
             actions.contiguous(), rewards.contiguous(),
             discounts.contiguous(), values.contiguous(),
             bootstrap_value.contiguous(),
             float(clip_rho), float(clip_c), float(lambda_))
-        probs, target_lp, entropy_sum, vs, pg_adv = out
         pg = -(target_lp * pg_adv.flatten()).sum()
-        vdiff = vs - values  # (T,B)
+        vdiff = vs - values.detach()  # (T,B)
         bl = 0.5 * (vdiff * vdiff).sum()
-        ent = entropy_sum  # scalar tensor: sum of p*logp (negative entropy)
+        ent = row_negent.sum()  # sum of p*logp (negative entropy)
         loss = pg + baseline_cost * bl + entropy_cost * ent
-        ctx.save_for_backward(probs, actions, pg_adv, vdiff)
-        ctx.dims = (T, B, A)
+        ctx.save_for_backward(logp, row_negent, actions, pg_adv, vdiff)
         ctx.coefs = (float(baseline_cost), float(entropy_cost))
         ctx.mark_non_differentiable(vs, pg_adv)
         return loss, pg.detach(), bl.detach(), ent.detach(), vs, pg_adv
@@ -114,13 +112,13 @@ class _FusedVtraceLoss(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss, *unused):
         ext = ext_hip()
-        probs, actions, pg_adv, vdiff = ctx.saved_tensors
-        T, B, A = ctx.dims
+        logp, row_negent, actions, pg_adv, vdiff = ctx.saved_tensors
         baseline_cost, entropy_cost = ctx.coefs
+        grad_scale = dloss.reshape(1).to(
+            device=logp.device, dtype=torch.float32).contiguous()
         dlogits, dvalues = ext.vtrace_loss_bwd(
-            probs, actions, pg_adv.contiguous(), vdiff.contiguous(),
-            float(dloss.item()) if dloss.dim() == 0 else 1.0,
-            baseline_cost, entropy_cost, T, B, A)
+            logp, row_negent, actions, pg_adv.contiguous(),
+            vdiff.contiguous(), grad_scale, baseline_cost, entropy_cost)
         return (dlogits, dvalues) + (None,) * 10
 
 

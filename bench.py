@@ -34,7 +34,7 @@ def main():
     p.add_argument("--batch-size", type=int, default=32)
     p.add_argument("--unroll-length", type=int, default=20)
     p.add_argument("--num-actors", type=int, default=32)
-    p.add_argument("--envs-per-actor", type=int, default=1)
+    p.add_argument("--envs-per-actor", type=int, default=8)
     p.add_argument("--use-lstm", action="store_true", default=False)
     p.add_argument("--model", default="auto")
     p.add_argument("--no-hip-kernels", action="store_true")

@@ -8,7 +8,7 @@
 #include <torch/extension.h>
 
 #include <c10/hip/HIPStream.h>
-#include <c10/hip/HIPGuard.h>
+
 #include <hip/hip_runtime.h>
 
 #include <vector>
@@ -30,6 +30,13 @@ void launch_vtrace_scan(const float*, const float*, const float*, const float*,
 void launch_vtrace_loss_bwd(const float*, const float*, const long*,
                             const float*, const float*, const float*, float,
                             float, int, int, float*, float*, hipStream_t);
+void launch_lstm_seq_fwd(const float*, const float*, const float*,
+                         const float*, const float*, int, int, float*, float*,
+                         float*, float*, float*, float*, int, hipStream_t);
+void launch_lstm_seq_bwd(const float*, const float*, const float*,
+                         const float*, const float*, const float*,
+                         const float*, const float*, int, int, float*, float*,
+                         float*, hipStream_t);
 }
 
 namespace {
@@ -43,6 +50,19 @@ inline void check_f32(const torch::Tensor& t, const char* name) {
 hipStream_t current_stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
+
+// torch-on-ROCm reports device type "cuda" (masquerading), which
+// c10::hip::HIPGuard rejects — use a plain hipSetDevice scope instead.
+struct DeviceScope {
+  int old_ = -1;
+  explicit DeviceScope(int dev) {
+    HIP_CHECK(hipGetDevice(&old_));
+    if (old_ != dev) HIP_CHECK(hipSetDevice(dev));
+  }
+  ~DeviceScope() {
+    if (old_ >= 0) (void)hipSetDevice(old_);
+  }
+};
 
 // ---------------------------------------------------------------------------
 // pinned-host registration (the "pinned-host ring buffer" half of the
@@ -91,7 +111,7 @@ std::vector<torch::Tensor> vtrace_loss_fwd(
   const int A = target_logits.size(2);
   const int N = T * B;
   TORCH_CHECK(A <= 4096, "action dim too large");
-  c10::hip::HIPGuard guard(target_logits.device());
+  DeviceScope guard(target_logits.get_device());
   auto opts = target_logits.options();
   auto logp = torch::empty({T, B, A}, opts);
   auto tgt_lp = torch::empty({N}, opts);
@@ -122,7 +142,7 @@ std::vector<torch::Tensor> vtrace_scan(
   check_f32(log_rhos, "log_rhos");
   const int T = log_rhos.size(0);
   const int B = log_rhos.size(1);
-  c10::hip::HIPGuard guard(log_rhos.device());
+  DeviceScope guard(log_rhos.get_device());
   auto vs = torch::empty_like(log_rhos);
   auto pg_adv = torch::empty_like(log_rhos);
   launch_vtrace_scan(
@@ -145,7 +165,7 @@ std::vector<torch::Tensor> vtrace_loss_bwd(
   const int B = logp.size(1);
   const int A = logp.size(2);
   const int N = T * B;
-  c10::hip::HIPGuard guard(logp.device());
+  DeviceScope guard(logp.get_device());
   auto dlogits = torch::empty({T, B, A}, logp.options());
   auto dvalues = torch::empty({T, B}, logp.options());
   launch_vtrace_loss_bwd(
@@ -158,6 +178,78 @@ std::vector<torch::Tensor> vtrace_loss_bwd(
   return {dlogits, dvalues};
 }
 
+// ---------------------------------------------------------------------------
+// LSTM core sequence (hidden fixed at 256 — the IMPALA core).
+// fwd: (Xg (T,B,1024), W_hhT (256,1024), notdone (T,B), h0, c0 (B,256),
+//       save_for_bwd) -> (H, act, C, Hp, hT, cT)
+// bwd: (dH, dhT, dcT, act, C, c0, notdone, W_hh (1024,256))
+//       -> (dG (T,B,1024), dh0, dc0)
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> lstm_seq_fwd(
+    torch::Tensor Xg, torch::Tensor W_hhT, torch::Tensor notdone,
+    torch::Tensor h0, torch::Tensor c0, bool save_for_bwd) {
+  check_f32(Xg, "Xg");
+  check_f32(W_hhT, "W_hhT");
+  check_f32(notdone, "notdone");
+  check_f32(h0, "h0");
+  check_f32(c0, "c0");
+  const int T = Xg.size(0);
+  const int B = Xg.size(1);
+  TORCH_CHECK(Xg.size(2) == 1024 && W_hhT.size(0) == 256 &&
+                  W_hhT.size(1) == 1024 && h0.size(1) == 256,
+              "lstm kernels are specialized for hidden=256");
+  DeviceScope guard(Xg.get_device());
+  auto opts = Xg.options();
+  auto H = torch::empty({T, B, 256}, opts);
+  auto hT = torch::empty({B, 256}, opts);
+  auto cT = torch::empty({B, 256}, opts);
+  torch::Tensor act, C, Hp;
+  if (save_for_bwd) {
+    act = torch::empty({T, B, 1024}, opts);
+    C = torch::empty({T, B, 256}, opts);
+    Hp = torch::empty({T, B, 256}, opts);
+  } else {
+    act = torch::empty({1}, opts);
+    C = torch::empty({1}, opts);
+    Hp = torch::empty({1}, opts);
+  }
+  launch_lstm_seq_fwd(Xg.data_ptr<float>(), W_hhT.data_ptr<float>(),
+                      notdone.data_ptr<float>(), h0.data_ptr<float>(),
+                      c0.data_ptr<float>(), T, B, H.data_ptr<float>(),
+                      act.data_ptr<float>(), C.data_ptr<float>(),
+                      Hp.data_ptr<float>(), hT.data_ptr<float>(),
+                      cT.data_ptr<float>(), save_for_bwd ? 1 : 0,
+                      current_stream());
+  HIP_CHECK(hipGetLastError());
+  return {H, act, C, Hp, hT, cT};
+}
+
+std::vector<torch::Tensor> lstm_seq_bwd(
+    torch::Tensor dH, torch::Tensor dhT, torch::Tensor dcT,
+    torch::Tensor act, torch::Tensor C, torch::Tensor c0,
+    torch::Tensor notdone, torch::Tensor W_hh) {
+  check_f32(dH, "dH");
+  check_f32(act, "act");
+  check_f32(W_hh, "W_hh");
+  const int T = dH.size(0);
+  const int B = dH.size(1);
+  TORCH_CHECK(dH.size(2) == 256 && W_hh.size(0) == 1024 && W_hh.size(1) == 256,
+              "lstm kernels are specialized for hidden=256");
+  DeviceScope guard(dH.get_device());
+  auto opts = dH.options();
+  auto dG = torch::empty({T, B, 1024}, opts);
+  auto dh0 = torch::empty({B, 256}, opts);
+  auto dc0 = torch::empty({B, 256}, opts);
+  launch_lstm_seq_bwd(dH.data_ptr<float>(), dhT.data_ptr<float>(),
+                      dcT.data_ptr<float>(), act.data_ptr<float>(),
+                      C.data_ptr<float>(), c0.data_ptr<float>(),
+                      notdone.data_ptr<float>(), W_hh.data_ptr<float>(), T, B,
+                      dG.data_ptr<float>(), dh0.data_ptr<float>(),
+                      dc0.data_ptr<float>(), current_stream());
+  HIP_CHECK(hipGetLastError());
+  return {dG, dh0, dc0};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -167,4 +259,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vtrace_loss_fwd", &vtrace_loss_fwd);
   m.def("vtrace_scan", &vtrace_scan);
   m.def("vtrace_loss_bwd", &vtrace_loss_bwd);
+  m.def("lstm_seq_fwd", &lstm_seq_fwd);
+  m.def("lstm_seq_bwd", &lstm_seq_bwd);
 }

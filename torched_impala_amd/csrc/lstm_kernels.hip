@@ -1,0 +1,229 @@
+// gfx950 LSTM-core sequence kernels (BASELINE.json:5 "LSTM cell" kernel).
+//
+// The IMPALA core is LSTMCell(hidden=256) iterated over the unroll with
+// done-mask state resets (SURVEY.md §2c row lstm_cell: "the
+// latency-critical kernel" — sequential over T). Eager PyTorch runs
+// ~2 kernel launches per step plus Python overhead; here the WHOLE
+// unroll is ONE kernel launch per direction:
+//
+//   * the input-to-gate GEMM for all T steps is hoisted into one batched
+//     GEMM outside (Xg = x @ W_ih^T + b_ih + b_hh, torch/rocBLAS),
+//   * the sequential recurrence partitions by BATCH ROW: one workgroup
+//     owns one batch row for the whole T loop, so there are NO
+//     inter-workgroup dependencies, no grid barriers, no weight
+//     re-publication — h/c live in LDS across steps and W_hh streams
+//     from L2 (1 MiB, read by every workgroup, L2/L3-resident).
+//
+// Layout choices (CDNA4): the forward reads W_hh TRANSPOSED (256, 1024)
+// so that lane j's float2 gate-column reads are coalesced; the backward
+// reads W_hh in its native (1024, 256) layout so lane k's dh_prev reads
+// are coalesced. 512 threads (8 waves) per workgroup; LDS ~12 KiB.
+//
+// dW_hh / dW_ih / db are NOT computed here: the backward kernel emits
+// dGates (T,B,4H), and the host reduces them with two big GEMMs
+// (dW_hh = dG^T @ Hp, dXg = dG feeding autograd) — that work is
+// GEMM-shaped and belongs on MFMA via the BLAS path, not in this
+// latency-critical loop.
+
+#include <hip/hip_runtime.h>
+
+namespace impala {
+
+constexpr int H = 256;        // hidden size (IMPALA core is fixed at 256)
+constexpr int G4 = 4 * H;     // 1024 gate pre-activations
+constexpr int THREADS = 512;  // 8 waves
+
+__device__ __forceinline__ float sigmoidf(float x) {
+  return 1.0f / (1.0f + __expf(-x));
+}
+
+// ---------------------------------------------------------------------------
+// forward: grid = (B) blocks x 512 threads
+//   Xg      (T, B, 1024)  x-part of gates incl. both biases
+//   W_hhT   (256, 1024)   W_hh transposed
+//   notdone (T, B)        1 - done, fp32
+//   h0, c0  (B, 256)
+// outputs:
+//   Hout    (T, B, 256)   h_t
+//   act     (T, B, 1024)  post-activation i,f,g,o (saved for bwd)
+//   Cout    (T, B, 256)   c_t (saved for bwd)
+//   Hp      (T, B, 256)   masked h input at step t (saved for bwd + dW_hh)
+//   hT, cT  (B, 256)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(THREADS)
+lstm_seq_fwd(const float* __restrict__ Xg, const float* __restrict__ W_hhT,
+             const float* __restrict__ notdone, const float* __restrict__ h0,
+             const float* __restrict__ c0, int T, int B,
+             float* __restrict__ Hout, float* __restrict__ act,
+             float* __restrict__ Cout, float* __restrict__ Hp,
+             float* __restrict__ hT, float* __restrict__ cT,
+             int save_for_bwd) {
+  __shared__ float s_h[H];
+  __shared__ float s_cm[H];
+  __shared__ float s_c[H];
+  __shared__ float s_g[G4];
+
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  if (tid < H) {
+    s_h[tid] = h0[b * H + tid];
+    s_c[tid] = c0[b * H + tid];
+  }
+  __syncthreads();
+
+  for (int t = 0; t < T; ++t) {
+    const float nd = notdone[t * B + b];
+    const size_t rowH = ((size_t)t * B + b) * H;
+    const size_t rowG = ((size_t)t * B + b) * G4;
+    if (tid < H) {
+      const float hm = s_h[tid] * nd;
+      s_h[tid] = hm;  // masked input h' for this step
+      s_cm[tid] = s_c[tid] * nd;
+      if (save_for_bwd) Hp[rowH + tid] = hm;
+    }
+    __syncthreads();
+    // gate GEMV: thread j computes gate columns (2j, 2j+1)
+    {
+      float acc0 = Xg[rowG + 2 * tid];
+      float acc1 = Xg[rowG + 2 * tid + 1];
+      const float2* wrow = reinterpret_cast<const float2*>(W_hhT) + tid;
+      #pragma unroll 8
+      for (int k = 0; k < H; ++k) {
+        const float hk = s_h[k];
+        const float2 w = wrow[k * (G4 / 2)];
+        acc0 = fmaf(hk, w.x, acc0);
+        acc1 = fmaf(hk, w.y, acc1);
+      }
+      s_g[2 * tid] = acc0;
+      s_g[2 * tid + 1] = acc1;
+    }
+    __syncthreads();
+    if (tid < H) {
+      const float gi = sigmoidf(s_g[tid]);
+      const float gf = sigmoidf(s_g[H + tid]);
+      const float gg = tanhf(s_g[2 * H + tid]);
+      const float go = sigmoidf(s_g[3 * H + tid]);
+      const float c = fmaf(gf, s_cm[tid], gi * gg);
+      const float h = go * tanhf(c);
+      if (save_for_bwd) {
+        act[rowG + tid] = gi;
+        act[rowG + H + tid] = gf;
+        act[rowG + 2 * H + tid] = gg;
+        act[rowG + 3 * H + tid] = go;
+        Cout[rowH + tid] = c;
+      }
+      Hout[rowH + tid] = h;
+      s_h[tid] = h;
+      s_c[tid] = c;
+    }
+    __syncthreads();
+  }
+  if (tid < H) {
+    hT[b * H + tid] = s_h[tid];
+    cT[b * H + tid] = s_c[tid];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward: grid = (B) blocks x 512 threads, reverse time loop.
+// Emits dG (T,B,1024) and dh0/dc0; weight grads are host-side GEMMs.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(THREADS)
+lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
+             const float* __restrict__ dcT, const float* __restrict__ act,
+             const float* __restrict__ Cout, const float* __restrict__ c0,
+             const float* __restrict__ notdone,
+             const float* __restrict__ W_hh,  // (1024, 256) native layout
+             int T, int B, float* __restrict__ dG,
+             float* __restrict__ dh0, float* __restrict__ dc0) {
+  __shared__ float s_dh[H];
+  __shared__ float s_dc[H];
+  __shared__ float s_dg[G4];
+  __shared__ float s_part[THREADS];
+
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  if (tid < H) {
+    s_dh[tid] = dhT[b * H + tid];
+    s_dc[tid] = dcT[b * H + tid];
+  }
+  __syncthreads();
+
+  for (int t = T - 1; t >= 0; --t) {
+    const float nd = notdone[t * B + b];
+    const size_t rowH = ((size_t)t * B + b) * H;
+    const size_t rowG = ((size_t)t * B + b) * G4;
+    if (tid < H) {
+      const float c_t = Cout[rowH + tid];
+      const float tc = tanhf(c_t);
+      const float cprev =
+          (t > 0 ? Cout[rowH + tid - (size_t)B * H] : c0[b * H + tid]) * nd;
+      const float gi = act[rowG + tid];
+      const float gf = act[rowG + H + tid];
+      const float gg = act[rowG + 2 * H + tid];
+      const float go = act[rowG + 3 * H + tid];
+      const float dh = dH[rowH + tid] + s_dh[tid];
+      const float dct = s_dc[tid] + dh * go * (1.f - tc * tc);
+      const float dgi = (dct * gg) * gi * (1.f - gi);
+      const float dgf = (dct * cprev) * gf * (1.f - gf);
+      const float dgg = (dct * gi) * (1.f - gg * gg);
+      const float dgo = (dh * tc) * go * (1.f - go);
+      s_dg[tid] = dgi;
+      s_dg[H + tid] = dgf;
+      s_dg[2 * H + tid] = dgg;
+      s_dg[3 * H + tid] = dgo;
+      dG[rowG + tid] = dgi;
+      dG[rowG + H + tid] = dgf;
+      dG[rowG + 2 * H + tid] = dgg;
+      dG[rowG + 3 * H + tid] = dgo;
+      s_dc[tid] = dct * gf * nd;  // dc_{t-1} (masked)
+    }
+    __syncthreads();
+    // dh_prev[k] = sum_j s_dg[j] * W_hh[j, k]; split j over two thread halves
+    {
+      const int k = tid & (H - 1);
+      const int j0 = (tid < H) ? 0 : (G4 / 2);
+      float acc = 0.f;
+      #pragma unroll 8
+      for (int j = 0; j < G4 / 2; ++j) {
+        acc = fmaf(s_dg[j0 + j], W_hh[(size_t)(j0 + j) * H + k], acc);
+      }
+      s_part[tid] = acc;
+    }
+    __syncthreads();
+    if (tid < H) {
+      s_dh[tid] = (s_part[tid] + s_part[H + tid]) * nd;
+    }
+    __syncthreads();
+  }
+  if (tid < H) {
+    dh0[b * H + tid] = s_dh[tid];
+    dc0[b * H + tid] = s_dc[tid];
+  }
+}
+
+}  // namespace impala
+
+extern "C" {
+
+void launch_lstm_seq_fwd(const float* Xg, const float* W_hhT,
+                         const float* notdone, const float* h0,
+                         const float* c0, int T, int B, float* Hout,
+                         float* act, float* Cout, float* Hp, float* hT,
+                         float* cT, int save_for_bwd, hipStream_t stream) {
+  hipLaunchKernelGGL(impala::lstm_seq_fwd, dim3(B), dim3(impala::THREADS), 0,
+                     stream, Xg, W_hhT, notdone, h0, c0, T, B, Hout, act,
+                     Cout, Hp, hT, cT, save_for_bwd);
+}
+
+void launch_lstm_seq_bwd(const float* dH, const float* dhT, const float* dcT,
+                         const float* act, const float* Cout, const float* c0,
+                         const float* notdone, const float* W_hh, int T,
+                         int B, float* dG, float* dh0, float* dc0,
+                         hipStream_t stream) {
+  hipLaunchKernelGGL(impala::lstm_seq_bwd, dim3(B), dim3(impala::THREADS), 0,
+                     stream, dH, dhT, dcT, act, Cout, c0, notdone, W_hh, T, B,
+                     dG, dh0, dc0);
+}
+
+}  // extern "C"

@@ -261,6 +261,7 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
 
     obs_shape, obs_dtype, num_actions = env_spec(flags)
     model = build_model(flags, obs_shape, num_actions).to(device)
+    model.force_eager_core = not flags.use_hip_kernels
     if world > 1:
         for p in model.parameters():
             torch.distributed.broadcast(p.data, src=0)

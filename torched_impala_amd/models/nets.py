@@ -61,6 +61,13 @@ class _CoreMixin(nn.Module):
         if not self.use_lstm:
             return _flatten_time(core_input), tuple()
         T, B = core_input.shape[:2]
+        if (core_input.is_cuda and T > 1 and self.core.hidden_size == 256
+                and not getattr(self, "force_eager_core", False)):
+            from torched_impala_amd import ops
+            if ops.hip_available():
+                from torched_impala_amd.ops.lstm import lstm_core_forward
+                return lstm_core_forward(self.core, core_input, done,
+                                         core_state)
         notdone = (~done).to(core_input.dtype)
         outputs = []
         h, c = core_state

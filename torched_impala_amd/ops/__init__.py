@@ -137,7 +137,9 @@ def fused_vtrace_loss(target_logits, values, bootstrap_value,
             target_logits, values, bootstrap_value, behavior_logits,
             actions, rewards, discounts, clip_rho, clip_c, lambda_,
             baseline_cost, entropy_cost)
-        return loss, dict(pg_loss=pg, baseline_loss=bl, entropy_loss=ent), vs, adv
+        stats = dict(pg_loss=pg.detach(), baseline_loss=bl.detach(),
+                     entropy_loss=ent.detach())
+        return loss, stats, vs, adv
 
     # eager composition (CPU tests / escape hatch)
     from torched_impala_amd.algo import vtrace as VT, losses as L

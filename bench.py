@@ -83,7 +83,10 @@ def main():
             barrier_sync()
             state["t1"] = time.perf_counter()
 
-    learner.train(flags, max_learner_steps=W + K, on_step=on_step)
+    train_stats = learner.train(flags, max_learner_steps=W + K,
+                                on_step=on_step)
+    print("[bench] stage timers (whole run):",
+          train_stats.get("timers"), file=sys.stderr)
 
     elapsed = state["t1"] - state["t0"]
     rank = int(os.environ.get("RANK", "0"))

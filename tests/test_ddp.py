@@ -108,3 +108,116 @@ def test_gradreducer_matches_mean_of_local_grads():
     for p in procs:
         p.join(timeout=30)
     torch.testing.assert_close(torch.from_numpy(got), expected, rtol=1e-6, atol=1e-7)
+
+
+def _worker_flat(rank, world, port, q):
+    """FlatRMSprop + FlatGradReducer: ranks converge to identical params."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from torched_impala_amd.ops.optim import FlatRMSprop
+        from torched_impala_amd.parallel import FlatGradReducer
+        torch.manual_seed(0)
+        model = MLPPolicy((4,), 2)
+        opt = FlatRMSprop(model.parameters(), lr=0.01, max_grad_norm=40.0,
+                          world_size=world)
+        reducer = FlatGradReducer(model, opt, bucket_bytes=1 << 10)
+        for it in range(3):
+            torch.manual_seed(1000 * (rank + 1) + it)  # different data
+            x = torch.randn(2, 3, 4)
+            inputs = dict(frame=x, reward=torch.zeros(2, 3),
+                          done=torch.zeros(2, 3, dtype=torch.bool),
+                          last_action=torch.zeros(2, 3, dtype=torch.long))
+            opt.zero_grad()
+            reducer.prepare()
+            out, _ = model(inputs, ())
+            (out["policy_logits"].pow(2).sum()
+             + out["baseline"].pow(2).sum()).backward()
+            reducer.finalize()
+            opt.step()
+        q.put((rank, opt.flat_param.numpy().copy()))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_flat_reducer_gloo_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_flat, args=(r, 2, 29617, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(2):
+        rank, flat = q.get(timeout=120)
+        res[rank] = flat
+    for p in procs:
+        p.join(timeout=30)
+    # identical data-parallel updates on both ranks
+    torch.testing.assert_close(torch.from_numpy(res[0]),
+                               torch.from_numpy(res[1]),
+                               rtol=1e-6, atol=1e-7)
+
+
+def _worker_flat_ref(rank, world, port, q):
+    """Flat DP world-2 result == single-process run on the combined data."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from torched_impala_amd.ops.optim import FlatRMSprop
+        from torched_impala_amd.parallel import FlatGradReducer
+        torch.manual_seed(0)
+        model = MLPPolicy((4,), 2)
+        opt = FlatRMSprop(model.parameters(), lr=0.01, max_grad_norm=0.0,
+                          world_size=world)
+        reducer = FlatGradReducer(model, opt)
+        x = torch.full((1, 1, 4), float(rank + 1))
+        inputs = dict(frame=x, reward=torch.zeros(1, 1),
+                      done=torch.zeros(1, 1, dtype=torch.bool),
+                      last_action=torch.zeros(1, 1, dtype=torch.long))
+        opt.zero_grad()
+        reducer.prepare()
+        out, _ = model(inputs, ())
+        out["baseline"].sum().backward()
+        reducer.finalize()
+        opt.step()
+        if rank == 0:
+            q.put(opt.flat_param.numpy().copy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_flat_reducer_equals_mean_gradient_update():
+    from torched_impala_amd.ops.optim import FlatRMSprop
+    # reference: single process, grad = mean of the two ranks' grads
+    torch.manual_seed(0)
+    ref = MLPPolicy((4,), 2)
+    opt_ref = FlatRMSprop(ref.parameters(), lr=0.01, max_grad_norm=0.0)
+    grads = []
+    for val in (1.0, 2.0):
+        opt_ref.zero_grad()
+        inputs = dict(frame=torch.full((1, 1, 4), val),
+                      reward=torch.zeros(1, 1),
+                      done=torch.zeros(1, 1, dtype=torch.bool),
+                      last_action=torch.zeros(1, 1, dtype=torch.long))
+        out, _ = ref(inputs, ())
+        out["baseline"].sum().backward()
+        grads.append(opt_ref.flat_grad.clone())
+    opt_ref.flat_grad.copy_((grads[0] + grads[1]) / 2)
+    opt_ref.step()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_flat_ref, args=(r, 2, 29618, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=120)
+    for p in procs:
+        p.join(timeout=30)
+    torch.testing.assert_close(torch.from_numpy(got), opt_ref.flat_param,
+                               rtol=1e-6, atol=1e-7)

@@ -49,7 +49,7 @@ class Rollout:
         self.carry = None
         self._advance()  # produce the first carried row
 
-    @torch.no_grad()
+    @torch.inference_mode()
     def _forward(self):
         inputs = dict(
             frame=self.frame.unsqueeze(0),

@@ -63,7 +63,8 @@ def build_ring(force: bool = False) -> str:
     return out
 
 
-HIP_SOURCES = ["hip_ext.cpp", "vtrace_kernels.hip", "lstm_kernels.hip"]
+HIP_SOURCES = ["hip_ext.cpp", "vtrace_kernels.hip", "lstm_kernels.hip",
+               "optim_kernels.hip"]
 
 
 def build_hip(force: bool = False) -> str:

@@ -28,8 +28,11 @@ log = logging.getLogger(__name__)
 def save(path: str, model, optimizer, scheduler, step: int, flags):
     os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
     tmp = path + ".tmp"
+    # clone: FlatRMSprop params are views into one flat storage — cloning
+    # keeps each checkpoint entry self-contained (and small)
     torch.save({
-        "model_state_dict": model.state_dict(),
+        "model_state_dict": {k: v.detach().clone()
+                             for k, v in model.state_dict().items()},
         "optimizer_state_dict": optimizer.state_dict() if optimizer else None,
         "scheduler_state_dict": scheduler.state_dict() if scheduler else None,
         "step": int(step),

@@ -37,6 +37,11 @@ void launch_lstm_seq_bwd(const float*, const float*, const float*,
                          const float*, const float*, const float*,
                          const float*, const float*, int, int, float*, float*,
                          float*, hipStream_t);
+void launch_grad_sqnorm(const float*, long, const float*, float*,
+                        hipStream_t);
+void launch_rmsprop_update(float*, const float*, float*, long, const float*,
+                           const float*, const float*, float, float, float,
+                           hipStream_t);
 }
 
 namespace {
@@ -250,6 +255,36 @@ std::vector<torch::Tensor> lstm_seq_bwd(
   return {dG, dh0, dc0};
 }
 
+// ---------------------------------------------------------------------------
+// fused flat-arena optimizer: clip-by-global-norm + RMSProp in 2 kernels.
+// norm_sq must be zeroed by the caller (a cheap zero_ on the stream).
+// lr and scale are device scalars so the whole step is graph-capturable.
+// ---------------------------------------------------------------------------
+void fused_clip_rmsprop(torch::Tensor flat_p, torch::Tensor flat_g,
+                        torch::Tensor flat_sqavg, torch::Tensor lr,
+                        torch::Tensor norm_sq, torch::Tensor scale,
+                        double max_norm, double alpha, double eps) {
+  check_f32(flat_p, "flat_p");
+  check_f32(flat_g, "flat_g");
+  check_f32(flat_sqavg, "flat_sqavg");
+  check_f32(lr, "lr");
+  check_f32(norm_sq, "norm_sq");
+  check_f32(scale, "scale");
+  const long n = flat_p.numel();
+  DeviceScope guard(flat_p.get_device());
+  hipStream_t stream = current_stream();
+  if (max_norm > 0.0) {
+    launch_grad_sqnorm(flat_g.data_ptr<float>(), n, scale.data_ptr<float>(),
+                       norm_sq.data_ptr<float>(), stream);
+  }
+  launch_rmsprop_update(flat_p.data_ptr<float>(), flat_g.data_ptr<float>(),
+                        flat_sqavg.data_ptr<float>(), n,
+                        lr.data_ptr<float>(), norm_sq.data_ptr<float>(),
+                        scale.data_ptr<float>(), (float)max_norm,
+                        (float)alpha, (float)eps, stream);
+  HIP_CHECK(hipGetLastError());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -261,4 +296,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vtrace_loss_bwd", &vtrace_loss_bwd);
   m.def("lstm_seq_fwd", &lstm_seq_fwd);
   m.def("lstm_seq_bwd", &lstm_seq_bwd);
+  m.def("fused_clip_rmsprop", &fused_clip_rmsprop);
 }

@@ -27,7 +27,9 @@ from torched_impala_amd.config import Flags
 from torched_impala_amd.envs import env_spec
 from torched_impala_amd.metrics import MetricsLogger, StageTimers
 from torched_impala_amd.models import build_model
-from torched_impala_amd.parallel import GradReducer, init_distributed
+from torched_impala_amd.ops.optim import FlatRMSprop
+from torched_impala_amd.parallel import (FlatGradReducer, GradReducer,
+                                         init_distributed)
 from torched_impala_amd.transport import SlotQueue, TrajectoryBuffers, WeightMirror
 
 log = logging.getLogger(__name__)
@@ -216,17 +218,27 @@ def learn_step(flags: Flags, model, batch: Dict[str, torch.Tensor],
     if reducer:
         reducer.finalize()
     t.start("optimizer")
-    torch.nn.utils.clip_grad_norm_(model.parameters(), flags.grad_norm_clipping)
-    optimizer.step()
+    if isinstance(optimizer, FlatRMSprop):
+        optimizer.step()  # fused: clip-by-global-norm folded in
+    else:
+        torch.nn.utils.clip_grad_norm_(model.parameters(),
+                                       flags.grad_norm_clipping)
+        optimizer.step()
     scheduler.step()
     t.stop()
     return dict(total_loss=loss.detach(), **stats)
 
 
 def build_optimizer(flags: Flags, model, world_size: int = 1):
-    optimizer = torch.optim.RMSprop(
-        model.parameters(), lr=flags.learning_rate, alpha=flags.alpha,
-        momentum=flags.momentum, eps=flags.epsilon)
+    if flags.momentum == 0.0:
+        optimizer = FlatRMSprop(
+            model.parameters(), lr=flags.learning_rate, alpha=flags.alpha,
+            eps=flags.epsilon, max_grad_norm=flags.grad_norm_clipping,
+            world_size=world_size)
+    else:  # momentum needs the torch path (flat kernel is momentum-free)
+        optimizer = torch.optim.RMSprop(
+            model.parameters(), lr=flags.learning_rate, alpha=flags.alpha,
+            momentum=flags.momentum, eps=flags.epsilon)
     frames_per_step = flags.unroll_length * flags.batch_size * world_size
 
     def lr_lambda(step):
@@ -257,6 +269,8 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     rank, world = init_distributed(device.type)
     if device.type == "cuda":
         device = torch.device("cuda", torch.cuda.current_device())
+        # don't let the learner's CPU-side OMP pool fight the actors
+        torch.set_num_threads(min(8, os.cpu_count() or 8))
     torch.manual_seed(flags.seed + rank)
 
     obs_shape, obs_dtype, num_actions = env_spec(flags)
@@ -282,7 +296,12 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     procs = start_actors(*actor_args)
 
     optimizer, scheduler = build_optimizer(flags, model, world)
-    reducer = GradReducer(model) if world > 1 else None
+    if world > 1:
+        reducer = (FlatGradReducer(model, optimizer)
+                   if isinstance(optimizer, FlatRMSprop)
+                   else GradReducer(model))
+    else:
+        reducer = None
     batcher = Batcher(flags, buffers, free_queue, full_queue, device)
     metrics = MetricsLogger(flags.metrics_jsonl)
     timers = StageTimers()
@@ -316,10 +335,7 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
 
             if learner_steps % flags.weight_sync_interval == 0:
                 timers.start("weight_publish")
-                sd = model.state_dict()
-                cpu_model.load_state_dict(
-                    {k: v.detach().cpu() for k, v in sd.items()})
-                mirror.publish(cpu_model)
+                mirror.publish_state_dict(model.state_dict())
                 timers.stop()
 
             # episode metrics (cheap on CPU; sync-inducing on GPU -> sparse)
@@ -377,7 +393,8 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     out = {k: float(v) for k, v in stats.items()}
     out.update(step_frames=step_frames, learner_steps=learner_steps,
                mean_return=metrics.mean_return or float("nan"),
-               fps=step_frames / max(time.time() - t_start, 1e-9))
+               fps=step_frames / max(time.time() - t_start, 1e-9),
+               timers={k: round(v, 4) for k, v in timers.drain().items()})
     return out
 
 

@@ -21,8 +21,9 @@ class _LstmSeq(torch.autograd.Function):
     @staticmethod
     def forward(ctx, Xg, W_hh, h0, c0, notdone):
         ext = ops.ext_hip()
-        need_bwd = torch.is_grad_enabled() and (
-            Xg.requires_grad or W_hh.requires_grad)
+        # NB: grad mode is disabled inside Function.forward —
+        # needs_input_grad is the correct signal here.
+        need_bwd = ctx.needs_input_grad[0] or ctx.needs_input_grad[1]
         W_hhT = W_hh.detach().t().contiguous()
         H, act, C, Hp, hT, cT = ext.lstm_seq_fwd(
             Xg.contiguous(), W_hhT, notdone.contiguous(),

@@ -1,3 +1,5 @@
-from torched_impala_amd.parallel.ddp import GradReducer, init_distributed
+from torched_impala_amd.parallel.ddp import (
+    FlatGradReducer, GradReducer, init_distributed,
+)
 
-__all__ = ["GradReducer", "init_distributed"]
+__all__ = ["FlatGradReducer", "GradReducer", "init_distributed"]

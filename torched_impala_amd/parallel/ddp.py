@@ -39,6 +39,75 @@ def init_distributed(device: str) -> tuple[int, int]:
     return rank, dist.get_world_size()
 
 
+class FlatGradReducer:
+    """Zero-copy bucketed all-reduce over a FlatRMSprop gradient arena.
+
+    Buckets are contiguous slices of ``optimizer.flat_grad`` aligned to
+    parameter boundaries in flat (reverse-parameter ≈ backward) order;
+    post-accumulate-grad hooks launch each bucket's async all-reduce the
+    moment its last grad lands, overlapping the rest of backward. No
+    copy-in/copy-out; the 1/world mean is folded into the optimizer's
+    device-side scale (FlatRMSprop(world_size=...)).
+    """
+
+    def __init__(self, model, optimizer, bucket_bytes: int = 25 << 20,
+                 process_group=None):
+        del model
+        self.group = process_group
+        self.opt = optimizer
+        self.world_size = dist.get_world_size(process_group) if (
+            dist.is_available() and dist.is_initialized()) else 1
+        self.enabled = self.world_size > 1
+        self.buckets = []
+        self._param_bucket = {}
+        if not self.enabled:
+            return
+        flat = optimizer.flat_grad
+        cur_params, cur_start, cur_end = [], 0, 0
+        for p in optimizer._order:
+            off, n = optimizer.slices[p]
+            cur_params.append(p)
+            cur_end = off + n
+            if (cur_end - cur_start) * 4 >= bucket_bytes:
+                self.buckets.append(
+                    dict(params=list(cur_params), view=flat[cur_start:cur_end],
+                         ready=0, work=None))
+                cur_params, cur_start = [], cur_end
+        if cur_params:
+            self.buckets.append(
+                dict(params=list(cur_params), view=flat[cur_start:cur_end],
+                     ready=0, work=None))
+        for b in self.buckets:
+            for p in b["params"]:
+                self._param_bucket[p] = b
+                p.register_post_accumulate_grad_hook(self._hook)
+        log.info("FlatGradReducer: %d buckets, world=%d", len(self.buckets),
+                 self.world_size)
+
+    def _hook(self, p):
+        b = self._param_bucket[p]
+        b["ready"] += 1
+        if b["ready"] == len(b["params"]):
+            b["work"] = dist.all_reduce(b["view"], op=dist.ReduceOp.SUM,
+                                        group=self.group, async_op=True)
+
+    def prepare(self):
+        if not self.enabled:
+            return
+        for b in self.buckets:
+            b["ready"], b["work"] = 0, None
+
+    def finalize(self):
+        if not self.enabled:
+            return
+        for b in self.buckets:
+            if b["work"] is None:  # keep ranks collectively consistent
+                b["work"] = dist.all_reduce(b["view"], op=dist.ReduceOp.SUM,
+                                            group=self.group, async_op=True)
+        for b in self.buckets:
+            b["work"].wait()
+
+
 class _Bucket:
     __slots__ = ("params", "numel", "flat", "work", "ready")
 

@@ -40,6 +40,16 @@ class WeightMirror:
         self._write((v + 1) % 2, model)
         self.version.fill_(v + 1)
 
+    def publish_state_dict(self, sd):
+        """Direct publish from a (possibly GPU) state dict — one D2H copy
+        per tensor straight into the shared mirror, no CPU model hop."""
+        v = int(self.version.item())
+        b = (v + 1) % 2
+        with torch.no_grad():
+            for k, dst in zip(self.keys, self._bufs[b]):
+                dst.copy_(sd[k].detach(), non_blocking=False)
+        self.version.fill_(v + 1)
+
     def pull_into(self, model: torch.nn.Module, last_seen: int = -1) -> int:
         """Called by actors. Returns the version pulled (or last_seen if
         already current)."""

@@ -38,6 +38,7 @@ def main():
     p.add_argument("--use-lstm", action="store_true", default=False)
     p.add_argument("--model", default="auto")
     p.add_argument("--no-hip-kernels", action="store_true")
+    p.add_argument("--no-graphs", action="store_true")
     p.add_argument("--device", default="auto")
     args = p.parse_args()
 
@@ -60,6 +61,7 @@ def main():
         total_steps=1 << 60,
         disable_checkpoint=True,
         use_hip_kernels=not args.no_hip_kernels,
+        graph_capture=not args.no_graphs,
         device=args.device,
         weight_sync_interval=1,
         seed=1234,

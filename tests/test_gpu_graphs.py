@@ -1,0 +1,108 @@
+"""hipGraph-captured learn step == eager learn steps (bit-for-bit device
+math on identical staged batches) — marked gpu."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from torched_impala_amd.config import Flags
+from torched_impala_amd import learner as L
+from torched_impala_amd.models import build_model
+from torched_impala_amd.ops.optim import FlatRMSprop
+
+
+class _StubBatcher:
+    """Quacks like Batcher for GraphedStep: staging dict + _to_time_major."""
+
+    def __init__(self, flags, staging):
+        self.flags = flags
+        self.k = flags.batch_size // flags.envs_per_actor
+        self.staging = {0: staging}
+
+    _to_time_major = L.Batcher._to_time_major
+
+
+def _staging(flags, obs, A, device, seed=0):
+    torch.manual_seed(seed)
+    T = flags.unroll_length
+    k, n = flags.batch_size, 1
+    st = dict(
+        frame=torch.randint(0, 256, (k, T + 1, n, *obs), dtype=torch.uint8,
+                            device=device),
+        reward=torch.randn(k, T + 1, n, device=device),
+        done=torch.rand(k, T + 1, n, device=device) < 0.05,
+        action=torch.randint(0, A, (k, T + 1, n), device=device),
+        last_action=torch.randint(0, A, (k, T + 1, n), device=device),
+        behavior_logits=torch.randn(k, T + 1, n, A, device=device),
+        episode_return=torch.zeros(k, T + 1, n, device=device),
+        episode_step=torch.zeros(k, T + 1, n, dtype=torch.int32,
+                                 device=device),
+    )
+    if flags.use_lstm:
+        st["initial_core_state"] = torch.zeros(k, 2, n, 256, device=device)
+    return st
+
+
+@pytest.mark.parametrize("use_lstm", [False, True])
+@pytest.mark.timeout(600)
+def test_graphed_step_matches_eager(use_lstm):
+    device = torch.device("cuda", 0)
+    obs, A = (4, 84, 84), 6
+    flags = Flags(env="synthetic-atari", batch_size=8, unroll_length=10,
+                  use_lstm=use_lstm, total_steps=1 << 50)
+    n_steps = 2
+    total_updates = L.GraphedStep.WARMUP_ITERS + n_steps
+
+    def make():
+        torch.manual_seed(42)
+        model = build_model(flags, obs, A).to(device)
+        opt, sched = L.build_optimizer(flags, model)
+        assert isinstance(opt, FlatRMSprop)
+        return model, opt, sched
+
+    # graphed run: first step does WARMUP updates + 1 replay, then replays
+    model_g, opt_g, sched_g = make()
+    st = _staging(flags, obs, A, device)
+    stub = _StubBatcher(flags, st)
+    graphed = L.GraphedStep(flags, model_g, opt_g, stub)
+    for _ in range(n_steps):
+        stats_g = graphed.step(0, sched_g)
+    torch.cuda.synchronize()
+
+    # eager run: same number of parameter updates on the same batch
+    model_e, opt_e, sched_e = make()
+    batch = stub._to_time_major(st)
+    for _ in range(total_updates):
+        stats_e = L.learn_step(flags, model_e, dict(batch), opt_e, sched_e,
+                               None)
+    torch.cuda.synchronize()
+
+    torch.testing.assert_close(opt_g.flat_param, opt_e.flat_param,
+                               rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(opt_g.flat_sqavg, opt_e.flat_sqavg,
+                               rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(stats_g["total_loss"], stats_e["total_loss"],
+                               rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.timeout(600)
+def test_graphed_two_buffers_shared_pool():
+    """Both staging buffers get graphs; replays alternate correctly."""
+    device = torch.device("cuda", 0)
+    obs, A = (4, 84, 84), 6
+    flags = Flags(env="synthetic-atari", batch_size=4, unroll_length=6,
+                  use_lstm=True, total_steps=1 << 50)
+    torch.manual_seed(0)
+    model = build_model(flags, obs, A).to(device)
+    opt, sched = L.build_optimizer(flags, model)
+    st0 = _staging(flags, obs, A, device, seed=1)
+    st1 = _staging(flags, obs, A, device, seed=2)
+    stub = _StubBatcher(flags, st0)
+    stub.staging[1] = st1
+    graphed = L.GraphedStep(flags, model, opt, stub)
+    losses = []
+    for i in (0, 1, 0, 1, 0):
+        stats = graphed.step(i, sched)
+        losses.append(float(stats["total_loss"]))
+    assert all(x == x for x in losses), f"NaN loss in {losses}"
+    assert len(graphed.graphs) == 2

@@ -55,6 +55,7 @@ class Flags:
     # --- device / kernels ---
     device: str = "auto"              # auto | cuda | cpu  ("cuda" is ROCm/HIP)
     use_hip_kernels: bool = True      # escape hatch: False = eager PyTorch ops
+    graph_capture: bool = True        # hipGraph-capture the learn step
     compute_dtype: str = "fp32"       # fp32 | bf16 (encoder GEMM compute)
 
     # --- synthetic env shapes ---
@@ -111,6 +112,7 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--grad-norm-clipping", type=float, default=f.grad_norm_clipping)
     p.add_argument("--device", default=f.device)
     _add_bool(p, "use-hip-kernels", f.use_hip_kernels)
+    _add_bool(p, "graph-capture", f.graph_capture)
     p.add_argument("--compute-dtype", default=f.compute_dtype, choices=["fp32", "bf16"])
     p.add_argument("--synthetic-frame-hw", default=None)
     p.add_argument("--synthetic-num-actions", type=int, default=f.synthetic_num_actions)

@@ -58,10 +58,10 @@ lstm_seq_fwd(const float* __restrict__ Xg, const float* __restrict__ W_hhT,
              float* __restrict__ Cout, float* __restrict__ Hp,
              float* __restrict__ hT, float* __restrict__ cT,
              int save_for_bwd) {
-  __shared__ float s_h[H];
-  __shared__ float s_cm[H];
-  __shared__ float s_c[H];
-  __shared__ float s_g[G4];
+  __shared__ __align__(16) float s_h[H];
+  __shared__ __align__(16) float s_cm[H];
+  __shared__ __align__(16) float s_c[H];
+  __shared__ __align__(16) float s_g[G4];
 
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
@@ -82,17 +82,30 @@ lstm_seq_fwd(const float* __restrict__ Xg, const float* __restrict__ W_hhT,
       if (save_for_bwd) Hp[rowH + tid] = hm;
     }
     __syncthreads();
-    // gate GEMV: thread j computes gate columns (2j, 2j+1)
+    // gate GEMV: thread j computes gate columns (2j, 2j+1).
+    // k is chunked by 4 (one ds_read_b128 broadcast of h) with the four
+    // W row loads batched, so ~16 loads stay in flight per thread —
+    // a scalar k-loop drains vmcnt every iteration (~200 L2 cycles x 256,
+    // measured 15 us/step; this form is latency-pipelined).
     {
       float acc0 = Xg[rowG + 2 * tid];
       float acc1 = Xg[rowG + 2 * tid + 1];
       const float2* wrow = reinterpret_cast<const float2*>(W_hhT) + tid;
-      #pragma unroll 8
-      for (int k = 0; k < H; ++k) {
-        const float hk = s_h[k];
-        const float2 w = wrow[k * (G4 / 2)];
-        acc0 = fmaf(hk, w.x, acc0);
-        acc1 = fmaf(hk, w.y, acc1);
+      #pragma unroll 4
+      for (int k0 = 0; k0 < H; k0 += 4) {
+        const float4 hv = *reinterpret_cast<const float4*>(&s_h[k0]);
+        const float2 w0 = wrow[(k0 + 0) * (G4 / 2)];
+        const float2 w1 = wrow[(k0 + 1) * (G4 / 2)];
+        const float2 w2 = wrow[(k0 + 2) * (G4 / 2)];
+        const float2 w3 = wrow[(k0 + 3) * (G4 / 2)];
+        acc0 = fmaf(hv.x, w0.x, acc0);
+        acc1 = fmaf(hv.x, w0.y, acc1);
+        acc0 = fmaf(hv.y, w1.x, acc0);
+        acc1 = fmaf(hv.y, w1.y, acc1);
+        acc0 = fmaf(hv.z, w2.x, acc0);
+        acc1 = fmaf(hv.z, w2.y, acc1);
+        acc0 = fmaf(hv.w, w3.x, acc0);
+        acc1 = fmaf(hv.w, w3.y, acc1);
       }
       s_g[2 * tid] = acc0;
       s_g[2 * tid + 1] = acc1;
@@ -136,10 +149,10 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
              const float* __restrict__ W_hh,  // (1024, 256) native layout
              int T, int B, float* __restrict__ dG,
              float* __restrict__ dh0, float* __restrict__ dc0) {
-  __shared__ float s_dh[H];
-  __shared__ float s_dc[H];
-  __shared__ float s_dg[G4];
-  __shared__ float s_part[THREADS];
+  __shared__ __align__(16) float s_dh[H];
+  __shared__ __align__(16) float s_dc[H];
+  __shared__ __align__(16) float s_dg[G4];
+  __shared__ __align__(16) float s_part[THREADS];
 
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
@@ -179,14 +192,25 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
       s_dc[tid] = dct * gf * nd;  // dc_{t-1} (masked)
     }
     __syncthreads();
-    // dh_prev[k] = sum_j s_dg[j] * W_hh[j, k]; split j over two thread halves
+    // dh_prev[k] = sum_j s_dg[j] * W_hh[j, k]; split j over two thread
+    // halves; j chunked by 4 with batched W row loads (same latency
+    // pipelining as the forward GEMV).
     {
       const int k = tid & (H - 1);
       const int j0 = (tid < H) ? 0 : (G4 / 2);
+      const float* wcol = W_hh + (size_t)j0 * H + k;
       float acc = 0.f;
-      #pragma unroll 8
-      for (int j = 0; j < G4 / 2; ++j) {
-        acc = fmaf(s_dg[j0 + j], W_hh[(size_t)(j0 + j) * H + k], acc);
+      #pragma unroll 4
+      for (int j = 0; j < G4 / 2; j += 4) {
+        const float4 dg = *reinterpret_cast<const float4*>(&s_dg[j0 + j]);
+        const float w0 = wcol[(size_t)(j + 0) * H];
+        const float w1 = wcol[(size_t)(j + 1) * H];
+        const float w2 = wcol[(size_t)(j + 2) * H];
+        const float w3 = wcol[(size_t)(j + 3) * H];
+        acc = fmaf(dg.x, w0, acc);
+        acc = fmaf(dg.y, w1, acc);
+        acc = fmaf(dg.z, w2, acc);
+        acc = fmaf(dg.w, w3, acc);
       }
       s_part[tid] = acc;
     }

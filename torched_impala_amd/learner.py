@@ -119,9 +119,11 @@ class Batcher:
             slots.append(m)
         return slots
 
-    def next_batch(self) -> Dict[str, torch.Tensor]:
+    def next_batch(self, materialize: bool = True) -> Dict[str, torch.Tensor]:
         """Returns time-major batch: frame (T+1, B, *obs), ...,
-        initial_core_state (2, B, H) if present."""
+        initial_core_state (2, B, H) if present. With materialize=False
+        (graph mode) only drains into the staging buffer and returns its
+        index — the captured graph owns the time-major views."""
         if not self.is_cuda:
             return self._next_batch_cpu()
         self._drain_pending()
@@ -139,7 +141,7 @@ class Batcher:
             copy_done.record(self.copy_stream)
         torch.cuda.current_stream().wait_event(copy_done)
         self._pending.append((copy_done, slots))
-        batch = self._to_time_major(st)
+        batch = self._to_time_major(st) if materialize else {}
         batch["_staging_index"] = i
         return batch
 
@@ -175,10 +177,10 @@ class Batcher:
         self._drain_pending(block=True)
 
 
-def learn_step(flags: Flags, model, batch: Dict[str, torch.Tensor],
-               optimizer, scheduler, reducer: Optional[GradReducer],
-               timers: Optional[StageTimers] = None) -> Dict[str, float]:
-    """One IMPALA update on one batch (SURVEY.md §3c)."""
+def _fwd_bwd(flags: Flags, model, batch: Dict[str, torch.Tensor],
+             optimizer, reducer=None, timers: Optional[StageTimers] = None):
+    """Forward + V-trace/loss + backward: pure device work, every op
+    hipGraph-capturable (GraphedStep captures exactly this)."""
     t = timers or StageTimers()
     t.start("forward")
     core_state = ()
@@ -187,7 +189,7 @@ def learn_step(flags: Flags, model, batch: Dict[str, torch.Tensor],
         core_state = (ics[0], ics[1])
     inputs = dict(frame=batch["frame"], reward=batch["reward"],
                   done=batch["done"], last_action=batch["last_action"])
-    outputs, _ = model(inputs, core_state)
+    outputs, _ = model(inputs, core_state, need_action=False)
 
     logits_all = outputs["policy_logits"]      # (T+1, B, A)
     baseline_all = outputs["baseline"]         # (T+1, B)
@@ -214,6 +216,16 @@ def learn_step(flags: Flags, model, batch: Dict[str, torch.Tensor],
     if reducer:
         reducer.prepare()
     loss.backward()
+    t.stop()
+    return loss, stats
+
+
+def learn_step(flags: Flags, model, batch: Dict[str, torch.Tensor],
+               optimizer, scheduler, reducer: Optional[GradReducer],
+               timers: Optional[StageTimers] = None) -> Dict[str, float]:
+    """One IMPALA update on one batch (SURVEY.md §3c)."""
+    t = timers or StageTimers()
+    loss, stats = _fwd_bwd(flags, model, batch, optimizer, reducer, t)
     t.start("allreduce")
     if reducer:
         reducer.finalize()
@@ -227,6 +239,67 @@ def learn_step(flags: Flags, model, batch: Dict[str, torch.Tensor],
     scheduler.step()
     t.stop()
     return dict(total_loss=loss.detach(), **stats)
+
+
+class GraphedStep:
+    """hipGraph capture of the entire learn step (time-major staging views
+    -> forward -> fused V-trace loss -> backward -> fused clip+RMSProp),
+    one graph per staging buffer, shared memory pool.
+
+    MI355X-native rationale: the headline step is hundreds of small
+    kernels over (T+1)*B=672-row tensors — launch overhead dominates;
+    one hipGraph replay removes it ("HIP streams and graphs instead of a
+    tracing compiler"). lr reaches the captured optimizer through a
+    device scalar (FlatRMSprop.sync_lr runs outside the graph).
+
+    Preconditions (asserted by the caller): CUDA, world_size == 1 (DP
+    capture of RCCL is a later round), FlatRMSprop, HIP kernels on.
+    """
+
+    WARMUP_ITERS = 2
+
+    def __init__(self, flags, model, optimizer, batcher):
+        self.flags = flags
+        self.model = model
+        self.optimizer = optimizer
+        self.batcher = batcher
+        self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self.outputs: Dict[int, tuple] = {}
+        self.pool = None
+
+    def _fn(self, i: int):
+        batch = self.batcher._to_time_major(self.batcher.staging[i])
+        loss, stats = _fwd_bwd(self.flags, self.model, batch, self.optimizer)
+        self.optimizer.step_device()
+        return (loss.detach(), stats["pg_loss"], stats["baseline_loss"],
+                stats["entropy_loss"])
+
+    def _capture(self, i: int):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(self.WARMUP_ITERS):
+                self._fn(i)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        ctx = (torch.cuda.graph(g, pool=self.pool) if self.pool is not None
+               else torch.cuda.graph(g))
+        with ctx:
+            self.outputs[i] = self._fn(i)
+        if self.pool is None:
+            self.pool = g.pool()
+        self.graphs[i] = g
+        log.info("captured learn-step graph for staging buffer %d", i)
+
+    def step(self, i: int, scheduler) -> Dict[str, torch.Tensor]:
+        self.optimizer.sync_lr()
+        if i not in self.graphs:
+            self._capture(i)
+        self.graphs[i].replay()
+        scheduler.step()
+        loss, pg, bl, ent = self.outputs[i]
+        return dict(total_loss=loss, pg_loss=pg, baseline_loss=bl,
+                    entropy_loss=ent)
 
 
 def build_optimizer(flags: Flags, model, world_size: int = 1):
@@ -303,6 +376,10 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     else:
         reducer = None
     batcher = Batcher(flags, buffers, free_queue, full_queue, device)
+    graphed = None
+    if (flags.graph_capture and device.type == "cuda" and world == 1
+            and flags.use_hip_kernels and isinstance(optimizer, FlatRMSprop)):
+        graphed = GraphedStep(flags, model, optimizer, batcher)
     metrics = MetricsLogger(flags.metrics_jsonl)
     timers = StageTimers()
 
@@ -326,9 +403,23 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
             if max_learner_steps is not None and learner_steps >= max_learner_steps:
                 break
             timers.start("rollout_wait")
-            batch = batcher.next_batch()
-            stats = learn_step(flags, model, batch, optimizer, scheduler,
-                               reducer, timers)
+            batch = batcher.next_batch(materialize=graphed is None)
+            si = batch.get("_staging_index")
+            if graphed is not None:
+                timers.start("graph_step")
+                try:
+                    stats = graphed.step(si, scheduler)
+                except Exception:
+                    log.exception("graph capture failed; falling back to "
+                                  "eager learn steps")
+                    graphed = None
+                    batch.update(batcher._to_time_major(batcher.staging[si]))
+                    stats = learn_step(flags, model, batch, optimizer,
+                                       scheduler, reducer, timers)
+                timers.stop()
+            else:
+                stats = learn_step(flags, model, batch, optimizer, scheduler,
+                                   reducer, timers)
             batcher.mark_consumed(batch)
             learner_steps += 1
             step_frames += frames_per_batch * world
@@ -340,9 +431,14 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
 
             # episode metrics (cheap on CPU; sync-inducing on GPU -> sparse)
             if device.type != "cuda" or learner_steps % 20 == 0:
-                done = batch["done"][1:]
-                if done.any():
+                if graphed is not None and si is not None:
+                    st = batcher.staging[si]
+                    done = st["done"][:, 1:]
+                    rets = st["episode_return"][:, 1:][done]
+                else:
+                    done = batch["done"][1:]
                     rets = batch["episode_return"][1:][done]
+                if rets.numel():
                     metrics.record_episodes(rets.cpu().tolist())
 
             # respawn dead actors (SURVEY.md §5c failure detection)

@@ -86,19 +86,23 @@ class _HeadsMixin(nn.Module):
         self.baseline = nn.Linear(in_features, 1)
         self.num_actions = num_actions
 
-    def _heads(self, core_output: torch.Tensor, T: int, B: int, greedy: bool):
+    def _heads(self, core_output: torch.Tensor, T: int, B: int, greedy: bool,
+               need_action: bool = True):
         policy_logits = self.policy(core_output)
         baseline = self.baseline(core_output)
-        if greedy:
-            action = torch.argmax(policy_logits, dim=-1)
-        else:
-            action = torch.multinomial(
-                F.softmax(policy_logits.float(), dim=-1), num_samples=1).squeeze(-1)
-        return dict(
+        out = dict(
             policy_logits=policy_logits.view(T, B, self.num_actions).float(),
             baseline=baseline.view(T, B).float(),
-            action=action.view(T, B),
         )
+        if need_action:  # the learner skips sampling (keeps RNG out of graphs)
+            if greedy:
+                action = torch.argmax(policy_logits, dim=-1)
+            else:
+                action = torch.multinomial(
+                    F.softmax(policy_logits.float(), dim=-1),
+                    num_samples=1).squeeze(-1)
+            out["action"] = action.view(T, B)
+        return out
 
 
 def _core_extras(inputs, num_actions: int, dtype: torch.dtype):
@@ -123,7 +127,8 @@ class MLPPolicy(_CoreMixin, _HeadsMixin):
         self._init_core(core_in, use_lstm)
         self._init_heads(self.core_output_size, num_actions)
 
-    def forward(self, inputs, core_state=(), greedy: bool = False):
+    def forward(self, inputs, core_state=(), greedy: bool = False,
+                need_action: bool = True):
         frame = inputs["frame"]
         T, B = frame.shape[:2]
         x = _flatten_time(frame).float()
@@ -133,7 +138,7 @@ class MLPPolicy(_CoreMixin, _HeadsMixin):
         core_input = torch.cat(
             [x.view(T, B, -1), reward, last_action], dim=-1)
         core_output, core_state = self._run_core(core_input, inputs["done"], core_state)
-        return self._heads(core_output, T, B, greedy), core_state
+        return self._heads(core_output, T, B, greedy, need_action), core_state
 
 
 class ShallowNet(_CoreMixin, _HeadsMixin):
@@ -169,14 +174,15 @@ class ShallowNet(_CoreMixin, _HeadsMixin):
         x = F.relu(self.conv3(x))
         return F.relu(self.fc(x.flatten(1)))
 
-    def forward(self, inputs, core_state=(), greedy: bool = False):
+    def forward(self, inputs, core_state=(), greedy: bool = False,
+                need_action: bool = True):
         frame = inputs["frame"]
         T, B = frame.shape[:2]
         x = self.encode(_flatten_time(frame))
         reward, last_action = _core_extras(inputs, self.num_actions, x.dtype)
         core_input = torch.cat([x.view(T, B, -1), reward, last_action], dim=-1)
         core_output, core_state = self._run_core(core_input, inputs["done"], core_state)
-        return self._heads(core_output, T, B, greedy), core_state
+        return self._heads(core_output, T, B, greedy, need_action), core_state
 
 
 class _ResBlock(nn.Module):
@@ -241,14 +247,15 @@ class DeepResNet(_CoreMixin, _HeadsMixin):
         x = F.relu(x)
         return F.relu(self.fc(x.flatten(1)))
 
-    def forward(self, inputs, core_state=(), greedy: bool = False):
+    def forward(self, inputs, core_state=(), greedy: bool = False,
+                need_action: bool = True):
         frame = inputs["frame"]
         T, B = frame.shape[:2]
         x = self.encode(_flatten_time(frame))
         reward, last_action = _core_extras(inputs, self.num_actions, x.dtype)
         core_input = torch.cat([x.view(T, B, -1), reward, last_action], dim=-1)
         core_output, core_state = self._run_core(core_input, inputs["done"], core_state)
-        return self._heads(core_output, T, B, greedy), core_state
+        return self._heads(core_output, T, B, greedy, need_action), core_state
 
 
 def build_model(flags, obs_shape, num_actions):

@@ -210,3 +210,49 @@ def test_lstm_model_fast_vs_eager_core():
                                rtol=1e-4, atol=1e-4)
     for a, b in zip(st_fast, st_eager):
         torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# Fused shallow-encoder conv kernels: parity vs eager fp32 (fwd + grads)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("N", [3, 64])
+def test_shallow_encoder_parity(N):
+    from torched_impala_amd.models import ShallowNet
+    torch.manual_seed(0)
+    dev = _dev()
+    net = ShallowNet((4, 84, 84), 6).to(dev)
+    x = torch.randint(0, 256, (N, 4, 84, 84), dtype=torch.uint8, device=dev)
+
+    out_fast = net.encode(x)
+    w = torch.arange(out_fast.numel(), device=dev).view_as(out_fast)
+    loss = (out_fast * torch.sin(w.float())).sum()
+    loss.backward()
+    g_fast = {k: p.grad.clone() for k, p in net.named_parameters()
+              if p.grad is not None}
+    net.zero_grad()
+
+    net.force_eager_core = True
+    out_eager = net.encode(x)
+    loss_e = (out_eager * torch.sin(w.float())).sum()
+    loss_e.backward()
+
+    torch.testing.assert_close(out_fast, out_eager, rtol=2e-4, atol=2e-4)
+    for k, p in net.named_parameters():
+        if p.grad is None:
+            continue
+        torch.testing.assert_close(
+            g_fast[k], p.grad, rtol=2e-3, atol=2e-3,
+            msg=lambda m: f"grad mismatch for {k}: {m}")
+
+
+def test_shallow_encoder_inference_no_saves():
+    """Inference path (no grad) must not allocate the saved activations."""
+    from torched_impala_amd.models import ShallowNet
+    dev = _dev()
+    net = ShallowNet((4, 84, 84), 6).to(dev)
+    x = torch.randint(0, 256, (5, 4, 84, 84), dtype=torch.uint8, device=dev)
+    with torch.no_grad():
+        out = net.encode(x)
+    assert out.shape == (5, 256)
+    assert torch.isfinite(out).all()

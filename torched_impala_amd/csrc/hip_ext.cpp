@@ -42,6 +42,15 @@ void launch_grad_sqnorm(const float*, long, const float*, float*,
 void launch_rmsprop_update(float*, const float*, float*, long, const float*,
                            const float*, const float*, float, float, float,
                            hipStream_t);
+void launch_shallow_encoder_fwd(const unsigned char*, const float*,
+                                const float*, const float*, const float*,
+                                const float*, const float*, int, float*,
+                                float*, float*, int, hipStream_t);
+void launch_shallow_encoder_bwd(const unsigned char*, const float*,
+                                const float*, const float*, const float*,
+                                const float*, const float*, int, float*,
+                                hipStream_t);
+int shallow_encoder_slab_elems();
 }
 
 namespace {
@@ -285,6 +294,61 @@ void fused_clip_rmsprop(torch::Tensor flat_p, torch::Tensor flat_g,
   HIP_CHECK(hipGetLastError());
 }
 
+// ---------------------------------------------------------------------------
+// fused shallow-encoder convolutions (one workgroup per 84x84x4 image;
+// conv1->conv2->conv3 LDS-resident; see conv_kernels.hip).
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> shallow_encoder_fwd(
+    torch::Tensor frames, torch::Tensor w1, torch::Tensor b1,
+    torch::Tensor w2, torch::Tensor b2, torch::Tensor w3, torch::Tensor b3,
+    bool save_for_bwd) {
+  TORCH_CHECK(frames.scalar_type() == torch::kUInt8 && frames.is_cuda() &&
+                  frames.is_contiguous(), "frames must be cuda u8 contiguous");
+  TORCH_CHECK(frames.size(1) == 4 && frames.size(2) == 84 &&
+                  frames.size(3) == 84, "specialized for 4x84x84");
+  check_f32(w1, "w1");
+  check_f32(w2, "w2");
+  check_f32(w3, "w3");
+  const int N = frames.size(0);
+  DeviceScope guard(frames.get_device());
+  auto opts = w1.options();
+  torch::Tensor a1, a2;
+  if (save_for_bwd) {
+    a1 = torch::empty({N, 16, 20, 20}, opts);
+    a2 = torch::empty({N, 32, 9, 9}, opts);
+  } else {
+    a1 = torch::empty({1}, opts);
+    a2 = torch::empty({1}, opts);
+  }
+  auto a3 = torch::empty({N, 1568}, opts);
+  launch_shallow_encoder_fwd(
+      frames.data_ptr<unsigned char>(), w1.data_ptr<float>(),
+      b1.data_ptr<float>(), w2.data_ptr<float>(), b2.data_ptr<float>(),
+      w3.data_ptr<float>(), b3.data_ptr<float>(), N, a1.data_ptr<float>(),
+      a2.data_ptr<float>(), a3.data_ptr<float>(), save_for_bwd ? 1 : 0,
+      current_stream());
+  HIP_CHECK(hipGetLastError());
+  return {a1, a2, a3};
+}
+
+torch::Tensor shallow_encoder_bwd(
+    torch::Tensor frames, torch::Tensor a1, torch::Tensor a2,
+    torch::Tensor a3, torch::Tensor d_a3, torch::Tensor w2,
+    torch::Tensor w3) {
+  const int N = frames.size(0);
+  check_f32(d_a3, "d_a3");
+  DeviceScope guard(frames.get_device());
+  auto slab = torch::empty({N, (long)shallow_encoder_slab_elems()},
+                           d_a3.options());
+  launch_shallow_encoder_bwd(
+      frames.data_ptr<unsigned char>(), a1.data_ptr<float>(),
+      a2.data_ptr<float>(), a3.data_ptr<float>(), d_a3.data_ptr<float>(),
+      w2.data_ptr<float>(), w3.data_ptr<float>(), N,
+      slab.data_ptr<float>(), current_stream());
+  HIP_CHECK(hipGetLastError());
+  return slab;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -297,4 +361,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_fwd", &lstm_seq_fwd);
   m.def("lstm_seq_bwd", &lstm_seq_bwd);
   m.def("fused_clip_rmsprop", &fused_clip_rmsprop);
+  m.def("shallow_encoder_fwd", &shallow_encoder_fwd);
+  m.def("shallow_encoder_bwd", &shallow_encoder_bwd);
 }

@@ -168,6 +168,13 @@ class ShallowNet(_CoreMixin, _HeadsMixin):
 
     def encode(self, x: torch.Tensor) -> torch.Tensor:
         """(N, C, H, W) uint8/float -> (N, 256). The HIP-kernel hot path."""
+        if (x.is_cuda and x.dtype == torch.uint8
+                and self.obs_shape == (4, 84, 84)
+                and not getattr(self, "force_eager_core", False)):
+            from torched_impala_amd import ops
+            if ops.hip_available():
+                from torched_impala_amd.ops.encoder import shallow_encode_conv
+                return F.relu(self.fc(shallow_encode_conv(self, x)))
         x = x.float() / 255.0
         x = F.relu(self.conv1(x))
         x = F.relu(self.conv2(x))

@@ -52,7 +52,7 @@ __device__ __forceinline__ float relu(float x) { return x > 0.f ? x : 0.f; }
 // ---------------------------------------------------------------------------
 // forward: grid = (N) blocks x 256 threads
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS)
+__global__ void __launch_bounds__(NTHREADS, 2)
 shallow_encoder_fwd(const unsigned char* __restrict__ img,
                     const float* __restrict__ w1, const float* __restrict__ b1,
                     const float* __restrict__ w2, const float* __restrict__ b2,
@@ -83,14 +83,14 @@ shallow_encoder_fwd(const unsigned char* __restrict__ img,
     const int lane = tid & 15;
     const float* wf = w1 + f * (C0 * K1 * K1);
     const float bias = b1[f];
+    #pragma unroll 1
     for (int px = lane; px < H1 * W1; px += 16) {
       const int oy = px / W1, ox = px - oy * W1;
       float acc = 0.f;
-      #pragma unroll
       for (int c = 0; c < C0; ++c) {
         const unsigned char* ic = s_img + c * (H0 * W0);
         const float* wc = wf + c * (K1 * K1);
-        #pragma unroll
+        #pragma unroll 2
         for (int ky = 0; ky < K1; ++ky) {
           // 8 u8 pixels = 2 aligned dwords (4*ox is dword-aligned)
           const unsigned char* row = ic + (S1 * oy + ky) * W0 + S1 * ox;
@@ -123,6 +123,7 @@ shallow_encoder_fwd(const unsigned char* __restrict__ img,
     const int lane = tid & 7;
     const float* wf = w2 + f * (F1 * K2 * K2);
     const float bias = b2[f];
+    #pragma unroll 1
     for (int px = lane; px < H2 * W2; px += 8) {
       const int oy = px / W2, ox = px - oy * W2;
       float acc = bias;
@@ -153,6 +154,7 @@ shallow_encoder_fwd(const unsigned char* __restrict__ img,
     const int lane = tid & 7;
     const float* wf = w3 + f * (F2 * K3 * K3);
     const float bias = b3[f];
+    #pragma unroll 1
     for (int px = lane; px < H3 * W3; px += 8) {
       const int oy = px / W3, ox = px - oy * W3;
       float acc = bias;
@@ -179,8 +181,14 @@ shallow_encoder_fwd(const unsigned char* __restrict__ img,
 // In: d_a3 (N,1568) = d(flat conv3 out), saved a1/a2/a3, frame, weights.
 // Out: per-image slab [dW1|dW2|dW3|db1|db2|db3] (N, SLAB_N); reduced
 // over N outside. No d(frame).
+//
+// LDS is a phased arena (78.4 KiB -> 2 workgroups/CU): the u8 frame is
+// loaded LAST, over the then-dead a2/d3/d2 regions, after db3/db2 are
+// banked — a static per-buffer layout needs 106 KiB and halves
+// occupancy.
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS)
+constexpr int ARENA_F = (A2_N + A3_N + A2_N + A1_N + A1_N);  // floats
+__global__ void __launch_bounds__(NTHREADS, 2)
 shallow_encoder_bwd(const unsigned char* __restrict__ img,
                     const float* __restrict__ a1g,
                     const float* __restrict__ a2g,
@@ -188,18 +196,22 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
                     const float* __restrict__ d_a3g,
                     const float* __restrict__ w2, const float* __restrict__ w3,
                     int N, float* __restrict__ slab) {
-  __shared__ __align__(16) float s_d3[A3_N];    // masked d(conv3 out)
-  __shared__ __align__(16) float s_a2[A2_N];
-  __shared__ __align__(16) float s_d2[A2_N];    // masked d(conv2 out)
-  __shared__ __align__(16) float s_a1[A1_N];
-  __shared__ __align__(16) float s_d1[A1_N];    // masked d(conv1 out)
-  __shared__ __align__(16) unsigned char s_img[IMG_BYTES];
+  __shared__ __align__(16) float s_arena[ARENA_F];
+  float* s_a2 = s_arena;                 // A2_N
+  float* s_d3 = s_a2 + A2_N;             // A3_N
+  float* s_d2 = s_d3 + A3_N;             // A2_N
+  float* s_a1 = s_d2 + A2_N;             // A1_N
+  float* s_d1 = s_a1 + A1_N;             // A1_N
+  // phase-5 overlay: the u8 frame covers [s_a2 .. s_a2+28224B) — the
+  // dead a2/d3/d2 (+ start of a1) regions
+  unsigned char* s_img = reinterpret_cast<unsigned char*>(s_arena);
 
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   float* my_slab = slab + (size_t)n * SLAB_N;
+  const int DB_OFF = DW1_N + DW2_N + DW3_N;
 
-  // stage a2, d3 (relu-masked), a1, img
+  // stage a2, d3 (relu-masked), a1
   for (int i = tid; i < A2_N; i += NTHREADS)
     s_a2[i] = a2g[(size_t)n * A2_N + i];
   for (int i = tid; i < A3_N; i += NTHREADS) {
@@ -208,15 +220,10 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
   }
   for (int i = tid; i < A1_N; i += NTHREADS)
     s_a1[i] = a1g[(size_t)n * A1_N + i];
-  {
-    const unsigned int* src =
-        reinterpret_cast<const unsigned int*>(img + (size_t)n * IMG_BYTES);
-    unsigned int* dst = reinterpret_cast<unsigned int*>(s_img);
-    for (int i = tid; i < IMG_BYTES / 4; i += NTHREADS) dst[i] = src[i];
-  }
   __syncthreads();
 
   // ---- dW3[f][c][ky][kx] = sum_px s_a2[c][oy+ky][ox+kx] * s_d3[f][px] ----
+  #pragma unroll 1
   for (int e = tid; e < DW3_N; e += NTHREADS) {
     const int f = e / (F2 * K3 * K3);
     const int r = e - f * (F2 * K3 * K3);
@@ -226,7 +233,6 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     const float* dc = s_d3 + f * (H3 * W3);
     const float* ac = s_a2 + c * (H2 * W2);
     float acc = 0.f;
-    #pragma unroll
     for (int oy = 0; oy < H3; ++oy) {
       const float* arow = ac + (oy + ky) * W2 + kx;
       const float* drow = dc + oy * W3;
@@ -236,8 +242,8 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     my_slab[DW1_N + DW2_N + e] = acc;
   }
 
-  // ---- d_a2[c][y][x] = sum_f sum_{valid ky,kx} d3[f][y-ky][x-kx] w3[f][c][ky][kx]
-  //      masked by a2>0 ----
+  // ---- d_a2 (masked by a2>0) ----
+  #pragma unroll 1
   for (int e = tid; e < A2_N; e += NTHREADS) {
     const int c = e / (H2 * W2);
     const int p = e - c * (H2 * W2);
@@ -246,11 +252,9 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     for (int f = 0; f < F3; ++f) {
       const float* dc = s_d3 + f * (H3 * W3);
       const float* wc = w3 + (f * F2 + c) * (K3 * K3);
-      #pragma unroll
       for (int ky = 0; ky < K3; ++ky) {
         const int oy = y - ky;
         if (oy < 0 || oy >= H3) continue;
-        #pragma unroll
         for (int kx = 0; kx < K3; ++kx) {
           const int ox = x - kx;
           if (ox < 0 || ox >= W3) continue;
@@ -262,7 +266,8 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
   }
   __syncthreads();
 
-  // ---- dW2[f][c][ky][kx] = sum_px s_a1[c][2oy+ky][2ox+kx] * s_d2[f][px] ----
+  // ---- dW2 ----
+  #pragma unroll 1
   for (int e = tid; e < DW2_N; e += NTHREADS) {
     const int f = e / (F1 * K2 * K2);
     const int r = e - f * (F1 * K2 * K2);
@@ -272,7 +277,6 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     const float* dc = s_d2 + f * (H2 * W2);
     const float* ac = s_a1 + c * (H1 * W1);
     float acc = 0.f;
-    #pragma unroll
     for (int oy = 0; oy < H2; ++oy) {
       const float* arow = ac + (S2 * oy + ky) * W1 + kx;
       const float* drow = dc + oy * W2;
@@ -283,8 +287,8 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     my_slab[DW1_N + e] = acc;
   }
 
-  // ---- d_a1[c][y][x] = sum_f sum_{2oy+ky=y, 2ox+kx=x} d2[f][oy][ox] w2[f][c][ky][kx]
-  //      masked by a1>0 ----
+  // ---- d_a1 (masked by a1>0) ----
+  #pragma unroll 1
   for (int e = tid; e < A1_N; e += NTHREADS) {
     const int c = e / (H1 * W1);
     const int p = e - c * (H1 * W1);
@@ -293,13 +297,11 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     for (int f = 0; f < F2; ++f) {
       const float* dc = s_d2 + f * (H2 * W2);
       const float* wc = w2 + (f * F1 + c) * (K2 * K2);
-      #pragma unroll
       for (int ky = 0; ky < K2; ++ky) {
         const int ty = y - ky;
         if (ty < 0 || (ty & 1)) continue;
         const int oy = ty >> 1;
         if (oy >= H2) continue;
-        #pragma unroll
         for (int kx = 0; kx < K2; ++kx) {
           const int tx = x - kx;
           if (tx < 0 || (tx & 1)) continue;
@@ -313,9 +315,40 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
   }
   __syncthreads();
 
-  // ---- dW1[f][c][ky][kx] = sum_px (img[c][4oy+ky][4ox+kx]/255) * d1[f][px] ----
+  // ---- bank db3/db2/db1 BEFORE the frame overlays their buffers ----
+  for (int f = tid; f < F3; f += NTHREADS) {
+    float acc = 0.f;
+    const float* dc = s_d3 + f * (H3 * W3);
+    for (int p = 0; p < H3 * W3; ++p) acc += dc[p];
+    my_slab[DB_OFF + F1 + F2 + f] = acc;
+  }
+  for (int f = tid; f < F2; f += NTHREADS) {
+    float acc = 0.f;
+    const float* dc = s_d2 + f * (H2 * W2);
+    for (int p = 0; p < H2 * W2; ++p) acc += dc[p];
+    my_slab[DB_OFF + F1 + f] = acc;
+  }
+  for (int f = tid; f < F1; f += NTHREADS) {
+    float acc = 0.f;
+    const float* dc = s_d1 + f * (H1 * W1);
+    for (int p = 0; p < H1 * W1; ++p) acc += dc[p];
+    my_slab[DB_OFF + f] = acc;
+  }
+  __syncthreads();
+
+  // ---- overlay: stage the u8 frame over the dead a2/d3/d2 regions ----
+  {
+    const unsigned int* src =
+        reinterpret_cast<const unsigned int*>(img + (size_t)n * IMG_BYTES);
+    unsigned int* dst = reinterpret_cast<unsigned int*>(s_img);
+    for (int i = tid; i < IMG_BYTES / 4; i += NTHREADS) dst[i] = src[i];
+  }
+  __syncthreads();
+
+  // ---- dW1 from the u8 frame (fused /255) ----
   {
     const float inv255 = 1.0f / 255.0f;
+    #pragma unroll 1
     for (int e = tid; e < DW1_N; e += NTHREADS) {
       const int f = e / (C0 * K1 * K1);
       const int r = e - f * (C0 * K1 * K1);
@@ -325,36 +358,15 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
       const float* dc = s_d1 + f * (H1 * W1);
       const unsigned char* ic = s_img + c * (H0 * W0);
       float acc = 0.f;
-      #pragma unroll
       for (int oy = 0; oy < H1; ++oy) {
         const unsigned char* irow = ic + (S1 * oy + ky) * W0 + kx;
         const float* drow = dc + oy * W1;
-        #pragma unroll
+        #pragma unroll 4
         for (int ox = 0; ox < W1; ++ox)
           acc = fmaf((float)irow[S1 * ox], drow[ox], acc);
       }
       my_slab[e] = acc * inv255;
     }
-  }
-
-  // ---- bias grads: db = sum_px d(pre-act) over each filter map ----
-  for (int f = tid; f < F1; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d1 + f * (H1 * W1);
-    for (int p = 0; p < H1 * W1; ++p) acc += dc[p];
-    my_slab[DW1_N + DW2_N + DW3_N + f] = acc;
-  }
-  for (int f = tid; f < F2; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d2 + f * (H2 * W2);
-    for (int p = 0; p < H2 * W2; ++p) acc += dc[p];
-    my_slab[DW1_N + DW2_N + DW3_N + F1 + f] = acc;
-  }
-  for (int f = tid; f < F3; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d3 + f * (H3 * W3);
-    for (int p = 0; p < H3 * W3; ++p) acc += dc[p];
-    my_slab[DW1_N + DW2_N + DW3_N + F1 + F2 + f] = acc;
   }
 }
 

@@ -385,13 +385,20 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
 
     run_dir = None
     ckpt_path = None
+    resume_frames = 0
     if flags.xpid and rank == 0:
         run_dir = os.path.join(os.path.expanduser(flags.savedir), flags.xpid)
         os.makedirs(run_dir, exist_ok=True)
         ckpt_path = os.path.join(run_dir, "model.tar")
+        if os.path.exists(ckpt_path) and not flags.disable_checkpoint:
+            ckpt = ckpt_mod.load(ckpt_path, model, optimizer, scheduler,
+                                 map_location=device)
+            resume_frames = int(ckpt.get("step", 0))
+            log.info("resumed from %s at step=%d", ckpt_path, resume_frames)
+            mirror.publish_state_dict(model.state_dict())
 
     frames_per_batch = flags.unroll_length * flags.batch_size
-    step_frames = 0
+    step_frames = resume_frames
     learner_steps = 0
     last_ckpt = time.time()
     t_start = time.time()

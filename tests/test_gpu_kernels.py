@@ -256,3 +256,43 @@ def test_shallow_encoder_inference_no_saves():
         out = net.encode(x)
     assert out.shape == (5, 256)
     assert torch.isfinite(out).all()
+
+
+# ---------------------------------------------------------------------------
+# MFMA linear kernel parity
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("M,N,K", [(672, 256, 1568), (64, 64, 64),
+                                   (100, 7, 256), (33, 256, 263)])
+def test_linear_mfma_parity(M, N, K):
+    from torched_impala_amd.ops.linear import _LinearReLU
+    torch.manual_seed(0)
+    dev = _dev()
+    x = torch.randn(M, K, device=dev, requires_grad=True)
+    lin = torch.nn.Linear(K, N).to(dev)
+
+    y = _LinearReLU.apply(x, lin.weight, lin.bias)
+    s = torch.sin(torch.arange(y.numel(), device=dev).float()).view_as(y)
+    (y * s).sum().backward()
+    gx, gw, gb = x.grad.clone(), lin.weight.grad.clone(), lin.bias.grad.clone()
+    x.grad = None
+    lin.zero_grad()
+
+    x2 = x.detach().clone().requires_grad_()
+    y2 = torch.relu(torch.nn.functional.linear(x2, lin.weight, lin.bias))
+    (y2 * s).sum().backward()
+
+    torch.testing.assert_close(y, y2, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(gx, x2.grad, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(gw, lin.weight.grad, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(gb, lin.bias.grad, rtol=1e-3, atol=1e-4)
+
+
+def test_linear_mfma_plain_mode():
+    dev = _dev()
+    torch.manual_seed(1)
+    A = torch.randn(70, 100, device=dev)
+    B = torch.randn(50, 100, device=dev)
+    from torched_impala_amd import ops
+    C = ops.ext_hip().linear_mfma(A, B, None, None, 0)
+    torch.testing.assert_close(C, A @ B.t(), rtol=1e-4, atol=1e-4)

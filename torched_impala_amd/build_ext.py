@@ -64,7 +64,7 @@ def build_ring(force: bool = False) -> str:
 
 
 HIP_SOURCES = ["hip_ext.cpp", "vtrace_kernels.hip", "lstm_kernels.hip",
-               "optim_kernels.hip", "conv_kernels.hip"]
+               "optim_kernels.hip", "conv_kernels.hip", "linear_kernels.hip"]
 
 
 def build_hip(force: bool = False) -> str:

@@ -51,6 +51,9 @@ void launch_shallow_encoder_bwd(const unsigned char*, const float*,
                                 const float*, const float*, int, float*,
                                 hipStream_t);
 int shallow_encoder_slab_elems();
+void launch_linear_mfma(const float*, const float*, const float*,
+                        const float*, int, int, int, int, float*,
+                        hipStream_t);
 }
 
 namespace {
@@ -349,6 +352,38 @@ torch::Tensor shallow_encoder_bwd(
   return slab;
 }
 
+// ---------------------------------------------------------------------------
+// MFMA linear: C = act(A @ B^T + bias). mode 0 plain / 1 relu /
+// 2 A-masked-by-(ymask>0) (fused ReLU backward for dX).
+// ---------------------------------------------------------------------------
+torch::Tensor linear_mfma(torch::Tensor A, torch::Tensor B,
+                          c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> ymask, int64_t mode) {
+  check_f32(A, "A");
+  check_f32(B, "B");
+  const int M = A.size(0);
+  const int K = A.size(1);
+  const int N = B.size(0);
+  TORCH_CHECK(B.size(1) == K, "K mismatch");
+  DeviceScope guard(A.get_device());
+  auto C = torch::empty({M, N}, A.options());
+  const float* bias_p = nullptr;
+  const float* ymask_p = nullptr;
+  if (bias.has_value()) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  if (ymask.has_value()) {
+    check_f32(*ymask, "ymask");
+    ymask_p = ymask->data_ptr<float>();
+  }
+  launch_linear_mfma(A.data_ptr<float>(), B.data_ptr<float>(), bias_p,
+                     ymask_p, M, N, K, (int)mode, C.data_ptr<float>(),
+                     current_stream());
+  HIP_CHECK(hipGetLastError());
+  return C;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -363,4 +398,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_clip_rmsprop", &fused_clip_rmsprop);
   m.def("shallow_encoder_fwd", &shallow_encoder_fwd);
   m.def("shallow_encoder_bwd", &shallow_encoder_bwd);
+  m.def("linear_mfma", &linear_mfma, py::arg("A"), py::arg("B"),
+        py::arg("bias") = py::none(), py::arg("ymask") = py::none(),
+        py::arg("mode") = 0);
 }

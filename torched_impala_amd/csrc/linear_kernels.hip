@@ -1,0 +1,163 @@
+// gfx950 MFMA linear kernels (BASELINE.json:5 "MFMA for the conv/linear
+// GEMMs"; SURVEY.md §2c linear_fwd/bwd).
+//
+// f32-input MFMA (v_mfma_f32_16x16x4_f32 — exact f32 at the 157 TF f32
+// vector rate, cdna_hip_programming.md §3): the learner is fp32, so the
+// f32 "SGEMM"-class instructions are the right matrix-core path (no
+// xf32 on gfx950).
+//
+// One kernel shape serves both directions of the FC layer:
+//     C[i][j] = act( sum_t A[i][t] * B[j][t] + bias[j] )
+//   fwd:  A = X (M,K),  B = W (N,K),   act = ReLU, bias = b
+//   bwdX: A = dY masked by (Y>0) ON THE FLY during staging (fuses the
+//         ReLU backward), B = W^T (K,N) pre-transposed once, no bias.
+// dW/db are plain reductions over the big M dim — they stay on rocBLAS
+// (which consumes the transposed layouts without materialising them).
+//
+// Tiling: 64x64 block tile, 4 waves of 32x32 (2x2 16x16 fragments),
+// K staged in LDS 32 deep, double-buffered, +4-float row pad against
+// bank conflicts. Tail M/N guarded.
+
+#include <hip/hip_runtime.h>
+
+namespace impala {
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+constexpr int BM = 64, BN = 64, BK = 32;
+constexpr int LDS_STRIDE = BK + 4;  // pad: avoids b32 column conflicts
+constexpr int THREADS = 256;
+
+// mode: 0 = plain, 1 = relu epilogue, 2 = A-staging masked by Ymask>0
+__global__ void __launch_bounds__(THREADS, 2)
+linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
+            const float* __restrict__ bias, const float* __restrict__ Ymask,
+            int M, int N, int K, int mode, float* __restrict__ C) {
+  __shared__ __align__(16) float s_a[2][BM * LDS_STRIDE];
+  __shared__ __align__(16) float s_b[2][BN * LDS_STRIDE];
+
+  const int bm = blockIdx.x * BM;
+  const int bn = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;           // 4 waves
+  const int wm = (wave >> 1) * 32;     // wave row offset in tile
+  const int wn = (wave & 1) * 32;      // wave col offset in tile
+
+  f32x4 acc[2][2] = {};
+
+  // staging map: 256 threads x 2 rows each; thread handles row r = tid/4,
+  // 8 floats at k-offset (tid%4)*8 -> two float4 loads per row-half
+  const int srow = tid >> 2;           // 0..63
+  const int skoff = (tid & 3) * 8;     // 0,8,16,24
+
+  auto stage = [&](int buf, int k0) {
+    // A rows
+    {
+      const int gr = bm + srow;
+      float4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
+      if (gr < M) {
+        const float* src = A + (size_t)gr * K + k0 + skoff;
+        const int krem = K - (k0 + skoff);
+        if (krem >= 8) {
+          v0 = *reinterpret_cast<const float4*>(src);
+          v1 = *reinterpret_cast<const float4*>(src + 4);
+        } else {
+          for (int i = 0; i < 8 && i < krem; ++i)
+            (i < 4 ? (&v0.x)[i] : (&v1.x)[i - 4]) = src[i];
+        }
+        if (mode == 2) {  // fused ReLU-backward mask from Y
+          const float* ym = Ymask + (size_t)gr * K + k0 + skoff;
+          const int mrem = K - (k0 + skoff);
+          for (int i = 0; i < 4; ++i) {
+            if (i < mrem && ym[i] <= 0.f) (&v0.x)[i] = 0.f;
+            if (i + 4 < mrem && ym[i + 4] <= 0.f) (&v1.x)[i] = 0.f;
+          }
+        }
+      }
+      float* dst = s_a[buf] + srow * LDS_STRIDE + skoff;
+      *reinterpret_cast<float4*>(dst) = v0;
+      *reinterpret_cast<float4*>(dst + 4) = v1;
+    }
+    // B rows
+    {
+      const int gr = bn + srow;
+      float4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
+      if (gr < N) {
+        const float* src = B + (size_t)gr * K + k0 + skoff;
+        const int krem = K - (k0 + skoff);
+        if (krem >= 8) {
+          v0 = *reinterpret_cast<const float4*>(src);
+          v1 = *reinterpret_cast<const float4*>(src + 4);
+        } else {
+          for (int i = 0; i < 8 && i < krem; ++i)
+            (i < 4 ? (&v0.x)[i] : (&v1.x)[i - 4]) = src[i];
+        }
+      }
+      float* dst = s_b[buf] + srow * LDS_STRIDE + skoff;
+      *reinterpret_cast<float4*>(dst) = v0;
+      *reinterpret_cast<float4*>(dst + 4) = v1;
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  const int nk = (K + BK - 1) / BK;
+  for (int kt = 0; kt < nk; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nk) stage(cur ^ 1, (kt + 1) * BK);
+    // MFMA over this K tile: 16x16x4 fragments
+    const int ar = lane & 15;          // fragment row
+    const int ak = lane >> 4;          // k sub-index (0..3)
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      float a0 = s_a[cur][(wm + ar) * LDS_STRIDE + kk + ak];
+      float a1 = s_a[cur][(wm + 16 + ar) * LDS_STRIDE + kk + ak];
+      float b0 = s_b[cur][(wn + ar) * LDS_STRIDE + kk + ak];
+      float b1 = s_b[cur][(wn + 16 + ar) * LDS_STRIDE + kk + ak];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C[row][col] with 16x16 C-layout col=lane&15,
+  // row=(lane>>4)*4+reg (dtype-independent on gfx950)
+  const int ccol = lane & 15;
+  const int crow = (lane >> 4) * 4;
+  #pragma unroll
+  for (int r = 0; r < 2; ++r) {
+    #pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = bm + wm + r * 16 + crow + reg;
+        const int col = bn + wn + c * 16 + ccol;
+        if (row < M && col < N) {
+          float v = acc[r][c][reg];
+          if (bias != nullptr) v += bias[col];
+          if (mode == 1) v = v > 0.f ? v : 0.f;
+          C[(size_t)row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
+}  // namespace impala
+
+extern "C" {
+
+void launch_linear_mfma(const float* A, const float* B, const float* bias,
+                        const float* ymask, int M, int N, int K, int mode,
+                        float* C, hipStream_t stream) {
+  dim3 grid((M + impala::BM - 1) / impala::BM,
+            (N + impala::BN - 1) / impala::BN);
+  hipLaunchKernelGGL(impala::linear_mfma, grid, dim3(impala::THREADS), 0,
+                     stream, A, B, bias, ymask, M, N, K, mode, C);
+}
+
+}  // extern "C"

@@ -174,7 +174,8 @@ class ShallowNet(_CoreMixin, _HeadsMixin):
             from torched_impala_amd import ops
             if ops.hip_available():
                 from torched_impala_amd.ops.encoder import shallow_encode_conv
-                return F.relu(self.fc(shallow_encode_conv(self, x)))
+                from torched_impala_amd.ops.linear import linear_relu
+                return linear_relu(self.fc, shallow_encode_conv(self, x))
         x = x.float() / 255.0
         x = F.relu(self.conv1(x))
         x = F.relu(self.conv2(x))

@@ -1,0 +1,37 @@
+"""MFMA linear op: Y = relu(X @ W^T + b) on v_mfma_f32_16x16x4_f32
+(csrc/linear_kernels.hip). Forward fuses bias+ReLU; backward-dX fuses
+the ReLU mask into the A-staging pass. dW/db are rocBLAS reductions over
+the big M dim (transposed layouts consumed natively)."""
+from __future__ import annotations
+
+import torch
+
+from torched_impala_amd import ops
+
+
+class _LinearReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ext = ops.ext_hip()
+        y = ext.linear_mfma(x.contiguous(), w.contiguous(),
+                            b.contiguous(), None, 1)
+        ctx.save_for_backward(x, w, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops.ext_hip()
+        x, w, y = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = None
+        if ctx.needs_input_grad[0]:
+            wt = w.detach().t().contiguous()
+            dx = ext.linear_mfma(dy, wt, None, y, 2)  # mask fused
+        dym = dy * (y > 0)
+        dw = dym.t().mm(x) if ctx.needs_input_grad[1] else None
+        db = dym.sum(0) if ctx.needs_input_grad[2] else None
+        return dx, dw, db
+
+
+def linear_relu(module: torch.nn.Linear, x: torch.Tensor) -> torch.Tensor:
+    return _LinearReLU.apply(x, module.weight, module.bias)

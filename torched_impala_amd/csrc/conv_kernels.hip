@@ -2,29 +2,27 @@
 // "batched Conv2d encoder" / SURVEY.md §2c conv_encoder_fwd/bwd).
 //
 // MI355X-native design: the Atari frame is TINY (4x84x84 u8 = 28 KiB),
-// so instead of im2col GEMMs with NCHW<->NHWC transposes (what the
-// MIOpen path does — rocprof showed igemm + batched_transpose +
-// elementwise chains per layer), ONE workgroup owns ONE image and runs
-// the whole conv1->conv2->conv3 chain out of LDS:
+// so instead of im2col GEMMs with NCHW<->NHWC transposes (the MIOpen
+// path: igemm + batched_transpose + elementwise chains per layer), ONE
+// workgroup owns ONE image and runs conv1->conv2->conv3 out of LDS:
 //
 //   u8 frame (28 KiB, read directly — /255 fused, no float frame pass)
 //   -> conv1 16x8x8/s4 -> a1 (25.6 KiB LDS) -> conv2 32x4x4/s2
-//   -> a2 (10.4 KiB)   -> conv3 32x3x3    -> a3 (6.1 KiB) -> global.
+//   -> a2 (10.4 KiB)   -> conv3 32x3x3    -> a3 -> global.
 //
-// Peak LDS ~64 KiB -> 2 workgroups/CU; grid = N images (672 at the
-// headline batch) covers the chip. Weights stream from L2 (shared by
-// every workgroup; conv1 row fits L1 per filter). Bias+ReLU fused.
-// All math fp32, bit-comparable to the eager fp32 reference (u8->f32
-// convert and *(1/255) match torch's x.float()/255).
+// Register blocking (the v1 lesson, measured): weights must be loaded
+// once per (c,ky) chunk into registers and reused across an
+// ACCUMULATOR ARRAY of output pixels — a w-load per MAC turns the
+// whole kernel issue/L1-bound (321 us/step vs ~25 us of math).
+// Occupancy pinned at 2 workgroups/CU (launch_bounds + <=64 KiB LDS).
 //
-// Backward (same per-image structure, reverse order) emits per-image
-// weight-grad slabs (summed over images by one reduction outside) and
-// d(flat conv3 output) comes in from the fc backward; no d(frame) is
-// needed (frames are inputs). The ReLU masks come from the saved
-// activations.
+// Backward is THREE kernels (dW3+da2 | dW2+da1 | dW1), not one: a
+// single kernel's register pressure across phases forced ~200 B/lane
+// scratch spills inside the loops (2.4 ms/step measured). Intermediate
+// d2/d1 round-trip through global (~24 MB/step, ~8 us at HBM rate).
+// Per-image dW/db slabs are reduced by one sum outside (deterministic).
 //
-// The fc / heads GEMMs stay on the MFMA BLAS path (rocBLAS) — GEMM-
-// shaped work at M=672..52k belongs there; see ops/encoder.py.
+// fc / heads GEMMs: csrc/linear_kernels.hip (MFMA) / rocBLAS.
 
 #include <hip/hip_runtime.h>
 
@@ -46,6 +44,7 @@ constexpr int DW1_N = F1 * C0 * K1 * K1;           // 4096
 constexpr int DW2_N = F2 * F1 * K2 * K2;           // 8192
 constexpr int DW3_N = F3 * F2 * K3 * K3;           // 9216
 constexpr int SLAB_N = DW1_N + DW2_N + DW3_N + F1 + F2 + F3;  // 21584
+constexpr int DB_OFF = DW1_N + DW2_N + DW3_N;
 
 __device__ __forceinline__ float relu(float x) { return x > 0.f ? x : 0.f; }
 
@@ -71,44 +70,55 @@ shallow_encoder_fwd(const unsigned char* __restrict__ img,
     const unsigned int* src =
         reinterpret_cast<const unsigned int*>(img + (size_t)n * IMG_BYTES);
     unsigned int* dst = reinterpret_cast<unsigned int*>(s_img);
+    #pragma unroll 1
     for (int i = tid; i < IMG_BYTES / 4; i += NTHREADS) dst[i] = src[i];
   }
   __syncthreads();
 
-  // ---- conv1: 16 filters x 20x20, k=8 s=4, u8 input (fused /255) ----
-  // thread t: filter f = t/16, pixel-lane p = t%16 walks the 400 pixels.
+  // ---- conv1: thread = (filter f = tid/16, lane = tid%16); lane owns
+  //      25 pixels (px = lane, lane+16, ...). Weights per (c,ky): 8
+  //      floats in registers, reused across all 25 accumulators. ----
   {
     const float inv255 = 1.0f / 255.0f;
     const int f = tid >> 4;
     const int lane = tid & 15;
     const float* wf = w1 + f * (C0 * K1 * K1);
-    const float bias = b1[f];
-    #pragma unroll 1
-    for (int px = lane; px < H1 * W1; px += 16) {
-      const int oy = px / W1, ox = px - oy * W1;
-      float acc = 0.f;
-      for (int c = 0; c < C0; ++c) {
-        const unsigned char* ic = s_img + c * (H0 * W0);
-        const float* wc = wf + c * (K1 * K1);
-        #pragma unroll 2
-        for (int ky = 0; ky < K1; ++ky) {
-          // 8 u8 pixels = 2 aligned dwords (4*ox is dword-aligned)
+    float acc[25];
+    #pragma unroll
+    for (int i = 0; i < 25; ++i) acc[i] = 0.f;
+    for (int c = 0; c < C0; ++c) {
+      const unsigned char* ic = s_img + c * (H0 * W0);
+      const float* wc = wf + c * (K1 * K1);
+      #pragma unroll 1
+      for (int ky = 0; ky < K1; ++ky) {
+        const float4 wlo = *reinterpret_cast<const float4*>(wc + ky * K1);
+        const float4 whi = *reinterpret_cast<const float4*>(wc + ky * K1 + 4);
+        #pragma unroll
+        for (int i = 0; i < 25; ++i) {
+          const int px = lane + i * 16;
+          const int oy = px / W1, ox = px - oy * W1;
           const unsigned char* row = ic + (S1 * oy + ky) * W0 + S1 * ox;
           const unsigned int lo = *reinterpret_cast<const unsigned int*>(row);
           const unsigned int hi =
               *reinterpret_cast<const unsigned int*>(row + 4);
-          const float* wr = wc + ky * K1;
-          acc = fmaf((float)(lo & 0xff), wr[0], acc);
-          acc = fmaf((float)((lo >> 8) & 0xff), wr[1], acc);
-          acc = fmaf((float)((lo >> 16) & 0xff), wr[2], acc);
-          acc = fmaf((float)(lo >> 24), wr[3], acc);
-          acc = fmaf((float)(hi & 0xff), wr[4], acc);
-          acc = fmaf((float)((hi >> 8) & 0xff), wr[5], acc);
-          acc = fmaf((float)((hi >> 16) & 0xff), wr[6], acc);
-          acc = fmaf((float)(hi >> 24), wr[7], acc);
+          float a = acc[i];
+          a = fmaf((float)(lo & 0xff), wlo.x, a);
+          a = fmaf((float)((lo >> 8) & 0xff), wlo.y, a);
+          a = fmaf((float)((lo >> 16) & 0xff), wlo.z, a);
+          a = fmaf((float)(lo >> 24), wlo.w, a);
+          a = fmaf((float)(hi & 0xff), whi.x, a);
+          a = fmaf((float)((hi >> 8) & 0xff), whi.y, a);
+          a = fmaf((float)((hi >> 16) & 0xff), whi.z, a);
+          a = fmaf((float)(hi >> 24), whi.w, a);
+          acc[i] = a;
         }
       }
-      const float v = relu(fmaf(acc, inv255, bias));
+    }
+    const float bias = b1[f];
+    #pragma unroll
+    for (int i = 0; i < 25; ++i) {
+      const int px = lane + i * 16;
+      const float v = relu(fmaf(acc[i], inv255, bias));
       const int idx = f * (H1 * W1) + px;
       s_a1[idx] = v;
       if (save_for_bwd) a1g[(size_t)n * A1_N + idx] = v;
@@ -116,113 +126,118 @@ shallow_encoder_fwd(const unsigned char* __restrict__ img,
   }
   __syncthreads();
 
-  // ---- conv2: 32 filters x 9x9, k=4 s=2, fp32 from LDS ----
-  // thread t: filter f = t/8, pixel-lane p = t%8 walks the 81 pixels.
+  // ---- conv2: thread = (f = tid/8, lane = tid%8); 11 accumulators
+  //      (last partially valid). Weights per (c,ky): 4 floats. ----
   {
     const int f = tid >> 3;
     const int lane = tid & 7;
     const float* wf = w2 + f * (F1 * K2 * K2);
-    const float bias = b2[f];
-    #pragma unroll 1
-    for (int px = lane; px < H2 * W2; px += 8) {
-      const int oy = px / W2, ox = px - oy * W2;
-      float acc = bias;
-      for (int c = 0; c < F1; ++c) {
-        const float* ic = s_a1 + c * (H1 * W1);
-        const float* wc = wf + c * (K2 * K2);
+    float acc[11];
+    #pragma unroll
+    for (int i = 0; i < 11; ++i) acc[i] = 0.f;
+    for (int c = 0; c < F1; ++c) {
+      const float* ic = s_a1 + c * (H1 * W1);
+      const float* wc = wf + c * (K2 * K2);
+      #pragma unroll 1
+      for (int ky = 0; ky < K2; ++ky) {
+        const float4 w4 = *reinterpret_cast<const float4*>(wc + ky * K2);
         #pragma unroll
-        for (int ky = 0; ky < K2; ++ky) {
-          const float* row = ic + (S2 * oy + ky) * W1 + S2 * ox;
-          const float* wr = wc + ky * K2;
-          acc = fmaf(row[0], wr[0], acc);
-          acc = fmaf(row[1], wr[1], acc);
-          acc = fmaf(row[2], wr[2], acc);
-          acc = fmaf(row[3], wr[3], acc);
+        for (int i = 0; i < 11; ++i) {
+          const int px = lane + i * 8;
+          if (px < H2 * W2) {
+            const int oy = px / W2, ox = px - oy * W2;
+            const float* row = ic + (S2 * oy + ky) * W1 + S2 * ox;
+            float a = acc[i];
+            a = fmaf(row[0], w4.x, a);
+            a = fmaf(row[1], w4.y, a);
+            a = fmaf(row[2], w4.z, a);
+            a = fmaf(row[3], w4.w, a);
+            acc[i] = a;
+          }
         }
       }
-      const float v = relu(acc);
-      const int idx = f * (H2 * W2) + px;
-      s_a2[idx] = v;
-      if (save_for_bwd) a2g[(size_t)n * A2_N + idx] = v;
+    }
+    const float bias = b2[f];
+    #pragma unroll
+    for (int i = 0; i < 11; ++i) {
+      const int px = lane + i * 8;
+      if (px < H2 * W2) {
+        const float v = relu(acc[i] + bias);
+        const int idx = f * (H2 * W2) + px;
+        s_a2[idx] = v;
+        if (save_for_bwd) a2g[(size_t)n * A2_N + idx] = v;
+      }
     }
   }
   __syncthreads();
 
-  // ---- conv3: 32 filters x 7x7, k=3 s=1 ----
+  // ---- conv3: thread = (f = tid/8, lane = tid%8); 7 accumulators. ----
   {
     const int f = tid >> 3;
     const int lane = tid & 7;
     const float* wf = w3 + f * (F2 * K3 * K3);
-    const float bias = b3[f];
-    #pragma unroll 1
-    for (int px = lane; px < H3 * W3; px += 8) {
-      const int oy = px / W3, ox = px - oy * W3;
-      float acc = bias;
-      for (int c = 0; c < F2; ++c) {
-        const float* ic = s_a2 + c * (H2 * W2);
-        const float* wc = wf + c * (K3 * K3);
+    float acc[7];
+    #pragma unroll
+    for (int i = 0; i < 7; ++i) acc[i] = 0.f;
+    for (int c = 0; c < F2; ++c) {
+      const float* ic = s_a2 + c * (H2 * W2);
+      const float* wc = wf + c * (K3 * K3);
+      #pragma unroll 1
+      for (int ky = 0; ky < K3; ++ky) {
+        const float wa = wc[ky * K3 + 0];
+        const float wb = wc[ky * K3 + 1];
+        const float wcc = wc[ky * K3 + 2];
         #pragma unroll
-        for (int ky = 0; ky < K3; ++ky) {
-          const float* row = ic + (oy + ky) * W2 + ox;
-          const float* wr = wc + ky * K3;
-          acc = fmaf(row[0], wr[0], acc);
-          acc = fmaf(row[1], wr[1], acc);
-          acc = fmaf(row[2], wr[2], acc);
+        for (int i = 0; i < 7; ++i) {
+          const int px = lane + i * 8;
+          if (px < H3 * W3) {
+            const int oy = px / W3, ox = px - oy * W3;
+            const float* row = ic + (oy + ky) * W2 + ox;
+            float a = acc[i];
+            a = fmaf(row[0], wa, a);
+            a = fmaf(row[1], wb, a);
+            a = fmaf(row[2], wcc, a);
+            acc[i] = a;
+          }
         }
       }
-      // a3 layout matches torch flatten(1): [f][oy][ox]
-      a3g[(size_t)n * A3_N + f * (H3 * W3) + px] = relu(acc);
+    }
+    const float bias = b3[f];
+    #pragma unroll
+    for (int i = 0; i < 7; ++i) {
+      const int px = lane + i * 8;
+      if (px < H3 * W3)
+        a3g[(size_t)n * A3_N + f * (H3 * W3) + px] = relu(acc[i] + bias);
     }
   }
 }
 
 // ---------------------------------------------------------------------------
-// backward: grid = (N) blocks x 256 threads.
-// In: d_a3 (N,1568) = d(flat conv3 out), saved a1/a2/a3, frame, weights.
-// Out: per-image slab [dW1|dW2|dW3|db1|db2|db3] (N, SLAB_N); reduced
-// over N outside. No d(frame).
-//
-// LDS is a phased arena (78.4 KiB -> 2 workgroups/CU): the u8 frame is
-// loaded LAST, over the then-dead a2/d3/d2 regions, after db3/db2 are
-// banked — a static per-buffer layout needs 106 KiB and halves
-// occupancy.
+// backward kernel 1: dW3 + db3 + d_a2 (relu-masked) -> d2g (N, A2_N)
+// LDS: a2 + d3 + d2 = 27 KiB
 // ---------------------------------------------------------------------------
-constexpr int ARENA_F = (A2_N + A3_N + A2_N + A1_N + A1_N);  // floats
 __global__ void __launch_bounds__(NTHREADS, 2)
-shallow_encoder_bwd(const unsigned char* __restrict__ img,
-                    const float* __restrict__ a1g,
-                    const float* __restrict__ a2g,
-                    const float* __restrict__ a3g,
-                    const float* __restrict__ d_a3g,
-                    const float* __restrict__ w2, const float* __restrict__ w3,
-                    int N, float* __restrict__ slab) {
-  __shared__ __align__(16) float s_arena[ARENA_F];
-  float* s_a2 = s_arena;                 // A2_N
-  float* s_d3 = s_a2 + A2_N;             // A3_N
-  float* s_d2 = s_d3 + A3_N;             // A2_N
-  float* s_a1 = s_d2 + A2_N;             // A1_N
-  float* s_d1 = s_a1 + A1_N;             // A1_N
-  // phase-5 overlay: the u8 frame covers [s_a2 .. s_a2+28224B) — the
-  // dead a2/d3/d2 (+ start of a1) regions
-  unsigned char* s_img = reinterpret_cast<unsigned char*>(s_arena);
-
+shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
+               const float* __restrict__ d_a3g, const float* __restrict__ w3,
+               int N, float* __restrict__ slab, float* __restrict__ d2g) {
+  __shared__ __align__(16) float s_a2[A2_N];
+  __shared__ __align__(16) float s_d3[A3_N];
+  __shared__ __align__(16) float s_d2[A2_N];
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   float* my_slab = slab + (size_t)n * SLAB_N;
-  const int DB_OFF = DW1_N + DW2_N + DW3_N;
 
-  // stage a2, d3 (relu-masked), a1
+  #pragma unroll 1
   for (int i = tid; i < A2_N; i += NTHREADS)
     s_a2[i] = a2g[(size_t)n * A2_N + i];
+  #pragma unroll 1
   for (int i = tid; i < A3_N; i += NTHREADS) {
     const float m = a3g[(size_t)n * A3_N + i] > 0.f ? 1.f : 0.f;
     s_d3[i] = m * d_a3g[(size_t)n * A3_N + i];
   }
-  for (int i = tid; i < A1_N; i += NTHREADS)
-    s_a1[i] = a1g[(size_t)n * A1_N + i];
   __syncthreads();
 
-  // ---- dW3[f][c][ky][kx] = sum_px s_a2[c][oy+ky][ox+kx] * s_d3[f][px] ----
+  // dW3[f][c][ky][kx] = sum_px a2[c][oy+ky][ox+kx] * d3[f][px]
   #pragma unroll 1
   for (int e = tid; e < DW3_N; e += NTHREADS) {
     const int f = e / (F2 * K3 * K3);
@@ -233,6 +248,7 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     const float* dc = s_d3 + f * (H3 * W3);
     const float* ac = s_a2 + c * (H2 * W2);
     float acc = 0.f;
+    #pragma unroll 1
     for (int oy = 0; oy < H3; ++oy) {
       const float* arow = ac + (oy + ky) * W2 + kx;
       const float* drow = dc + oy * W3;
@@ -241,32 +257,70 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     }
     my_slab[DW1_N + DW2_N + e] = acc;
   }
+  for (int f = tid; f < F3; f += NTHREADS) {
+    float acc = 0.f;
+    const float* dc = s_d3 + f * (H3 * W3);
+    for (int p = 0; p < H3 * W3; ++p) acc += dc[p];
+    my_slab[DB_OFF + F1 + F2 + f] = acc;
+  }
 
-  // ---- d_a2 (masked by a2>0) ----
+  // d_a2[c][y][x]: bounds computed once per element, branch-free inner
   #pragma unroll 1
   for (int e = tid; e < A2_N; e += NTHREADS) {
     const int c = e / (H2 * W2);
     const int p = e - c * (H2 * W2);
     const int y = p / W2, x = p - y * W2;
+    const int ky_lo = max(0, y - (H3 - 1)), ky_hi = min(K3 - 1, y);
+    const int kx_lo = max(0, x - (W3 - 1)), kx_hi = min(K3 - 1, x);
     float acc = 0.f;
+    #pragma unroll 1
     for (int f = 0; f < F3; ++f) {
       const float* dc = s_d3 + f * (H3 * W3);
       const float* wc = w3 + (f * F2 + c) * (K3 * K3);
-      for (int ky = 0; ky < K3; ++ky) {
-        const int oy = y - ky;
-        if (oy < 0 || oy >= H3) continue;
-        for (int kx = 0; kx < K3; ++kx) {
-          const int ox = x - kx;
-          if (ox < 0 || ox >= W3) continue;
-          acc = fmaf(dc[oy * W3 + ox], wc[ky * K3 + kx], acc);
-        }
+      for (int ky = ky_lo; ky <= ky_hi; ++ky) {
+        const float* drow = dc + (y - ky) * W3 + x;
+        const float* wrow = wc + ky * K3;
+        for (int kx = kx_lo; kx <= kx_hi; ++kx)
+          acc = fmaf(drow[-kx], wrow[kx], acc);
       }
     }
-    s_d2[e] = s_a2[e] > 0.f ? acc : 0.f;
+    const float v = s_a2[e] > 0.f ? acc : 0.f;
+    s_d2[e] = v;
+    d2g[(size_t)n * A2_N + e] = v;
   }
   __syncthreads();
+  for (int f = tid; f < F2; f += NTHREADS) {
+    float acc = 0.f;
+    const float* dc = s_d2 + f * (H2 * W2);
+    for (int p = 0; p < H2 * W2; ++p) acc += dc[p];
+    my_slab[DB_OFF + F1 + f] = acc;
+  }
+}
 
-  // ---- dW2 ----
+// ---------------------------------------------------------------------------
+// backward kernel 2: dW2 + d_a1 (relu-masked) -> d1g (N, A1_N) + db1
+// LDS: a1 + d2 + d1 = 61.6 KiB
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(NTHREADS, 2)
+shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
+               const float* __restrict__ w2, int N,
+               float* __restrict__ slab, float* __restrict__ d1g) {
+  __shared__ __align__(16) float s_a1[A1_N];
+  __shared__ __align__(16) float s_d2[A2_N];
+  __shared__ __align__(16) float s_d1[A1_N];
+  const int n = blockIdx.x;
+  const int tid = threadIdx.x;
+  float* my_slab = slab + (size_t)n * SLAB_N;
+
+  #pragma unroll 1
+  for (int i = tid; i < A1_N; i += NTHREADS)
+    s_a1[i] = a1g[(size_t)n * A1_N + i];
+  #pragma unroll 1
+  for (int i = tid; i < A2_N; i += NTHREADS)
+    s_d2[i] = d2g[(size_t)n * A2_N + i];
+  __syncthreads();
+
+  // dW2[f][c][ky][kx] = sum_px a1[c][2oy+ky][2ox+kx] * d2[f][px]
   #pragma unroll 1
   for (int e = tid; e < DW2_N; e += NTHREADS) {
     const int f = e / (F1 * K2 * K2);
@@ -277,6 +331,7 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     const float* dc = s_d2 + f * (H2 * W2);
     const float* ac = s_a1 + c * (H1 * W1);
     float acc = 0.f;
+    #pragma unroll 1
     for (int oy = 0; oy < H2; ++oy) {
       const float* arow = ac + (S2 * oy + ky) * W1 + kx;
       const float* drow = dc + oy * W2;
@@ -287,86 +342,92 @@ shallow_encoder_bwd(const unsigned char* __restrict__ img,
     my_slab[DW1_N + e] = acc;
   }
 
-  // ---- d_a1 (masked by a1>0) ----
+  // d_a1[c][y][x]: stride-2 transposed conv — per axis exactly two
+  // parity-matched kernel offsets; bounds clamp, branch-light inner.
   #pragma unroll 1
   for (int e = tid; e < A1_N; e += NTHREADS) {
     const int c = e / (H1 * W1);
     const int p = e - c * (H1 * W1);
     const int y = p / W1, x = p - y * W1;
+    const int ky0 = (y & 1), kx0 = (x & 1);
     float acc = 0.f;
+    #pragma unroll 1
     for (int f = 0; f < F2; ++f) {
       const float* dc = s_d2 + f * (H2 * W2);
       const float* wc = w2 + (f * F1 + c) * (K2 * K2);
-      for (int ky = 0; ky < K2; ++ky) {
-        const int ty = y - ky;
-        if (ty < 0 || (ty & 1)) continue;
-        const int oy = ty >> 1;
-        if (oy >= H2) continue;
-        for (int kx = 0; kx < K2; ++kx) {
-          const int tx = x - kx;
-          if (tx < 0 || (tx & 1)) continue;
-          const int ox = tx >> 1;
-          if (ox >= W2) continue;
+      #pragma unroll
+      for (int a = 0; a < 2; ++a) {
+        const int ky = ky0 + 2 * a;
+        const int oy = (y - ky) >> 1;
+        if (oy < 0 || oy >= H2) continue;
+        #pragma unroll
+        for (int bq = 0; bq < 2; ++bq) {
+          const int kx = kx0 + 2 * bq;
+          const int ox = (x - kx) >> 1;
+          if (ox < 0 || ox >= W2) continue;
           acc = fmaf(dc[oy * W2 + ox], wc[ky * K2 + kx], acc);
         }
       }
     }
-    s_d1[e] = s_a1[e] > 0.f ? acc : 0.f;
+    const float v = s_a1[e] > 0.f ? acc : 0.f;
+    s_d1[e] = v;
+    d1g[(size_t)n * A1_N + e] = v;
   }
   __syncthreads();
-
-  // ---- bank db3/db2/db1 BEFORE the frame overlays their buffers ----
-  for (int f = tid; f < F3; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d3 + f * (H3 * W3);
-    for (int p = 0; p < H3 * W3; ++p) acc += dc[p];
-    my_slab[DB_OFF + F1 + F2 + f] = acc;
-  }
-  for (int f = tid; f < F2; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d2 + f * (H2 * W2);
-    for (int p = 0; p < H2 * W2; ++p) acc += dc[p];
-    my_slab[DB_OFF + F1 + f] = acc;
-  }
   for (int f = tid; f < F1; f += NTHREADS) {
     float acc = 0.f;
     const float* dc = s_d1 + f * (H1 * W1);
     for (int p = 0; p < H1 * W1; ++p) acc += dc[p];
     my_slab[DB_OFF + f] = acc;
   }
-  __syncthreads();
+}
 
-  // ---- overlay: stage the u8 frame over the dead a2/d3/d2 regions ----
+// ---------------------------------------------------------------------------
+// backward kernel 3: dW1 from the u8 frame (fused /255)
+// LDS: img + d1 = 53.6 KiB
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(NTHREADS, 2)
+shallow_bwd_k3(const unsigned char* __restrict__ img,
+               const float* __restrict__ d1g, int N,
+               float* __restrict__ slab) {
+  __shared__ __align__(16) unsigned char s_img[IMG_BYTES];
+  __shared__ __align__(16) float s_d1[A1_N];
+  const int n = blockIdx.x;
+  const int tid = threadIdx.x;
+  float* my_slab = slab + (size_t)n * SLAB_N;
+
   {
     const unsigned int* src =
         reinterpret_cast<const unsigned int*>(img + (size_t)n * IMG_BYTES);
     unsigned int* dst = reinterpret_cast<unsigned int*>(s_img);
+    #pragma unroll 1
     for (int i = tid; i < IMG_BYTES / 4; i += NTHREADS) dst[i] = src[i];
   }
+  #pragma unroll 1
+  for (int i = tid; i < A1_N; i += NTHREADS)
+    s_d1[i] = d1g[(size_t)n * A1_N + i];
   __syncthreads();
 
-  // ---- dW1 from the u8 frame (fused /255) ----
-  {
-    const float inv255 = 1.0f / 255.0f;
+  const float inv255 = 1.0f / 255.0f;
+  #pragma unroll 1
+  for (int e = tid; e < DW1_N; e += NTHREADS) {
+    const int f = e / (C0 * K1 * K1);
+    const int r = e - f * (C0 * K1 * K1);
+    const int c = r / (K1 * K1);
+    const int k = r - c * (K1 * K1);
+    const int ky = k / K1, kx = k - ky * K1;
+    const float* dc = s_d1 + f * (H1 * W1);
+    const unsigned char* ic = s_img + c * (H0 * W0);
+    float acc = 0.f;
     #pragma unroll 1
-    for (int e = tid; e < DW1_N; e += NTHREADS) {
-      const int f = e / (C0 * K1 * K1);
-      const int r = e - f * (C0 * K1 * K1);
-      const int c = r / (K1 * K1);
-      const int k = r - c * (K1 * K1);
-      const int ky = k / K1, kx = k - ky * K1;
-      const float* dc = s_d1 + f * (H1 * W1);
-      const unsigned char* ic = s_img + c * (H0 * W0);
-      float acc = 0.f;
-      for (int oy = 0; oy < H1; ++oy) {
-        const unsigned char* irow = ic + (S1 * oy + ky) * W0 + kx;
-        const float* drow = dc + oy * W1;
-        #pragma unroll 4
-        for (int ox = 0; ox < W1; ++ox)
-          acc = fmaf((float)irow[S1 * ox], drow[ox], acc);
-      }
-      my_slab[e] = acc * inv255;
+    for (int oy = 0; oy < H1; ++oy) {
+      const unsigned char* irow = ic + (S1 * oy + ky) * W0 + kx;
+      const float* drow = dc + oy * W1;
+      #pragma unroll 4
+      for (int ox = 0; ox < W1; ++ox)
+        acc = fmaf((float)irow[S1 * ox], drow[ox], acc);
     }
+    my_slab[e] = acc * inv255;
   }
 }
 
@@ -385,14 +446,21 @@ void launch_shallow_encoder_fwd(const unsigned char* img, const float* w1,
                      w3, b3, N, a1, a2, a3, save_for_bwd);
 }
 
+// d2_tmp: (N, A2_N); d1_tmp: (N, A1_N) — caller-provided workspaces
 void launch_shallow_encoder_bwd(const unsigned char* img, const float* a1,
                                 const float* a2, const float* a3,
                                 const float* d_a3, const float* w2,
                                 const float* w3, int N, float* slab,
+                                float* d2_tmp, float* d1_tmp,
                                 hipStream_t stream) {
-  hipLaunchKernelGGL(impala::shallow_encoder_bwd, dim3(N),
-                     dim3(impala::NTHREADS), 0, stream, img, a1, a2, a3,
-                     d_a3, w2, w3, N, slab);
+  hipLaunchKernelGGL(impala::shallow_bwd_k1, dim3(N),
+                     dim3(impala::NTHREADS), 0, stream, a2, a3, d_a3, w3, N,
+                     slab, d2_tmp);
+  hipLaunchKernelGGL(impala::shallow_bwd_k2, dim3(N),
+                     dim3(impala::NTHREADS), 0, stream, a1, d2_tmp, w2, N,
+                     slab, d1_tmp);
+  hipLaunchKernelGGL(impala::shallow_bwd_k3, dim3(N),
+                     dim3(impala::NTHREADS), 0, stream, img, d1_tmp, N, slab);
 }
 
 int shallow_encoder_slab_elems() { return impala::SLAB_N; }

@@ -49,7 +49,7 @@ void launch_shallow_encoder_fwd(const unsigned char*, const float*,
 void launch_shallow_encoder_bwd(const unsigned char*, const float*,
                                 const float*, const float*, const float*,
                                 const float*, const float*, int, float*,
-                                hipStream_t);
+                                float*, float*, hipStream_t);
 int shallow_encoder_slab_elems();
 void launch_linear_mfma(const float*, const float*, const float*,
                         const float*, int, int, int, int, float*,
@@ -343,11 +343,14 @@ torch::Tensor shallow_encoder_bwd(
   DeviceScope guard(frames.get_device());
   auto slab = torch::empty({N, (long)shallow_encoder_slab_elems()},
                            d_a3.options());
+  auto d2_tmp = torch::empty({N, 32 * 9 * 9}, d_a3.options());
+  auto d1_tmp = torch::empty({N, 16 * 20 * 20}, d_a3.options());
   launch_shallow_encoder_bwd(
       frames.data_ptr<unsigned char>(), a1.data_ptr<float>(),
       a2.data_ptr<float>(), a3.data_ptr<float>(), d_a3.data_ptr<float>(),
       w2.data_ptr<float>(), w3.data_ptr<float>(), N,
-      slab.data_ptr<float>(), current_stream());
+      slab.data_ptr<float>(), d2_tmp.data_ptr<float>(),
+      d1_tmp.data_ptr<float>(), current_stream());
   HIP_CHECK(hipGetLastError());
   return slab;
 }

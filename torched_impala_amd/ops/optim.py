@@ -82,6 +82,9 @@ class FlatRMSprop(torch.optim.Optimizer):
         Call OUTSIDE any captured graph, once per step."""
         self.lr_t.fill_(self.param_groups[0]["lr"])
         self._steps += 1
+        # keep LRScheduler's step-order bookkeeping happy in graph mode,
+        # where optimizer.step() itself is inside the captured graph
+        self._step_count = getattr(self, "_step_count", 0) + 1
 
     @torch.no_grad()
     def step_device(self):

@@ -220,9 +220,12 @@ __global__ void __launch_bounds__(NTHREADS, 2)
 shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
                const float* __restrict__ d_a3g, const float* __restrict__ w3,
                int N, float* __restrict__ slab, float* __restrict__ d2g) {
+  // grid: (N, ESPLIT) — blockIdx.y picks an element-range slice so the
+  // per-image serial loops shrink (the loops are latency-bound; more
+  // blocks beat deeper ILP here)
+  const int esplit = gridDim.y, eslice = blockIdx.y;
   __shared__ __align__(16) float s_a2[A2_N];
   __shared__ __align__(16) float s_d3[A3_N];
-  __shared__ __align__(16) float s_d2[A2_N];
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   float* my_slab = slab + (size_t)n * SLAB_N;
@@ -239,7 +242,8 @@ shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
 
   // dW3[f][c][ky][kx] = sum_px a2[c][oy+ky][ox+kx] * d3[f][px]
   #pragma unroll 1
-  for (int e = tid; e < DW3_N; e += NTHREADS) {
+  for (int e = tid + eslice * NTHREADS; e < DW3_N;
+       e += NTHREADS * esplit) {
     const int f = e / (F2 * K3 * K3);
     const int r = e - f * (F2 * K3 * K3);
     const int c = r / (K3 * K3);
@@ -257,16 +261,19 @@ shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
     }
     my_slab[DW1_N + DW2_N + e] = acc;
   }
-  for (int f = tid; f < F3; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d3 + f * (H3 * W3);
-    for (int p = 0; p < H3 * W3; ++p) acc += dc[p];
-    my_slab[DB_OFF + F1 + F2 + f] = acc;
+  if (eslice == 0) {
+    for (int f = tid; f < F3; f += NTHREADS) {
+      float acc = 0.f;
+      const float* dc = s_d3 + f * (H3 * W3);
+      for (int p = 0; p < H3 * W3; ++p) acc += dc[p];
+      my_slab[DB_OFF + F1 + F2 + f] = acc;
+    }
   }
 
   // d_a2[c][y][x]: bounds computed once per element, branch-free inner
   #pragma unroll 1
-  for (int e = tid; e < A2_N; e += NTHREADS) {
+  for (int e = tid + eslice * NTHREADS; e < A2_N;
+       e += NTHREADS * esplit) {
     const int c = e / (H2 * W2);
     const int p = e - c * (H2 * W2);
     const int y = p / W2, x = p - y * W2;
@@ -285,15 +292,7 @@ shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
       }
     }
     const float v = s_a2[e] > 0.f ? acc : 0.f;
-    s_d2[e] = v;
     d2g[(size_t)n * A2_N + e] = v;
-  }
-  __syncthreads();
-  for (int f = tid; f < F2; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d2 + f * (H2 * W2);
-    for (int p = 0; p < H2 * W2; ++p) acc += dc[p];
-    my_slab[DB_OFF + F1 + f] = acc;
   }
 }
 
@@ -305,9 +304,9 @@ __global__ void __launch_bounds__(NTHREADS, 2)
 shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
                const float* __restrict__ w2, int N,
                float* __restrict__ slab, float* __restrict__ d1g) {
+  const int esplit = gridDim.y, eslice = blockIdx.y;
   __shared__ __align__(16) float s_a1[A1_N];
   __shared__ __align__(16) float s_d2[A2_N];
-  __shared__ __align__(16) float s_d1[A1_N];
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   float* my_slab = slab + (size_t)n * SLAB_N;
@@ -320,9 +319,19 @@ shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
     s_d2[i] = d2g[(size_t)n * A2_N + i];
   __syncthreads();
 
+  if (eslice == 0) {
+    for (int f = tid; f < F2; f += NTHREADS) {
+      float acc = 0.f;
+      const float* dc = s_d2 + f * (H2 * W2);
+      for (int p = 0; p < H2 * W2; ++p) acc += dc[p];
+      my_slab[DB_OFF + F1 + f] = acc;
+    }
+  }
+
   // dW2[f][c][ky][kx] = sum_px a1[c][2oy+ky][2ox+kx] * d2[f][px]
   #pragma unroll 1
-  for (int e = tid; e < DW2_N; e += NTHREADS) {
+  for (int e = tid + eslice * NTHREADS; e < DW2_N;
+       e += NTHREADS * esplit) {
     const int f = e / (F1 * K2 * K2);
     const int r = e - f * (F1 * K2 * K2);
     const int c = r / (K2 * K2);
@@ -345,7 +354,8 @@ shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
   // d_a1[c][y][x]: stride-2 transposed conv — per axis exactly two
   // parity-matched kernel offsets; bounds clamp, branch-light inner.
   #pragma unroll 1
-  for (int e = tid; e < A1_N; e += NTHREADS) {
+  for (int e = tid + eslice * NTHREADS; e < A1_N;
+       e += NTHREADS * esplit) {
     const int c = e / (H1 * W1);
     const int p = e - c * (H1 * W1);
     const int y = p / W1, x = p - y * W1;
@@ -370,15 +380,7 @@ shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
       }
     }
     const float v = s_a1[e] > 0.f ? acc : 0.f;
-    s_d1[e] = v;
     d1g[(size_t)n * A1_N + e] = v;
-  }
-  __syncthreads();
-  for (int f = tid; f < F1; f += NTHREADS) {
-    float acc = 0.f;
-    const float* dc = s_d1 + f * (H1 * W1);
-    for (int p = 0; p < H1 * W1; ++p) acc += dc[p];
-    my_slab[DB_OFF + f] = acc;
   }
 }
 
@@ -390,6 +392,7 @@ __global__ void __launch_bounds__(NTHREADS, 2)
 shallow_bwd_k3(const unsigned char* __restrict__ img,
                const float* __restrict__ d1g, int N,
                float* __restrict__ slab) {
+  const int esplit = gridDim.y, eslice = blockIdx.y;
   __shared__ __align__(16) unsigned char s_img[IMG_BYTES];
   __shared__ __align__(16) float s_d1[A1_N];
   const int n = blockIdx.x;
@@ -408,9 +411,19 @@ shallow_bwd_k3(const unsigned char* __restrict__ img,
     s_d1[i] = d1g[(size_t)n * A1_N + i];
   __syncthreads();
 
+  if (eslice == 0) {
+    for (int f = tid; f < F1; f += NTHREADS) {
+      float acc = 0.f;
+      const float* dc = s_d1 + f * (H1 * W1);
+      for (int p = 0; p < H1 * W1; ++p) acc += dc[p];
+      my_slab[DB_OFF + f] = acc;
+    }
+  }
+
   const float inv255 = 1.0f / 255.0f;
   #pragma unroll 1
-  for (int e = tid; e < DW1_N; e += NTHREADS) {
+  for (int e = tid + eslice * NTHREADS; e < DW1_N;
+       e += NTHREADS * esplit) {
     const int f = e / (C0 * K1 * K1);
     const int r = e - f * (C0 * K1 * K1);
     const int c = r / (K1 * K1);
@@ -453,13 +466,14 @@ void launch_shallow_encoder_bwd(const unsigned char* img, const float* a1,
                                 const float* w3, int N, float* slab,
                                 float* d2_tmp, float* d1_tmp,
                                 hipStream_t stream) {
-  hipLaunchKernelGGL(impala::shallow_bwd_k1, dim3(N),
+  const int ES = 4;  // element-range split: more blocks, shorter loops
+  hipLaunchKernelGGL(impala::shallow_bwd_k1, dim3(N, ES),
                      dim3(impala::NTHREADS), 0, stream, a2, a3, d_a3, w3, N,
                      slab, d2_tmp);
-  hipLaunchKernelGGL(impala::shallow_bwd_k2, dim3(N),
+  hipLaunchKernelGGL(impala::shallow_bwd_k2, dim3(N, ES),
                      dim3(impala::NTHREADS), 0, stream, a1, d2_tmp, w2, N,
                      slab, d1_tmp);
-  hipLaunchKernelGGL(impala::shallow_bwd_k3, dim3(N),
+  hipLaunchKernelGGL(impala::shallow_bwd_k3, dim3(N, ES),
                      dim3(impala::NTHREADS), 0, stream, img, d1_tmp, N, slab);
 }
 

@@ -34,12 +34,13 @@ def main():
     p.add_argument("--batch-size", type=int, default=32)
     p.add_argument("--unroll-length", type=int, default=20)
     p.add_argument("--num-actors", type=int, default=32)
-    p.add_argument("--envs-per-actor", type=int, default=8)
+    p.add_argument("--envs-per-actor", type=int, default=16)
     p.add_argument("--use-lstm", action="store_true", default=False)
     p.add_argument("--model", default="auto")
     p.add_argument("--no-hip-kernels", action="store_true")
     p.add_argument("--no-graphs", action="store_true")
     p.add_argument("--device", default="auto")
+    p.add_argument("--compute-dtype", default="fp32", choices=["fp32", "bf16"])
     args = p.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -62,6 +63,7 @@ def main():
         disable_checkpoint=True,
         use_hip_kernels=not args.no_hip_kernels,
         graph_capture=not args.no_graphs,
+        compute_dtype=args.compute_dtype,
         device=args.device,
         weight_sync_interval=1,
         seed=1234,
@@ -117,7 +119,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": flags.compute_dtype,
             "data": "synthetic",
             "config": {
                 "model": ("shallow+lstm" if flags.use_lstm else "shallow"),

@@ -35,6 +35,7 @@ def main():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--no-hip-kernels", action="store_true")
     p.add_argument("--num-actions", type=int, default=6)
+    p.add_argument("--compute-dtype", default="fp32", choices=["fp32", "bf16"])
     args = p.parse_args()
 
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
@@ -44,7 +45,8 @@ def main():
                   use_lstm=args.use_lstm,
                   batch_size=args.batch_size, unroll_length=args.unroll_length,
                   total_steps=1 << 50,
-                  use_hip_kernels=not args.no_hip_kernels)
+                  use_hip_kernels=not args.no_hip_kernels,
+                  compute_dtype=args.compute_dtype)
     torch.manual_seed(0)
     model = build_model(flags, obs, A).to(device)
     model.force_eager_core = not flags.use_hip_kernels
@@ -79,7 +81,7 @@ def main():
     fps = args.steps * T * B / dt
     print(json.dumps(dict(
         diagnostic="learner_capacity", model=args.model, lstm=args.use_lstm,
-        hip=not args.no_hip_kernels, T=T, B=B,
+        hip=not args.no_hip_kernels, dtype=args.compute_dtype, T=T, B=B,
         ms_per_step=round(ms, 3), frames_per_sec=round(fps, 1))))
 
 

@@ -189,7 +189,14 @@ def _fwd_bwd(flags: Flags, model, batch: Dict[str, torch.Tensor],
         core_state = (ics[0], ics[1])
     inputs = dict(frame=batch["frame"], reward=batch["reward"],
                   done=batch["done"], last_action=batch["last_action"])
-    outputs, _ = model(inputs, core_state, need_action=False)
+    if flags.compute_dtype == "bf16" and batch["frame"].is_cuda:
+        # bf16 matmul/conv compute (MFMA bf16 rate); heads re-cast to
+        # fp32, so V-trace math stays fp32. Mainly for the deep/big
+        # config — the headline benches run fp32.
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            outputs, _ = model(inputs, core_state, need_action=False)
+    else:
+        outputs, _ = model(inputs, core_state, need_action=False)
 
     logits_all = outputs["policy_logits"]      # (T+1, B, A)
     baseline_all = outputs["baseline"]         # (T+1, B)

@@ -18,7 +18,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def _worker(args, idx, q):
     import torch
-    torch.set_num_threads(1)
+    torch.set_num_threads(max(1, args.threads))
     from torched_impala_amd.config import Flags
     from torched_impala_amd.models import build_model
     from torched_impala_amd.actor import Rollout
@@ -43,6 +43,7 @@ def main():
     p.add_argument("--envs", type=int, default=16)
     p.add_argument("--seconds", type=float, default=8.0)
     p.add_argument("--use-lstm", action="store_true")
+    p.add_argument("--threads", type=int, default=1)
     args = p.parse_args()
     import torch.multiprocessing as mp
     ctx = mp.get_context("spawn")
@@ -57,6 +58,7 @@ def main():
     agg = sum(rates)
     print(json.dumps(dict(
         diagnostic="actor_probe", actors=args.actors, envs=args.envs,
+        threads=args.threads,
         lstm=args.use_lstm, per_actor_fps=round(agg / args.actors, 1),
         aggregate_fps=round(agg, 1))))
 

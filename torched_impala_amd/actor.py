@@ -116,7 +116,12 @@ def act(flags, actor_index: int, free_queue, full_queue, buffers, mirror,
         obs_shape, obs_dtype, num_actions):
     """Actor process entry point."""
     try:
-        torch.set_num_threads(1)
+        nthreads = flags.actor_threads
+        if nthreads <= 0:  # auto: share the cores, cap the oversubscription
+            import os as _os
+            nthreads = max(1, min(8, (_os.cpu_count() or 8)
+                                  // max(1, flags.num_actors)))
+        torch.set_num_threads(nthreads)
         torch.manual_seed(flags.seed + actor_index)
         model = build_model(flags, obs_shape, num_actions)
         model.eval()

@@ -31,6 +31,8 @@ class Flags:
     unroll_length: int = 20            # T
     num_buffers: Optional[int] = None  # ring slots; default max(2*B, B+num_actors)
     envs_per_actor: int = 1            # batched envs inside one actor process
+    actor_threads: int = 0             # CPU threads per actor; 0 = auto
+                                       # (cores/num_actors, capped at 8)
 
     # --- model ---
     use_lstm: bool = False
@@ -97,6 +99,7 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--unroll-length", type=int, default=f.unroll_length)
     p.add_argument("--num-buffers", type=int, default=None)
     p.add_argument("--envs-per-actor", type=int, default=f.envs_per_actor)
+    p.add_argument("--actor-threads", type=int, default=f.actor_threads)
     _add_bool(p, "use-lstm", f.use_lstm)
     p.add_argument("--model", default=f.model,
                    choices=["auto", "mlp", "shallow", "deep"])

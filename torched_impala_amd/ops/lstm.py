@@ -61,10 +61,13 @@ def lstm_core_forward(core: torch.nn.LSTMCell, core_input: torch.Tensor,
     T, B, F = core_input.shape
     assert core.hidden_size == 256, "HIP LSTM core is specialized for 256"
     h0, c0 = core_state
-    bias = core.bias_ih + core.bias_hh
-    Xg = torch.addmm(bias, core_input.reshape(T * B, F).float(),
-                     core.weight_ih.t()).view(T, B, 1024)
-    notdone = (~done).float()
-    H, hT, cT = _LstmSeq.apply(Xg, core.weight_hh, h0.float(), c0.float(),
-                               notdone)
+    # the recurrence kernel is fp32; keep the gate GEMM out of any
+    # surrounding autocast region (which would emit bf16)
+    with torch.autocast("cuda", enabled=False):
+        bias = core.bias_ih + core.bias_hh
+        Xg = torch.addmm(bias, core_input.reshape(T * B, F).float(),
+                         core.weight_ih.t()).view(T, B, 1024)
+        notdone = (~done).float()
+        H, hT, cT = _LstmSeq.apply(Xg, core.weight_hh, h0.float(),
+                                   c0.float(), notdone)
     return H.reshape(T * B, 256), (hT, cT)

@@ -35,7 +35,10 @@ def init_distributed(device: str) -> tuple[int, int]:
         dist.init_process_group(backend=backend)
     rank = dist.get_rank()
     if device == "cuda":
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        local = int(os.environ.get("LOCAL_RANK", rank))
+        # modulo lets DP logic be exercised with ranks > GPUs (e.g. two
+        # RCCL ranks sharing the single gpurun GPU — SURVEY.md §4.5)
+        torch.cuda.set_device(local % max(1, torch.cuda.device_count()))
     return rank, dist.get_world_size()
 
 

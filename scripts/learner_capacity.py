@@ -36,6 +36,7 @@ def main():
     p.add_argument("--no-hip-kernels", action="store_true")
     p.add_argument("--num-actions", type=int, default=6)
     p.add_argument("--compute-dtype", default="fp32", choices=["fp32", "bf16"])
+    p.add_argument("--channels-last", action="store_true")
     args = p.parse_args()
 
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
@@ -49,6 +50,8 @@ def main():
                   compute_dtype=args.compute_dtype)
     torch.manual_seed(0)
     model = build_model(flags, obs, A).to(device)
+    if args.channels_last:
+        model = model.to(memory_format=torch.channels_last)
     model.force_eager_core = not flags.use_hip_kernels
     optimizer, scheduler = build_optimizer(flags, model)
 
@@ -81,7 +84,8 @@ def main():
     fps = args.steps * T * B / dt
     print(json.dumps(dict(
         diagnostic="learner_capacity", model=args.model, lstm=args.use_lstm,
-        hip=not args.no_hip_kernels, dtype=args.compute_dtype, T=T, B=B,
+        hip=not args.no_hip_kernels, dtype=args.compute_dtype,
+        channels_last=args.channels_last, T=T, B=B,
         ms_per_step=round(ms, 3), frames_per_sec=round(fps, 1))))
 
 

@@ -51,29 +51,21 @@ __device__ __forceinline__ float relu(float x) { return x > 0.f ? x : 0.f; }
 // ---------------------------------------------------------------------------
 // forward: grid = (N) blocks x 256 threads
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS, 2)
+__global__ void __launch_bounds__(NTHREADS, 4)
 shallow_encoder_fwd(const unsigned char* __restrict__ img,
                     const float* __restrict__ w1, const float* __restrict__ b1,
                     const float* __restrict__ w2, const float* __restrict__ b2,
                     const float* __restrict__ w3, const float* __restrict__ b3,
                     int N, float* __restrict__ a1g, float* __restrict__ a2g,
                     float* __restrict__ a3g, int save_for_bwd) {
-  __shared__ __align__(16) unsigned char s_img[IMG_BYTES];
   __shared__ __align__(16) float s_a1[A1_N];
   __shared__ __align__(16) float s_a2[A2_N];
 
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
-
-  // stage the u8 frame: 28224 B = 7056 dwords, coalesced
-  {
-    const unsigned int* src =
-        reinterpret_cast<const unsigned int*>(img + (size_t)n * IMG_BYTES);
-    unsigned int* dst = reinterpret_cast<unsigned int*>(s_img);
-    #pragma unroll 1
-    for (int i = tid; i < IMG_BYTES / 4; i += NTHREADS) dst[i] = src[i];
-  }
-  __syncthreads();
+  // the frame is read straight from L2 (28 KiB/image, PMC showed the
+  // LDS copy bought nothing and halved occupancy)
+  const unsigned char* s_img = img + (size_t)n * IMG_BYTES;
 
   // ---- conv1: thread = (filter f = tid/16, lane = tid%16); lane owns
   //      25 pixels (px = lane, lane+16, ...). Weights per (c,ky): 8
@@ -216,7 +208,7 @@ shallow_encoder_fwd(const unsigned char* __restrict__ img,
 // backward kernel 1: dW3 + db3 + d_a2 (relu-masked) -> d2g (N, A2_N)
 // LDS: a2 + d3 + d2 = 27 KiB
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS, 2)
+__global__ void __launch_bounds__(NTHREADS, 4)
 shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
                const float* __restrict__ d_a3g, const float* __restrict__ w3,
                int N, float* __restrict__ slab, float* __restrict__ d2g) {
@@ -300,7 +292,7 @@ shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
 // backward kernel 2: dW2 + d_a1 (relu-masked) -> d1g (N, A1_N) + db1
 // LDS: a1 + d2 + d1 = 61.6 KiB
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS, 2)
+__global__ void __launch_bounds__(NTHREADS, 4)
 shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
                const float* __restrict__ w2, int N,
                float* __restrict__ slab, float* __restrict__ d1g) {
@@ -388,24 +380,17 @@ shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
 // backward kernel 3: dW1 from the u8 frame (fused /255)
 // LDS: img + d1 = 53.6 KiB
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS, 2)
+__global__ void __launch_bounds__(NTHREADS, 4)
 shallow_bwd_k3(const unsigned char* __restrict__ img,
                const float* __restrict__ d1g, int N,
                float* __restrict__ slab) {
   const int esplit = gridDim.y, eslice = blockIdx.y;
-  __shared__ __align__(16) unsigned char s_img[IMG_BYTES];
   __shared__ __align__(16) float s_d1[A1_N];
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   float* my_slab = slab + (size_t)n * SLAB_N;
+  const unsigned char* s_img = img + (size_t)n * IMG_BYTES;
 
-  {
-    const unsigned int* src =
-        reinterpret_cast<const unsigned int*>(img + (size_t)n * IMG_BYTES);
-    unsigned int* dst = reinterpret_cast<unsigned int*>(s_img);
-    #pragma unroll 1
-    for (int i = tid; i < IMG_BYTES / 4; i += NTHREADS) dst[i] = src[i];
-  }
   #pragma unroll 1
   for (int i = tid; i < A1_N; i += NTHREADS)
     s_d1[i] = d1g[(size_t)n * A1_N + i];

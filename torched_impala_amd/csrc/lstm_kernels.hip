@@ -31,7 +31,8 @@ namespace impala {
 
 constexpr int H = 256;        // hidden size (IMPALA core is fixed at 256)
 constexpr int G4 = 4 * H;     // 1024 gate pre-activations
-constexpr int THREADS = 512;  // 8 waves
+constexpr int THREADS = 1024;  // 16 waves (4/SIMD: the GEMV is
+                               // L2-latency-bound; TLP covers it)
 
 // fast activations: libm tanhf lowers to an IEEE division
 // (v_div_scale/v_div_fmas/v_div_fixup + divergent special-case branches)
@@ -98,33 +99,24 @@ lstm_seq_fwd(const float* __restrict__ Xg, const float* __restrict__ W_hhT,
       if (save_for_bwd) Hp[rowH + tid] = hm;
     }
     __syncthreads();
-    // gate GEMV: thread j computes gate columns (2j, 2j+1).
-    // k is chunked by 4 (one ds_read_b128 broadcast of h) with the four
-    // W row loads batched, so ~16 loads stay in flight per thread —
-    // a scalar k-loop drains vmcnt every iteration (~200 L2 cycles x 256,
-    // measured 15 us/step; this form is latency-pipelined).
+    // gate GEMV: thread j computes gate column j. k chunked by 4 (one
+    // ds_read_b128 broadcast of h, four batched W row loads in flight).
     {
-      float acc0 = Xg[rowG + 2 * tid];
-      float acc1 = Xg[rowG + 2 * tid + 1];
-      const float2* wrow = reinterpret_cast<const float2*>(W_hhT) + tid;
+      float acc = Xg[rowG + tid];
+      const float* wcol = W_hhT + tid;
       #pragma unroll 4
       for (int k0 = 0; k0 < H; k0 += 4) {
         const float4 hv = *reinterpret_cast<const float4*>(&s_h[k0]);
-        const float2 w0 = wrow[(k0 + 0) * (G4 / 2)];
-        const float2 w1 = wrow[(k0 + 1) * (G4 / 2)];
-        const float2 w2 = wrow[(k0 + 2) * (G4 / 2)];
-        const float2 w3 = wrow[(k0 + 3) * (G4 / 2)];
-        acc0 = fmaf(hv.x, w0.x, acc0);
-        acc1 = fmaf(hv.x, w0.y, acc1);
-        acc0 = fmaf(hv.y, w1.x, acc0);
-        acc1 = fmaf(hv.y, w1.y, acc1);
-        acc0 = fmaf(hv.z, w2.x, acc0);
-        acc1 = fmaf(hv.z, w2.y, acc1);
-        acc0 = fmaf(hv.w, w3.x, acc0);
-        acc1 = fmaf(hv.w, w3.y, acc1);
+        const float w0 = wcol[(size_t)(k0 + 0) * G4];
+        const float w1 = wcol[(size_t)(k0 + 1) * G4];
+        const float w2 = wcol[(size_t)(k0 + 2) * G4];
+        const float w3 = wcol[(size_t)(k0 + 3) * G4];
+        acc = fmaf(hv.x, w0, acc);
+        acc = fmaf(hv.y, w1, acc);
+        acc = fmaf(hv.z, w2, acc);
+        acc = fmaf(hv.w, w3, acc);
       }
-      s_g[2 * tid] = acc0;
-      s_g[2 * tid + 1] = acc1;
+      s_g[tid] = acc;
     }
     __syncthreads();
     if (tid < H) {
@@ -208,16 +200,15 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
       s_dc[tid] = dct * gf * nd;  // dc_{t-1} (masked)
     }
     __syncthreads();
-    // dh_prev[k] = sum_j s_dg[j] * W_hh[j, k]; split j over two thread
-    // halves; j chunked by 4 with batched W row loads (same latency
-    // pipelining as the forward GEMV).
+    // dh_prev[k] = sum_j s_dg[j] * W_hh[j, k]; j split over four
+    // thread quarters, chunked by 4 with batched W row loads.
     {
       const int k = tid & (H - 1);
-      const int j0 = (tid < H) ? 0 : (G4 / 2);
+      const int j0 = (tid >> 8) * (G4 / 4);
       const float* wcol = W_hh + (size_t)j0 * H + k;
       float acc = 0.f;
       #pragma unroll 4
-      for (int j = 0; j < G4 / 2; j += 4) {
+      for (int j = 0; j < G4 / 4; j += 4) {
         const float4 dg = *reinterpret_cast<const float4*>(&s_dg[j0 + j]);
         const float w0 = wcol[(size_t)(j + 0) * H];
         const float w1 = wcol[(size_t)(j + 1) * H];
@@ -232,7 +223,8 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
     }
     __syncthreads();
     if (tid < H) {
-      s_dh[tid] = (s_part[tid] + s_part[H + tid]) * nd;
+      s_dh[tid] = (s_part[tid] + s_part[H + tid] + s_part[2 * H + tid] +
+                   s_part[3 * H + tid]) * nd;
     }
     __syncthreads();
   }

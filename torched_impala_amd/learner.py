@@ -109,12 +109,14 @@ class Batcher:
     def _pop_slots(self) -> List[int]:
         slots = []
         while len(slots) < self.k:
-            m = self.full_queue.pop_wait(timeout_ms=60000.0)
+            # generous: a cold 8-rank start spawns 256 actor processes
+            # that each import torch before the first slot lands
+            m = self.full_queue.pop_wait(timeout_ms=300000.0)
             if m == -2:
                 raise RuntimeError("transport aborted")
             if m < 0:
                 raise TimeoutError(
-                    "learner starved: no full rollout slot within 60 s "
+                    "learner starved: no full rollout slot within 300 s "
                     f"(got {len(slots)}/{self.k}) — actors dead or too slow")
             slots.append(m)
         return slots

@@ -296,3 +296,25 @@ def test_linear_mfma_plain_mode():
     from torched_impala_amd import ops
     C = ops.ext_hip().linear_mfma(A, B, None, None, 0)
     torch.testing.assert_close(C, A @ B.t(), rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.parametrize("clip_rho,clip_c,lam", [(0.8, 1.2, 0.95),
+                                                 (2.0, 0.7, 0.5)])
+def test_vtrace_scan_parity_nondefault_clips(clip_rho, clip_c, lam):
+    torch.manual_seed(5)
+    dev = _dev()
+    T, B = 20, 32
+    log_rhos = 0.6 * torch.randn(T, B, device=dev)
+    discounts = (torch.rand(T, B, device=dev) > 0.1).float() * 0.99
+    rewards = torch.randn(T, B, device=dev)
+    values = torch.randn(T, B, device=dev)
+    boot = torch.randn(B, device=dev)
+    vs, adv = ops.ext_hip().vtrace_scan(
+        log_rhos, discounts, rewards, values, boot, clip_rho, clip_c, lam)
+    ref = vtrace.from_importance_weights(
+        log_rhos.cpu(), discounts.cpu(), rewards.cpu(), values.cpu(),
+        boot.cpu(), clip_rho_threshold=clip_rho, clip_c_threshold=clip_c,
+        lambda_=lam, clip_pg_rho_threshold=clip_rho)
+    torch.testing.assert_close(vs.cpu(), ref.vs, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(adv.cpu(), ref.pg_advantages, rtol=1e-5,
+                               atol=1e-5)

@@ -99,3 +99,21 @@ def test_from_logits_consistency():
     lsm = torch.log_softmax(target, dim=-1)
     ref_lp = lsm.gather(2, actions.unsqueeze(-1)).squeeze(-1)
     torch.testing.assert_close(out.target_action_log_probs, ref_lp)
+
+
+@pytest.mark.parametrize("clip_rho,clip_c,lam", [(0.8, 1.2, 1.0),
+                                                 (1.5, 0.7, 0.95),
+                                                 (2.0, 2.0, 0.5)])
+def test_scan_matches_naive_nondefault_clips(clip_rho, clip_c, lam):
+    log_rhos, discounts, rewards, values, boot = _rand_inputs(seed=9)
+    vs_ref, adv_ref = naive_vtrace(log_rhos, discounts, rewards, values,
+                                   boot, clip_rho=clip_rho, clip_c=clip_c,
+                                   lam=lam)
+    out = vtrace.from_importance_weights(
+        torch.tensor(log_rhos), torch.tensor(discounts),
+        torch.tensor(rewards), torch.tensor(values), torch.tensor(boot),
+        clip_rho_threshold=clip_rho, clip_c_threshold=clip_c, lambda_=lam,
+        clip_pg_rho_threshold=clip_rho)
+    np.testing.assert_allclose(out.vs.numpy(), vs_ref, rtol=1e-6, atol=1e-6)
+    np.testing.assert_allclose(out.pg_advantages.numpy(), adv_ref,
+                               rtol=1e-6, atol=1e-6)

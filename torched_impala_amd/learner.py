@@ -11,7 +11,6 @@ MI355X specifics:
 """
 from __future__ import annotations
 
-import dataclasses
 import logging
 import os
 import time

@@ -15,7 +15,7 @@ per-cell sequence numbers) are stress-tested in tests/test_ringbuffer.py.
 from __future__ import annotations
 
 import logging
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 

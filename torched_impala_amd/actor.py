@@ -81,8 +81,20 @@ class Rollout:
             self.episode_step[finished] = 0
         self.core_state = new_core
         self.last_action = action
-        # step all envs with the chosen actions (auto-reset on done)
-        obs_list, rew, done = [], [], []
+        # step all envs with the chosen actions (auto-reset on done);
+        # preallocated numpy buffers keep this loop allocation-free.
+        # The obs buffer is DOUBLE-buffered: self.carry["frame"] aliases
+        # the buffer the previous step wrote, which must stay intact
+        # until fill_slot has copied it out.
+        if not hasattr(self, "_np_obs"):
+            shape = (self.n, *self.envs[0].obs_shape)
+            dt = self.envs[0].obs_dtype
+            self._np_obs = [np.empty(shape, dtype=dt), np.empty(shape, dtype=dt)]
+            self._obs_flip = 0
+        self._obs_flip ^= 1
+        obs_buf = self._np_obs[self._obs_flip]
+        rew = np.empty(self.n, dtype=np.float32)
+        done = np.empty(self.n, dtype=bool)
         act = action.tolist()
         for i, env in enumerate(self.envs):
             o, r, d, _ = env.step(act[i])
@@ -90,12 +102,12 @@ class Rollout:
             self.episode_step[i] += 1
             if d:
                 o = env.reset()
-            obs_list.append(o)
-            rew.append(r)
-            done.append(d)
-        self.frame = torch.from_numpy(np.stack(obs_list))
-        self.reward = torch.tensor(rew, dtype=torch.float32)
-        self.done = torch.tensor(done, dtype=torch.bool)
+            obs_buf[i] = o
+            rew[i] = r
+            done[i] = d
+        self.frame = torch.from_numpy(obs_buf)
+        self.reward = torch.from_numpy(rew)
+        self.done = torch.from_numpy(done)
 
     def fill_slot(self, slot, T: int):
         """Write rows 0..T (shape (T+1, n, ...)) of one slot."""

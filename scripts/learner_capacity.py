@@ -82,11 +82,14 @@ def main():
     dt = time.perf_counter() - t0
     ms = dt / args.steps * 1000
     fps = args.steps * T * B / dt
+    mem_gb = (torch.cuda.max_memory_allocated() / 2**30
+              if device.type == "cuda" else 0.0)
     print(json.dumps(dict(
         diagnostic="learner_capacity", model=args.model, lstm=args.use_lstm,
         hip=not args.no_hip_kernels, dtype=args.compute_dtype,
         channels_last=args.channels_last, T=T, B=B,
-        ms_per_step=round(ms, 3), frames_per_sec=round(fps, 1))))
+        ms_per_step=round(ms, 3), frames_per_sec=round(fps, 1),
+        peak_hbm_gib=round(mem_gb, 2))))
 
 
 if __name__ == "__main__":

@@ -479,6 +479,10 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
                 metrics.write(step=step_frames, fps=fps,
                               mean_return=mr,
                               timers=timers.drain(),
+                              peak_hbm_gib=(round(
+                                  torch.cuda.max_memory_allocated() / 2**30, 2)
+                                  if device.type == "cuda" else 0.0),
+                              queue_full_depth=full_queue.size_approx(),
                               **{k: float(v) for k, v in stats.items()})
                 last_log, last_log_frames = now, step_frames
 

@@ -22,3 +22,42 @@ def test_mirror_roundtrip():
         assert torch.equal(a, b)
     # no new version -> no copy needed
     assert mirror.pull_into(dst, last_seen=v2) == v2
+
+
+def _mirror_writer(mirror, n_pub):
+    src = MLPPolicy((4,), 2)
+    for k in range(1, n_pub + 1):
+        with torch.no_grad():
+            for p in src.parameters():
+                p.fill_(float(k))
+        mirror.publish(src)
+
+
+def test_mirror_no_torn_weights_under_fast_writer():
+    """Writer publishes constant-stamped models as fast as it can; every
+    read must be internally consistent (all params one stamp)."""
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    src = MLPPolicy((4,), 2)
+    with torch.no_grad():
+        for p in src.parameters():
+            p.fill_(0.5)
+    mirror = WeightMirror(src)
+    dst = MLPPolicy((4,), 2)
+    n_pub = 400
+    w = ctx.Process(target=_mirror_writer, args=(mirror, n_pub))
+    w.start()
+    seen = set()
+    last = -1
+    try:
+        while w.is_alive() or int(mirror.version.item()) != last:
+            last = mirror.pull_into(dst, last_seen=last)
+            vals = torch.cat([p.flatten() for p in dst.parameters()])
+            uniq = torch.unique(vals)
+            assert uniq.numel() == 1, f"torn weights: {uniq.tolist()[:5]}"
+            seen.add(float(uniq[0]))
+            if float(uniq[0]) >= n_pub:
+                break
+    finally:
+        w.join(timeout=30)
+    assert len(seen) > 3, "reader never observed multiple versions"

@@ -85,6 +85,7 @@ class FlatRMSprop(torch.optim.Optimizer):
         # keep LRScheduler's step-order bookkeeping happy in graph mode,
         # where optimizer.step() itself is inside the captured graph
         self._step_count = getattr(self, "_step_count", 0) + 1
+        self._opt_called = True  # torch>=2.4 scheduler-order check
 
     @torch.no_grad()
     def step_device(self):

@@ -63,6 +63,9 @@ class WeightMirror:
                 for k, s in zip(self.keys, src):
                     sd[k].copy_(s)
             v2 = int(self.version.item())
-            if v2 == v or (v2 % 2) == (v % 2):
+            # version v's buffer stays intact only until the writer
+            # publishes v+2 (which rewrites buffer v%2): exactly one
+            # newer version is tolerable, two means our copy may be torn
+            if v2 <= v + 1:
                 return v
-            v = v2  # writer moved onto our buffer mid-copy; retry
+            v = v2  # writer lapped us; retry from the newest version

@@ -70,3 +70,26 @@ def test_lstm_initial_state_replay():
         out, _ = model(inp, core)
     torch.testing.assert_close(out["policy_logits"], s["behavior_logits"],
                                rtol=1e-5, atol=1e-6)
+
+
+def test_batcher_time_major_layout():
+    """_to_time_major: (k, T+1, n, ...) slot-major -> (T+1, k*n, ...)."""
+    from torched_impala_amd.learner import Batcher
+
+    class _Stub:
+        pass
+    flags = Flags(batch_size=6, envs_per_actor=3)
+    stub = _Stub()
+    stub.flags = flags
+    stub.k = 2
+    st = dict(
+        frame=torch.arange(2 * 5 * 3 * 4).reshape(2, 5, 3, 4),
+        initial_core_state=torch.arange(2 * 2 * 3 * 8).reshape(2, 2, 3, 8),
+    )
+    out = Batcher._to_time_major(stub, st)
+    assert out["frame"].shape == (5, 6, 4)
+    assert out["initial_core_state"].shape == (2, 6, 8)
+    # element check: slot k=1, t=2, env n=0 -> batch column 1*3+0=3
+    assert torch.equal(out["frame"][2, 3], st["frame"][1, 2, 0])
+    assert torch.equal(out["initial_core_state"][1, 4],
+                       st["initial_core_state"][1, 1, 1])

@@ -33,9 +33,6 @@ from torched_impala_amd.transport import SlotQueue, TrajectoryBuffers, WeightMir
 
 log = logging.getLogger(__name__)
 
-BATCH_KEYS = ("frame", "reward", "done", "action", "behavior_logits",
-              "last_action", "episode_return", "episode_step")
-
 
 def _np_to_torch_dtype(np_dtype):
     import numpy as np

@@ -16,8 +16,14 @@ import torch
 import torch.nn.functional as F
 
 
+def _fp(x: torch.Tensor) -> torch.Tensor:
+    """Upcast low-precision logits to fp32; keep fp64 as-is (tests use
+    fp64 for finite-difference checks)."""
+    return x if x.dtype == torch.float64 else x.float()
+
+
 def pg_loss(target_logits, actions, advantages):
-    logp = F.log_softmax(target_logits.flatten(0, 1).float(), dim=-1)
+    logp = F.log_softmax(_fp(target_logits.flatten(0, 1)), dim=-1)
     lp_a = logp.gather(
         1, actions.flatten(0, 1).long().unsqueeze(1)).squeeze(1)
     return -(lp_a * advantages.flatten(0, 1).detach()).sum()
@@ -29,7 +35,7 @@ def baseline_loss(vs, values):
 
 def entropy_loss(target_logits):
     """Negative entropy summed over (T, B) — added with positive coef."""
-    logits = target_logits.flatten(0, 1).float()
+    logits = _fp(target_logits.flatten(0, 1))
     logp = F.log_softmax(logits, dim=-1)
     p = F.softmax(logits, dim=-1)
     return (p * logp).sum()

@@ -122,3 +122,27 @@ def test_buffers_nbytes_and_shapes():
     assert b.arenas["frame"].shape == (64, 21, 1, 4, 84, 84)
     assert b.arenas["initial_core_state"].shape == (64, 2, 1, 256)
     assert b.nbytes() > 64 * 21 * 4 * 84 * 84
+
+
+@pytest.mark.timeout(600)
+def test_ring_core_tsan():
+    """the C++ queue core under ThreadSanitizer (4 producers
+    x 4 consumers; exactly-once delivery). Skipped if g++/TSAN missing."""
+    import shutil
+    import subprocess
+    import sys
+    import tempfile
+    if shutil.which("g++") is None:
+        pytest.skip("no g++")
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    src = os.path.join(root, "tests", "native", "tsan_ring.cpp")
+    with tempfile.TemporaryDirectory() as td:
+        exe = os.path.join(td, "tsan_ring")
+        b = subprocess.run(["g++", "-O1", "-g", "-std=c++17",
+                            "-fsanitize=thread", "-pthread", src, "-o", exe],
+                           capture_output=True, text=True)
+        if b.returncode != 0:
+            pytest.skip(f"TSAN unavailable: {b.stderr[-200:]}")
+        r = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, f"tsan_ring failed:\n{r.stdout}\n{r.stderr}"
+        assert "OK" in r.stdout

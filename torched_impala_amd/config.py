@@ -108,6 +108,11 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--discounting", type=float, default=f.discounting)
     p.add_argument("--reward-clipping", default=f.reward_clipping,
                    choices=["abs_one", "none"])
+    p.add_argument("--clip-rho-threshold", type=float,
+                   default=f.clip_rho_threshold)
+    p.add_argument("--clip-c-threshold", type=float,
+                   default=f.clip_c_threshold)
+    p.add_argument("--lambda", dest="lambda_", type=float, default=f.lambda_)
     p.add_argument("--learning-rate", type=float, default=f.learning_rate)
     p.add_argument("--alpha", type=float, default=f.alpha)
     p.add_argument("--momentum", type=float, default=f.momentum)

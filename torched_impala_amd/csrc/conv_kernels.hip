@@ -272,7 +272,7 @@ shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
     const int ky_lo = max(0, y - (H3 - 1)), ky_hi = min(K3 - 1, y);
     const int kx_lo = max(0, x - (W3 - 1)), kx_hi = min(K3 - 1, x);
     float acc = 0.f;
-    #pragma unroll 1
+    #pragma unroll 4
     for (int f = 0; f < F3; ++f) {
       const float* dc = s_d3 + f * (H3 * W3);
       const float* wc = w3 + (f * F2 + c) * (K3 * K3);
@@ -353,7 +353,7 @@ shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
     const int y = p / W1, x = p - y * W1;
     const int ky0 = (y & 1), kx0 = (x & 1);
     float acc = 0.f;
-    #pragma unroll 1
+    #pragma unroll 4
     for (int f = 0; f < F2; ++f) {
       const float* dc = s_d2 + f * (H2 * W2);
       const float* wc = w2 + (f * F1 + c) * (K2 * K2);

@@ -257,17 +257,26 @@ class GraphedStep:
     tracing compiler"). lr reaches the captured optimizer through a
     device scalar (FlatRMSprop.sync_lr runs outside the graph).
 
-    Preconditions (asserted by the caller): CUDA, world_size == 1 (DP
-    capture of RCCL is a later round), FlatRMSprop, HIP kernels on.
+    Data parallelism (world > 1): the RCCL collective stays OUTSIDE the
+    graph — the captured region is forward+loss+backward only (grads
+    land in the flat arena), then ONE flat all-reduce of flat_grad and
+    the eager fused optimizer step run per replay. This trades the
+    backward-overlapped bucketing of FlatGradReducer for graph-level
+    launch elimination (the grads are a few MB — latency-dominated
+    either way on xGMI).
+
+    Preconditions (asserted by the caller): CUDA, FlatRMSprop, HIP
+    kernels on.
     """
 
     WARMUP_ITERS = 2
 
-    def __init__(self, flags, model, optimizer, batcher):
+    def __init__(self, flags, model, optimizer, batcher, world_size: int = 1):
         self.flags = flags
         self.model = model
         self.optimizer = optimizer
         self.batcher = batcher
+        self.world = world_size
         self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self.outputs: Dict[int, tuple] = {}
         self.pool = None
@@ -275,7 +284,8 @@ class GraphedStep:
     def _fn(self, i: int):
         batch = self.batcher._to_time_major(self.batcher.staging[i])
         loss, stats = _fwd_bwd(self.flags, self.model, batch, self.optimizer)
-        self.optimizer.step_device()
+        if self.world == 1:
+            self.optimizer.step_device()  # world>1: runs eagerly after
         return (loss.detach(), stats["pg_loss"], stats["baseline_loss"],
                 stats["entropy_loss"])
 
@@ -301,6 +311,11 @@ class GraphedStep:
         if i not in self.graphs:
             self._capture(i)
         self.graphs[i].replay()
+        if self.world > 1:
+            # SUM; the 1/world mean is the optimizer's device-side scale
+            torch.distributed.all_reduce(self.optimizer.flat_grad,
+                                         op=torch.distributed.ReduceOp.SUM)
+            self.optimizer.step_device()
         scheduler.step()
         loss, pg, bl, ent = self.outputs[i]
         return dict(total_loss=loss, pg_loss=pg, baseline_loss=bl,
@@ -374,17 +389,22 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     procs = start_actors(*actor_args)
 
     optimizer, scheduler = build_optimizer(flags, model, world)
-    if world > 1:
+    batcher = Batcher(flags, buffers, free_queue, full_queue, device)
+    graphed = None
+    reducer = None
+    use_graphs = (flags.graph_capture and device.type == "cuda"
+                  and flags.use_hip_kernels
+                  and isinstance(optimizer, FlatRMSprop))
+    if use_graphs:
+        # world>1: the collective runs OUTSIDE the graph (GraphedStep) —
+        # no hook-based reducer (its all_reduce calls must not be
+        # captured inside backward)
+        graphed = GraphedStep(flags, model, optimizer, batcher,
+                              world_size=world)
+    elif world > 1:
         reducer = (FlatGradReducer(model, optimizer)
                    if isinstance(optimizer, FlatRMSprop)
                    else GradReducer(model))
-    else:
-        reducer = None
-    batcher = Batcher(flags, buffers, free_queue, full_queue, device)
-    graphed = None
-    if (flags.graph_capture and device.type == "cuda" and world == 1
-            and flags.use_hip_kernels and isinstance(optimizer, FlatRMSprop)):
-        graphed = GraphedStep(flags, model, optimizer, batcher)
     metrics = MetricsLogger(flags.metrics_jsonl)
     timers = StageTimers()
 
@@ -425,6 +445,8 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
                     log.exception("graph capture failed; falling back to "
                                   "eager learn steps")
                     graphed = None
+                    if world > 1 and reducer is None:
+                        reducer = FlatGradReducer(model, optimizer)
                     batch.update(batcher._to_time_major(batcher.staging[si]))
                     stats = learn_step(flags, model, batch, optimizer,
                                        scheduler, reducer, timers)

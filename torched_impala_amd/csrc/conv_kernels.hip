@@ -292,20 +292,20 @@ shallow_bwd_k1(const float* __restrict__ a2g, const float* __restrict__ a3g,
 // backward kernel 2: dW2 + d_a1 (relu-masked) -> d1g (N, A1_N) + db1
 // LDS: a1 + d2 + d1 = 61.6 KiB
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS, 4)
+__global__ void __launch_bounds__(NTHREADS, 8)
 shallow_bwd_k2(const float* __restrict__ a1g, const float* __restrict__ d2g,
                const float* __restrict__ w2, int N,
                float* __restrict__ slab, float* __restrict__ d1g) {
   const int esplit = gridDim.y, eslice = blockIdx.y;
-  __shared__ __align__(16) float s_a1[A1_N];
+  // a1 reads come straight from L2 (25.6 KiB/image, re-read ~8x, the
+  // kernel is latency- not bandwidth-bound): dropping its LDS copy cuts
+  // LDS to 10.4 KiB -> 8 blocks/CU, doubling latency coverage
   __shared__ __align__(16) float s_d2[A2_N];
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   float* my_slab = slab + (size_t)n * SLAB_N;
+  const float* s_a1 = a1g + (size_t)n * A1_N;
 
-  #pragma unroll 1
-  for (int i = tid; i < A1_N; i += NTHREADS)
-    s_a1[i] = a1g[(size_t)n * A1_N + i];
   #pragma unroll 1
   for (int i = tid; i < A2_N; i += NTHREADS)
     s_d2[i] = d2g[(size_t)n * A2_N + i];

@@ -1,0 +1,49 @@
+"""bench.py is the driver contract — verify the JSON line schema and the
+CPU execution path end-to-end (subprocess, tiny config)."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.timeout(420)
+def test_bench_cpu_json_schema():
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"), "--device", "cpu",
+         "--steps", "4", "--warmup", "2", "--num-actors", "2",
+         "--batch-size", "4", "--unroll-length", "8",
+         "--envs-per-actor", "2"],
+        capture_output=True, text=True, cwd=ROOT, timeout=360)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.strip().splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, f"missing contract key {key}"
+    assert d["metric"].startswith("env frames/sec (learner)")
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["data"] == "synthetic"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    # value is the whole-job aggregate: steps*T*B*n_gpus / elapsed
+    assert abs(d["value"] * d["ms_per_step"] / 1000.0
+               - 8 * 4 * d["n_gpus"]) < 1e-6
+    cfg = d["config"]
+    for key in ("model", "global_batch", "seq_len", "parallelism"):
+        assert key in cfg
+
+
+@pytest.mark.timeout(420)
+def test_cli_train_entry():
+    r = subprocess.run(
+        [sys.executable, "-m", "torched_impala_amd.cli", "--env", "cartpole",
+         "--device", "cpu", "--num-actors", "1", "--batch-size", "2",
+         "--unroll-length", "5", "--total-steps", "40",
+         "--disable-checkpoint"],
+        capture_output=True, text=True, cwd=ROOT, timeout=360)
+    assert r.returncode == 0, r.stderr[-2000:]

@@ -28,6 +28,61 @@ ROW_KEYS = ("frame", "reward", "done", "action", "behavior_logits",
             "last_action", "episode_return", "episode_step")
 
 
+class _FlatActorFwd(torch.nn.Module):
+    """Feedforward-policy actor forward as one traceable graph: encode ->
+    core-input concat -> policy head -> sample. Skips the baseline head
+    (actors never use values) and all module-dispatch glue."""
+
+    def __init__(self, net):
+        super().__init__()
+        self.net = net
+
+    def forward(self, frame, reward, last_action):
+        import torch.nn.functional as F
+        x = self.net.encode(frame)
+        r = torch.clamp(reward, -1.0, 1.0).unsqueeze(-1)
+        oh = F.one_hot(last_action, self.net.num_actions).float()
+        ci = torch.cat([x, r, oh], dim=-1)
+        logits = self.net.policy(ci)
+        probs = F.softmax(logits.float(), dim=-1)
+        action = torch.multinomial(probs, num_samples=1).squeeze(-1)
+        return logits.float(), action
+
+
+class FastActorPolicy:
+    """torch.jit fast path for feedforward (non-LSTM) actor policies.
+
+    mode "safe":   trace + freeze with parameters PRESERVED — weight
+                   pulls propagate (the params are the same tensors);
+                   ~1.15x the eager flat forward.
+    mode "frozen": full freeze (weights folded as constants, conv+relu
+                   fusion etc.; ~1.6x) — caller must refresh() after
+                   weight pulls; each refresh costs a few ms + ~50 ms of
+                   profiling warmup, so refreshes are rate-limited by
+                   flags.actor_jit_refresh claims.
+    """
+
+    def __init__(self, model, example, mode: str = "safe"):
+        self.mode = mode
+        self._wrapper = _FlatActorFwd(model).eval()
+        self._traced = torch.jit.trace(self._wrapper, example,
+                                       check_trace=False)
+        self._frozen = None
+        self.refresh()
+
+    def refresh(self):
+        if self.mode == "frozen":
+            self._frozen = torch.jit.freeze(self._traced)
+        elif self._frozen is None:
+            preserved = ["net." + k for k, _ in
+                         self._wrapper.net.named_parameters()]
+            self._frozen = torch.jit.freeze(self._traced,
+                                            preserved_attrs=preserved)
+
+    def __call__(self, frame, reward, last_action):
+        return self._frozen(frame, reward, last_action)
+
+
 class Rollout:
     """Rollout state over n env copies driven by one policy forward."""
 
@@ -47,6 +102,16 @@ class Rollout:
         self.episode_step = torch.zeros(n, dtype=torch.int32)
         self.core_before = tuple(t.clone() for t in self.core_state)
         self.carry = None
+        self.fast = None
+        jit_mode = getattr(flags, "actor_jit", "safe")
+        if (jit_mode != "off" and not getattr(model, "use_lstm", False)
+                and hasattr(model, "encode")):
+            try:
+                example = (self.frame, self.reward, self.last_action)
+                self.fast = FastActorPolicy(model, example, mode=jit_mode)
+            except Exception:
+                log.exception("actor jit fast path unavailable; using eager")
+                self.fast = None
         self._advance()  # produce the first carried row
 
     @torch.inference_mode()
@@ -60,10 +125,17 @@ class Rollout:
         out, new_core = self.model(inputs, self.core_state)
         return out, new_core
 
+    @torch.inference_mode()
+    def _fast_forward(self):
+        logits, action = self.fast(self.frame, self.reward, self.last_action)
+        return {"policy_logits": logits.unsqueeze(0),
+                "action": action.unsqueeze(0)}, ()
+
     def _advance(self):
         """One timestep: forward -> record row -> step envs.
         Sets self.carry to the recorded row."""
-        out, new_core = self._forward()
+        out, new_core = (self._fast_forward() if self.fast is not None
+                         else self._forward())
         action = out["action"][0]
         self.carry = dict(
             frame=self.frame, reward=self.reward.clone(),
@@ -140,11 +212,19 @@ def act(flags, actor_index: int, free_queue, full_queue, buffers, mirror,
         version = mirror.pull_into(model)
         roll = Rollout(flags, actor_index, model)
         T = flags.unroll_length
+        refresh_every = max(1, getattr(flags, "actor_jit_refresh", 4))
+        claims = 0
         while True:
             m = free_queue.pop_wait()
             if m < 0:  # abort/sentinel
                 break
-            version = mirror.pull_into(model, last_seen=version)
+            new_version = mirror.pull_into(model, last_seen=version)
+            if (roll.fast is not None and roll.fast.mode == "frozen"
+                    and new_version != version
+                    and claims % refresh_every == 0):
+                roll.fast.refresh()  # re-bake the folded weights
+            version = new_version
+            claims += 1
             roll.fill_slot(buffers.slot(m), T)
             if not full_queue.push(m):
                 log.error("actor %d: full queue overflow", actor_index)

@@ -33,6 +33,11 @@ class Flags:
     envs_per_actor: int = 1            # batched envs inside one actor process
     actor_threads: int = 0             # CPU threads per actor; 0 = auto
                                        # (cores/num_actors, capped at 8)
+    actor_jit: str = "frozen"          # off | safe (preserved-freeze) |
+                                       # frozen (weights folded; fastest,
+                                       # refreshed every actor_jit_refresh
+                                       # slot claims)
+    actor_jit_refresh: int = 4
 
     # --- model ---
     use_lstm: bool = False
@@ -100,6 +105,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--num-buffers", type=int, default=None)
     p.add_argument("--envs-per-actor", type=int, default=f.envs_per_actor)
     p.add_argument("--actor-threads", type=int, default=f.actor_threads)
+    p.add_argument("--actor-jit", default=f.actor_jit,
+                   choices=["off", "safe", "frozen"])
+    p.add_argument("--actor-jit-refresh", type=int, default=f.actor_jit_refresh)
     _add_bool(p, "use-lstm", f.use_lstm)
     p.add_argument("--model", default=f.model,
                    choices=["auto", "mlp", "shallow", "deep"])

@@ -49,6 +49,28 @@ class _FlatActorFwd(torch.nn.Module):
         return logits.float(), action
 
 
+class _FlatActorLstmFwd(torch.nn.Module):
+    """Same, for LSTM-core policies: one traced step carrying (h, c) with
+    the done-mask reset inlined."""
+
+    def __init__(self, net):
+        super().__init__()
+        self.net = net
+
+    def forward(self, frame, reward, done, last_action, h, c):
+        import torch.nn.functional as F
+        x = self.net.encode(frame)
+        r = torch.clamp(reward, -1.0, 1.0).unsqueeze(-1)
+        oh = F.one_hot(last_action, self.net.num_actions).float()
+        ci = torch.cat([x, r, oh], dim=-1)
+        nd = (~done).float().unsqueeze(-1)
+        h2, c2 = self.net.core(ci, (h * nd, c * nd))
+        logits = self.net.policy(h2)
+        probs = F.softmax(logits.float(), dim=-1)
+        action = torch.multinomial(probs, num_samples=1).squeeze(-1)
+        return logits.float(), action, h2, c2
+
+
 class FastActorPolicy:
     """torch.jit fast path for feedforward (non-LSTM) actor policies.
 
@@ -62,9 +84,12 @@ class FastActorPolicy:
                    flags.actor_jit_refresh claims.
     """
 
-    def __init__(self, model, example, mode: str = "safe"):
+    def __init__(self, model, example, mode: str = "safe",
+                 use_lstm: bool = False):
         self.mode = mode
-        self._wrapper = _FlatActorFwd(model).eval()
+        self.use_lstm = use_lstm
+        wrapper_cls = _FlatActorLstmFwd if use_lstm else _FlatActorFwd
+        self._wrapper = wrapper_cls(model).eval()
         self._traced = torch.jit.trace(self._wrapper, example,
                                        check_trace=False)
         self._frozen = None
@@ -79,8 +104,8 @@ class FastActorPolicy:
             self._frozen = torch.jit.freeze(self._traced,
                                             preserved_attrs=preserved)
 
-    def __call__(self, frame, reward, last_action):
-        return self._frozen(frame, reward, last_action)
+    def __call__(self, *args):
+        return self._frozen(*args)
 
 
 class Rollout:
@@ -104,11 +129,16 @@ class Rollout:
         self.carry = None
         self.fast = None
         jit_mode = getattr(flags, "actor_jit", "safe")
-        if (jit_mode != "off" and not getattr(model, "use_lstm", False)
-                and hasattr(model, "encode")):
+        use_lstm = getattr(model, "use_lstm", False)
+        if jit_mode != "off" and hasattr(model, "encode"):
             try:
-                example = (self.frame, self.reward, self.last_action)
-                self.fast = FastActorPolicy(model, example, mode=jit_mode)
+                if use_lstm:
+                    example = (self.frame, self.reward, self.done,
+                               self.last_action, *self.core_state)
+                else:
+                    example = (self.frame, self.reward, self.last_action)
+                self.fast = FastActorPolicy(model, example, mode=jit_mode,
+                                            use_lstm=use_lstm)
             except Exception:
                 log.exception("actor jit fast path unavailable; using eager")
                 self.fast = None
@@ -127,6 +157,12 @@ class Rollout:
 
     @torch.inference_mode()
     def _fast_forward(self):
+        if self.fast.use_lstm:
+            logits, action, h, c = self.fast(
+                self.frame, self.reward, self.done, self.last_action,
+                *self.core_state)
+            return {"policy_logits": logits.unsqueeze(0),
+                    "action": action.unsqueeze(0)}, (h, c)
         logits, action = self.fast(self.frame, self.reward, self.last_action)
         return {"policy_logits": logits.unsqueeze(0),
                 "action": action.unsqueeze(0)}, ()

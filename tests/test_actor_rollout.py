@@ -9,10 +9,10 @@ from torched_impala_amd.actor import Rollout
 from torched_impala_amd.transport import TrajectoryBuffers
 
 
-def _setup(n=4, use_lstm=False, seed=0):
+def _setup(n=4, use_lstm=False, seed=0, actor_jit="frozen"):
     torch.manual_seed(seed)
     flags = Flags(env="synthetic-atari", envs_per_actor=n,
-                  use_lstm=use_lstm, seed=seed)
+                  use_lstm=use_lstm, seed=seed, actor_jit=actor_jit)
     model = build_model(flags, (4, 84, 84), 6)
     model.eval()
     return flags, model, Rollout(flags, 0, model)
@@ -33,10 +33,15 @@ def test_slot_overlap_row_semantics():
         assert torch.equal(s0[k][T], s1[k][0]), f"overlap mismatch for {k}"
 
 
-def test_behavior_logits_match_replay():
+import pytest
+
+
+@pytest.mark.parametrize("jit", ["off", "safe", "frozen", "native"])
+def test_behavior_logits_match_replay(jit):
     """Stored behavior logits == re-running the policy on the stored row
-    (catches torn/aliased frames between the forward and the slot copy)."""
-    flags, model, roll = _setup(seed=3)
+    (catches torn/aliased frames between the forward and the slot copy;
+    parametrized over every actor fast path incl. the native C++ one)."""
+    flags, model, roll = _setup(seed=3, actor_jit=jit)
     T = 5
     bufs = TrajectoryBuffers(T, (4, 84, 84), torch.uint8, 6, 2,
                              envs_per_slot=4)

@@ -71,6 +71,40 @@ class _FlatActorLstmFwd(torch.nn.Module):
         return logits.float(), action, h2, c2
 
 
+class NativeActorPolicy:
+    """AVX2 C++ actor inference (csrc/actor_net.cpp) for the shallow net:
+    zero torch dispatch, bit-level parity ~3e-8, cheap weight repack on
+    refresh. Shallow-net (4,84,84) feedforward policies only."""
+
+    mode = "native"
+    use_lstm = False
+
+    def __init__(self, model, num_actions: int, seed: int):
+        import torched_impala_amd._impala_actor as ext
+        assert model.obs_shape == (4, 84, 84) and not model.use_lstm
+        self._ext = ext.ShallowActorNet(seed)
+        self._model = model
+        self._A = num_actions
+        self.refresh()
+
+    def refresh(self):
+        m = self._model
+        self._ext.load_weights(
+            m.conv1.weight.detach(), m.conv1.bias.detach(),
+            m.conv2.weight.detach(), m.conv2.bias.detach(),
+            m.conv3.weight.detach(), m.conv3.bias.detach(),
+            m.fc.weight.detach(), m.fc.bias.detach(),
+            m.policy.weight.detach(), m.policy.bias.detach())
+
+    def __call__(self, frame, reward, last_action):
+        n = frame.shape[0]
+        logits = torch.empty(n, self._A)
+        action = torch.empty(n, dtype=torch.int64)
+        self._ext.forward(frame.contiguous(), reward.contiguous(),
+                          last_action.contiguous(), logits, action)
+        return logits, action
+
+
 class FastActorPolicy:
     """torch.jit fast path for feedforward (non-LSTM) actor policies.
 
@@ -132,13 +166,29 @@ class Rollout:
         use_lstm = getattr(model, "use_lstm", False)
         if jit_mode != "off" and hasattr(model, "encode"):
             try:
-                if use_lstm:
-                    example = (self.frame, self.reward, self.done,
-                               self.last_action, *self.core_state)
+                if (jit_mode == "native" and not use_lstm
+                        and getattr(model, "obs_shape", None) == (4, 84, 84)):
+                    self.fast = NativeActorPolicy(
+                        model, model.num_actions,
+                        seed=flags.seed + 77777 * actor_index)
+                elif jit_mode == "native":
+                    # native covers the shallow feedforward net only;
+                    # other policies take the traced-frozen path
+                    example = ((self.frame, self.reward, self.done,
+                                self.last_action, *self.core_state)
+                               if use_lstm else
+                               (self.frame, self.reward, self.last_action))
+                    self.fast = FastActorPolicy(model, example,
+                                                mode="frozen",
+                                                use_lstm=use_lstm)
                 else:
-                    example = (self.frame, self.reward, self.last_action)
-                self.fast = FastActorPolicy(model, example, mode=jit_mode,
-                                            use_lstm=use_lstm)
+                    example = ((self.frame, self.reward, self.done,
+                                self.last_action, *self.core_state)
+                               if use_lstm else
+                               (self.frame, self.reward, self.last_action))
+                    self.fast = FastActorPolicy(model, example,
+                                                mode=jit_mode,
+                                                use_lstm=use_lstm)
             except Exception:
                 log.exception("actor jit fast path unavailable; using eager")
                 self.fast = None
@@ -255,10 +305,11 @@ def act(flags, actor_index: int, free_queue, full_queue, buffers, mirror,
             if m < 0:  # abort/sentinel
                 break
             new_version = mirror.pull_into(model, last_seen=version)
-            if (roll.fast is not None and roll.fast.mode == "frozen"
+            if (roll.fast is not None
+                    and roll.fast.mode in ("frozen", "native")
                     and new_version != version
                     and claims % refresh_every == 0):
-                roll.fast.refresh()  # re-bake the folded weights
+                roll.fast.refresh()  # re-bake/re-pack the weights
             version = new_version
             claims += 1
             roll.fill_slot(buffers.slot(m), T)

@@ -56,10 +56,22 @@ def _run(cmd):
 
 def build_ring(force: bool = False) -> str:
     out = os.path.join(PKG_DIR, "_impala_ring.so")
-    sources = [os.path.join(CSRC, "ring.cpp")]
+    sources = [os.path.join(CSRC, "ring.cpp"),
+               os.path.join(CSRC, "ring_core.h")]
     if force or _needs_build(out, sources):
         flags, link, _ = _common_flags("_impala_ring")
-        _run(["g++"] + flags + sources + link + ["-o", out])
+        _run(["g++"] + flags + [sources[0]] + link + ["-o", out])
+    return out
+
+
+def build_actor(force: bool = False) -> str:
+    """Native CPU actor inference: -mavx2 -mfma (EPYC-safe baseline)."""
+    out = os.path.join(PKG_DIR, "_impala_actor.so")
+    sources = [os.path.join(CSRC, "actor_net.cpp")]
+    if force or _needs_build(out, sources):
+        flags, link, _ = _common_flags("_impala_actor")
+        _run(["g++", "-mavx2", "-mfma", "-funroll-loops"] + flags + sources
+             + link + ["-o", out])
     return out
 
 
@@ -89,6 +101,7 @@ def build_hip(force: bool = False) -> str:
 
 def build_all(force: bool = False):
     build_ring(force)
+    build_actor(force)
     try:
         build_hip(force)
     except FileNotFoundError:

@@ -106,7 +106,7 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--envs-per-actor", type=int, default=f.envs_per_actor)
     p.add_argument("--actor-threads", type=int, default=f.actor_threads)
     p.add_argument("--actor-jit", default=f.actor_jit,
-                   choices=["off", "safe", "frozen"])
+                   choices=["off", "safe", "frozen", "native"])
     p.add_argument("--actor-jit-refresh", type=int, default=f.actor_jit_refresh)
     _add_bool(p, "use-lstm", f.use_lstm)
     p.add_argument("--model", default=f.model,

@@ -23,7 +23,7 @@ def _worker(args, idx, q):
     from torched_impala_amd.models import build_model
     from torched_impala_amd.actor import Rollout
     flags = Flags(env="synthetic-atari", envs_per_actor=args.envs,
-                  use_lstm=args.use_lstm, seed=idx)
+                  use_lstm=args.use_lstm, seed=idx, actor_jit=args.jit)
     model = build_model(flags, (4, 84, 84), 6)
     model.eval()
     roll = Rollout(flags, idx, model)
@@ -44,6 +44,7 @@ def main():
     p.add_argument("--seconds", type=float, default=8.0)
     p.add_argument("--use-lstm", action="store_true")
     p.add_argument("--threads", type=int, default=1)
+    p.add_argument("--jit", default="frozen", choices=["off", "safe", "frozen", "native"])
     args = p.parse_args()
     import torch.multiprocessing as mp
     ctx = mp.get_context("spawn")
@@ -58,7 +59,7 @@ def main():
     agg = sum(rates)
     print(json.dumps(dict(
         diagnostic="actor_probe", actors=args.actors, envs=args.envs,
-        threads=args.threads,
+        threads=args.threads, jit=args.jit,
         lstm=args.use_lstm, per_actor_fps=round(agg / args.actors, 1),
         aggregate_fps=round(agg, 1))))
 

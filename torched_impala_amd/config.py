@@ -33,7 +33,7 @@ class Flags:
     envs_per_actor: int = 1            # batched envs inside one actor process
     actor_threads: int = 0             # CPU threads per actor; 0 = auto
                                        # (cores/num_actors, capped at 8)
-    actor_jit: str = "frozen"          # off | safe (preserved-freeze) |
+    actor_jit: str = "native"          # off | safe | frozen |
                                        # frozen (weights folded; fastest,
                                        # refreshed every actor_jit_refresh
                                        # slot claims)

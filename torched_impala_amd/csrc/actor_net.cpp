@@ -51,6 +51,7 @@ struct ShallowActorNet {
   std::vector<float> img, a1, a2, a3, fc_out, ci, logits_pad;
 
   explicit ShallowActorNet(int64_t seed) : rng((uint64_t)seed) {
+    use512 = __builtin_cpu_supports("avx512f");
     img.resize(C0 * H0 * W0);
     a1.resize(F1 * H1 * W1);
     a2.resize(F2 * H2 * W2);
@@ -77,7 +78,7 @@ struct ShallowActorNet {
     check(fw, {FC, FLAT}, "fc.w");
     TORCH_CHECK(pw.size(1) == FC + 1 + a, "policy in-dim must be 256+1+A");
     A = a;
-    A_pad = (a + 7) & ~7;
+    A_pad = (a + 15) & ~15;
     pol_in = FC + 1 + a;
     // conv weights -> [c][ky][kx][f]
     auto pack_conv = [](std::vector<float>& dst, const torch::Tensor& src,
@@ -115,7 +116,160 @@ struct ShallowActorNet {
     std::memcpy(bpol.data(), pb.data_ptr<float>(), sizeof(float) * A);
   }
 
-  // --- per-frame stages (auto-vectorized: fixed bounds, restrict) ---
+  // --- per-frame stages: AVX-512 when the host has it (EPYC Zen4+ /
+  // this dev container), AVX2 otherwise; parity is exact (same fp32
+  // FMA contraction order per output element) ---
+
+  bool use512 = false;
+
+  void stage_conv1() { use512 ? stage_conv1_512() : stage_conv1_avx2(); }
+  void stage_conv2() { use512 ? stage_conv2_512() : stage_conv2_avx2(); }
+  void stage_conv3() { use512 ? stage_conv3_512() : stage_conv3_avx2(); }
+  void stage_fc() { use512 ? stage_fc_512() : stage_fc_avx2(); }
+
+  __attribute__((target("avx512f"))) void stage_conv1_512() {
+    const float* __restrict W = w1.data();
+    const float* __restrict I = img.data();
+    float* __restrict O = a1.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bia = _mm512_loadu_ps(b1.data());  // F1 == 16 == one zmm
+    for (int oy = 0; oy < H1; ++oy) {
+      for (int ox = 0; ox < W1; ox += 4) {
+        __m512 acc[4];
+        for (int p = 0; p < 4; ++p) acc[p] = zero;
+        for (int c = 0; c < C0; ++c) {
+          const float* __restrict ic = I + c * (H0 * W0);
+          for (int ky = 0; ky < K1; ++ky) {
+            const float* __restrict row = ic + (S1 * oy + ky) * W0 + S1 * ox;
+            const float* __restrict wr = W + ((c * K1 + ky) * K1) * F1;
+            for (int kx = 0; kx < K1; ++kx) {
+              const __m512 w = _mm512_loadu_ps(wr + kx * F1);
+              for (int p = 0; p < 4; ++p) {
+                const __m512 x = _mm512_set1_ps(row[S1 * p + kx]);
+                acc[p] = _mm512_fmadd_ps(x, w, acc[p]);
+              }
+            }
+          }
+        }
+        float buf[4][F1];
+        for (int p = 0; p < 4; ++p)
+          _mm512_storeu_ps(buf[p],
+                           _mm512_max_ps(_mm512_add_ps(acc[p], bia), zero));
+        for (int p = 0; p < 4; ++p) {
+          const int px = oy * W1 + ox + p;
+          for (int f = 0; f < F1; ++f) O[f * (H1 * W1) + px] = buf[p][f];
+        }
+      }
+    }
+  }
+
+  __attribute__((target("avx512f"))) void stage_conv2_512() {
+    const float* __restrict W = w2.data();
+    const float* __restrict I = a1.data();
+    float* __restrict O = a2.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bl = _mm512_loadu_ps(b2.data());
+    const __m512 bh = _mm512_loadu_ps(b2.data() + 16);
+    for (int px0 = 0; px0 < H2 * W2; px0 += 3) {  // 81 = 27*3
+      __m512 acc[3][2];
+      for (int p = 0; p < 3; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
+      int base[3];
+      for (int p = 0; p < 3; ++p) {
+        const int oy = (px0 + p) / W2, ox = (px0 + p) - oy * W2;
+        base[p] = S2 * oy * W1 + S2 * ox;
+      }
+      for (int c = 0; c < F1; ++c) {
+        const float* __restrict ic = I + c * (H1 * W1);
+        for (int ky = 0; ky < K2; ++ky) {
+          const float* __restrict wr = W + ((c * K2 + ky) * K2) * F2;
+          const int roff = ky * W1;
+          for (int kx = 0; kx < K2; ++kx) {
+            const __m512 w0 = _mm512_loadu_ps(wr + kx * F2);
+            const __m512 w1v = _mm512_loadu_ps(wr + kx * F2 + 16);
+            for (int p = 0; p < 3; ++p) {
+              const __m512 x = _mm512_set1_ps(ic[base[p] + roff + kx]);
+              acc[p][0] = _mm512_fmadd_ps(x, w0, acc[p][0]);
+              acc[p][1] = _mm512_fmadd_ps(x, w1v, acc[p][1]);
+            }
+          }
+        }
+      }
+      for (int p = 0; p < 3; ++p) {
+        float buf[F2];
+        _mm512_storeu_ps(buf,
+                         _mm512_max_ps(_mm512_add_ps(acc[p][0], bl), zero));
+        _mm512_storeu_ps(buf + 16,
+                         _mm512_max_ps(_mm512_add_ps(acc[p][1], bh), zero));
+        const int px = px0 + p;
+        for (int f = 0; f < F2; ++f) O[f * (H2 * W2) + px] = buf[f];
+      }
+    }
+  }
+
+  __attribute__((target("avx512f"))) void stage_conv3_512() {
+    const float* __restrict W = w3.data();
+    const float* __restrict I = a2.data();
+    float* __restrict O = a3.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bl = _mm512_loadu_ps(b3.data());
+    const __m512 bh = _mm512_loadu_ps(b3.data() + 16);
+    for (int px0 = 0; px0 < H3 * W3; px0 += 2) {  // 49 = 24*2 + 1
+      const int np = (px0 + 1 < H3 * W3) ? 2 : 1;
+      __m512 acc[2][2];
+      for (int p = 0; p < 2; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
+      int base[2] = {0, 0};
+      for (int p = 0; p < np; ++p) {
+        const int oy = (px0 + p) / W3, ox = (px0 + p) - oy * W3;
+        base[p] = oy * W2 + ox;
+      }
+      for (int c = 0; c < F2; ++c) {
+        const float* __restrict ic = I + c * (H2 * W2);
+        for (int ky = 0; ky < K3; ++ky) {
+          const float* __restrict wr = W + ((c * K3 + ky) * K3) * F3;
+          for (int kx = 0; kx < K3; ++kx) {
+            const __m512 w0 = _mm512_loadu_ps(wr + kx * F3);
+            const __m512 w1v = _mm512_loadu_ps(wr + kx * F3 + 16);
+            for (int p = 0; p < np; ++p) {
+              const __m512 x = _mm512_set1_ps(ic[base[p] + ky * W2 + kx]);
+              acc[p][0] = _mm512_fmadd_ps(x, w0, acc[p][0]);
+              acc[p][1] = _mm512_fmadd_ps(x, w1v, acc[p][1]);
+            }
+          }
+        }
+      }
+      for (int p = 0; p < np; ++p) {
+        float buf[F3];
+        _mm512_storeu_ps(buf,
+                         _mm512_max_ps(_mm512_add_ps(acc[p][0], bl), zero));
+        _mm512_storeu_ps(buf + 16,
+                         _mm512_max_ps(_mm512_add_ps(acc[p][1], bh), zero));
+        const int px = px0 + p;
+        for (int f = 0; f < F3; ++f) O[f * (H3 * W3) + px] = buf[f];
+      }
+    }
+  }
+
+  __attribute__((target("avx512f"))) void stage_fc_512() {
+    const float* __restrict X = a3.data();
+    const float* __restrict W = wfc.data();
+    float* __restrict O = fc_out.data();
+    const __m512 zero = _mm512_setzero_ps();
+    for (int f0 = 0; f0 < FC; f0 += 128) {  // 2 column blocks of 8 zmm
+      __m512 acc[8];
+      for (int q = 0; q < 8; ++q)
+        acc[q] = _mm512_loadu_ps(bfc.data() + f0 + 16 * q);
+      for (int k = 0; k < FLAT; ++k) {
+        const __m512 x = _mm512_set1_ps(X[k]);
+        const float* __restrict wv = W + (size_t)k * FC + f0;
+        for (int q = 0; q < 8; ++q)
+          acc[q] = _mm512_fmadd_ps(x, _mm512_loadu_ps(wv + 16 * q), acc[q]);
+      }
+      for (int q = 0; q < 8; ++q)
+        _mm512_storeu_ps(O + f0 + 16 * q, _mm512_max_ps(acc[q], zero));
+    }
+  }
+
+  // --- per-frame stages (AVX2 forms) ---
 
   void stage_convert(const unsigned char* __restrict fr) {
     constexpr float inv = 1.0f / 255.0f;
@@ -123,7 +277,7 @@ struct ShallowActorNet {
     for (int i = 0; i < C0 * H0 * W0; ++i) o[i] = (float)fr[i] * inv;
   }
 
-  void stage_conv1() {
+  void stage_conv1_avx2() {
     const float* __restrict W = w1.data();
     const float* __restrict I = img.data();
     float* __restrict O = a1.data();
@@ -165,7 +319,7 @@ struct ShallowActorNet {
     }
   }
 
-  void stage_conv2() {
+  void stage_conv2_avx2() {
     const float* __restrict W = w2.data();
     const float* __restrict I = a1.data();
     float* __restrict O = a2.data();
@@ -212,7 +366,7 @@ struct ShallowActorNet {
     }
   }
 
-  void stage_conv3() {
+  void stage_conv3_avx2() {
     const float* __restrict W = w3.data();
     const float* __restrict I = a2.data();
     float* __restrict O = a3.data();
@@ -260,7 +414,7 @@ struct ShallowActorNet {
     }
   }
 
-  void stage_fc() {
+  void stage_fc_avx2() {
     const float* __restrict X = a3.data();
     const float* __restrict W = wfc.data();
     float* __restrict O = fc_out.data();

@@ -72,16 +72,18 @@ class _FlatActorLstmFwd(torch.nn.Module):
 
 
 class NativeActorPolicy:
-    """AVX2 C++ actor inference (csrc/actor_net.cpp) for the shallow net:
-    zero torch dispatch, bit-level parity ~3e-8, cheap weight repack on
-    refresh. Shallow-net (4,84,84) feedforward policies only."""
+    """AVX-512/AVX2 C++ actor inference (csrc/actor_net.cpp): zero torch
+    dispatch, ~1e-7 parity, cheap weight repack on refresh. Covers the
+    shallow (4,84,84) net, feedforward or LSTM-256 core."""
 
     mode = "native"
-    use_lstm = False
 
     def __init__(self, model, num_actions: int, seed: int):
         import torched_impala_amd._impala_actor as ext
-        assert model.obs_shape == (4, 84, 84) and not model.use_lstm
+        assert model.obs_shape == (4, 84, 84)
+        self.use_lstm = bool(getattr(model, "use_lstm", False))
+        if self.use_lstm:
+            assert model.core.hidden_size == 256
         self._ext = ext.ShallowActorNet(seed)
         self._model = model
         self._A = num_actions
@@ -95,8 +97,26 @@ class NativeActorPolicy:
             m.conv3.weight.detach(), m.conv3.bias.detach(),
             m.fc.weight.detach(), m.fc.bias.detach(),
             m.policy.weight.detach(), m.policy.bias.detach())
+        if self.use_lstm:
+            self._ext.load_lstm(
+                m.core.weight_ih.detach(), m.core.weight_hh.detach(),
+                m.core.bias_ih.detach(), m.core.bias_hh.detach(),
+                m.policy.weight.detach(), m.policy.bias.detach())
 
-    def __call__(self, frame, reward, last_action):
+    def __call__(self, *args):
+        if self.use_lstm:
+            frame, reward, done, last_action, h, c = args
+            n = frame.shape[0]
+            logits = torch.empty(n, self._A)
+            action = torch.empty(n, dtype=torch.int64)
+            h2 = h.detach().clone().contiguous()
+            c2 = c.detach().clone().contiguous()
+            self._ext.forward_lstm(frame.contiguous(), reward.contiguous(),
+                                   done.contiguous(),
+                                   last_action.contiguous(), h2, c2,
+                                   logits, action)
+            return logits, action, h2, c2
+        frame, reward, last_action = args
         n = frame.shape[0]
         logits = torch.empty(n, self._A)
         action = torch.empty(n, dtype=torch.int64)
@@ -166,7 +186,7 @@ class Rollout:
         use_lstm = getattr(model, "use_lstm", False)
         if jit_mode != "off" and hasattr(model, "encode"):
             try:
-                if (jit_mode == "native" and not use_lstm
+                if (jit_mode == "native"
                         and getattr(model, "obs_shape", None) == (4, 84, 84)):
                     self.fast = NativeActorPolicy(
                         model, model.num_actions,

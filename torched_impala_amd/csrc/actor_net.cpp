@@ -76,10 +76,11 @@ struct ShallowActorNet {
     check(cw2, {F2, F1, K2, K2}, "w2");
     check(cw3, {F3, F2, K3, K3}, "w3");
     check(fw, {FC, FLAT}, "fc.w");
-    TORCH_CHECK(pw.size(1) == FC + 1 + a, "policy in-dim must be 256+1+A");
+    TORCH_CHECK(pw.size(1) == FC + 1 + a || pw.size(1) == FC,
+                "policy in-dim must be 256+1+A (feedforward) or 256 (lstm)");
     A = a;
     A_pad = (a + 15) & ~15;
-    pol_in = FC + 1 + a;
+    pol_in = (int)pw.size(1);
     // conv weights -> [c][ky][kx][f]
     auto pack_conv = [](std::vector<float>& dst, const torch::Tensor& src,
                         int F, int C, int K) {
@@ -105,15 +106,16 @@ struct ShallowActorNet {
         for (int k = 0; k < FLAT; ++k) wfc[(size_t)k * FC + f] = s[f][k];
     }
     bfc.assign(fb.data_ptr<float>(), fb.data_ptr<float>() + FC);
-    // policy -> [k][A_pad]
-    wpol.assign((size_t)pol_in * A_pad, 0.f);
-    {
+    // feedforward policy -> [k][A_pad] (the lstm head is packed by
+    // load_lstm instead)
+    if (pol_in == FC + 1 + a) {
+      wpol.assign((size_t)pol_in * A_pad, 0.f);
       auto s = pw.accessor<float, 2>();
       for (int j = 0; j < A; ++j)
         for (int k = 0; k < pol_in; ++k) wpol[(size_t)k * A_pad + j] = s[j][k];
+      bpol.assign(A_pad, 0.f);
+      std::memcpy(bpol.data(), pb.data_ptr<float>(), sizeof(float) * A);
     }
-    bpol.assign(A_pad, 0.f);
-    std::memcpy(bpol.data(), pb.data_ptr<float>(), sizeof(float) * A);
   }
 
   // --- per-frame stages: AVX-512 when the host has it (EPYC Zen4+ /
@@ -472,6 +474,140 @@ struct ShallowActorNet {
     return action;
   }
 
+  // ---------------- optional LSTM core (hidden 256) ----------------
+  // gates = W_ih @ [fc, r, onehot] + W_hh @ h + b;  i,f,g,o; done mask.
+  bool has_lstm = false;
+  std::vector<float> wih, whh, blstm;  // [263][1024], [256][1024], [1024]
+  std::vector<float> gates;            // [1024]
+  int lstm_in = 0;
+
+  void load_lstm(torch::Tensor w_ih, torch::Tensor w_hh, torch::Tensor b_ih,
+                 torch::Tensor b_hh, torch::Tensor pw, torch::Tensor pb) {
+    TORCH_CHECK(w_ih.size(0) == 1024 && w_hh.size(0) == 1024 &&
+                    w_hh.size(1) == FC, "LSTM must be 256-hidden");
+    lstm_in = w_ih.size(1);
+    TORCH_CHECK(lstm_in == FC + 1 + A, "core in-dim must be 256+1+A");
+    // k-major packs
+    wih.assign((size_t)lstm_in * 1024, 0.f);
+    {
+      auto sacc = w_ih.accessor<float, 2>();
+      for (int j = 0; j < 1024; ++j)
+        for (int k = 0; k < lstm_in; ++k) wih[(size_t)k * 1024 + j] = sacc[j][k];
+    }
+    whh.assign((size_t)FC * 1024, 0.f);
+    {
+      auto sacc = w_hh.accessor<float, 2>();
+      for (int j = 0; j < 1024; ++j)
+        for (int k = 0; k < FC; ++k) whh[(size_t)k * 1024 + j] = sacc[j][k];
+    }
+    blstm.assign(1024, 0.f);
+    for (int j = 0; j < 1024; ++j)
+      blstm[j] = b_ih.data_ptr<float>()[j] + b_hh.data_ptr<float>()[j];
+    // policy head over h (in-dim 256): reuse wpol storage k-major
+    TORCH_CHECK(pw.size(1) == FC, "lstm policy in-dim must be 256");
+    wpol.assign((size_t)FC * A_pad, 0.f);
+    {
+      auto sacc = pw.accessor<float, 2>();
+      for (int j = 0; j < A; ++j)
+        for (int k = 0; k < FC; ++k) wpol[(size_t)k * A_pad + j] = sacc[j][k];
+    }
+    bpol.assign(A_pad, 0.f);
+    std::memcpy(bpol.data(), pb.data_ptr<float>(), sizeof(float) * A);
+    gates.assign(1024, 0.f);
+    has_lstm = true;
+  }
+
+  // gates += W[k-major] rows weighted by x over [k0, k1)
+  void gemv_acc(const float* __restrict W, const float* __restrict x,
+                int k0, int k1, float* __restrict out) {
+    for (int k = k0; k < k1; ++k) {
+      const float xv = x[k - k0];
+      const float* __restrict wv = W + (size_t)k * 1024;
+      for (int j = 0; j < 1024; ++j) out[j] += xv * wv[j];
+    }
+  }
+
+  static float sigm(float x) { return 1.f / (1.f + std::exp(-x)); }
+
+  // one LSTM step for one frame; h/c updated in place (256 each)
+  void stage_lstm(float reward, int last_action, float nd,
+                  float* __restrict h, float* __restrict c,
+                  float* __restrict lg_out) {
+    float ci_tail[64];  // reward + onehot tail of the core input
+    const float rc = reward > 1.f ? 1.f : (reward < -1.f ? -1.f : reward);
+    const int tail = 1 + A;
+    for (int i = 0; i < tail; ++i) ci_tail[i] = 0.f;
+    ci_tail[0] = rc;
+    ci_tail[1 + last_action] = 1.f;
+    for (int j = 0; j < 1024; ++j) gates[j] = blstm[j];
+    gemv_acc(wih.data(), fc_out.data(), 0, FC, gates.data());
+    gemv_acc(wih.data(), ci_tail, FC, FC + tail, gates.data());
+    float hm[FC], cm[FC];
+    for (int k = 0; k < FC; ++k) { hm[k] = h[k] * nd; cm[k] = c[k] * nd; }
+    gemv_acc(whh.data(), hm, 0, FC, gates.data());
+    for (int k = 0; k < FC; ++k) {
+      const float gi = sigm(gates[k]);
+      const float gf = sigm(gates[FC + k]);
+      const float gg = std::tanh(gates[2 * FC + k]);
+      const float go = sigm(gates[3 * FC + k]);
+      const float cv = gf * cm[k] + gi * gg;
+      c[k] = cv;
+      h[k] = go * std::tanh(cv);
+    }
+    // policy over h
+    float acc[64];
+    for (int j = 0; j < A_pad; ++j) acc[j] = bpol[j];
+    for (int k = 0; k < FC; ++k) {
+      const float x = h[k];
+      const float* __restrict wv = wpol.data() + (size_t)k * A_pad;
+      for (int j = 0; j < A_pad; ++j) acc[j] += x * wv[j];
+    }
+    std::memcpy(lg_out, acc, sizeof(float) * A);
+  }
+
+  int sample(const float* __restrict lg) {
+    float mx = lg[0];
+    for (int j = 1; j < A; ++j) mx = lg[j] > mx ? lg[j] : mx;
+    float sum = 0.f, p[64];
+    for (int j = 0; j < A; ++j) { p[j] = std::exp(lg[j] - mx); sum += p[j]; }
+    std::uniform_real_distribution<float> U(0.f, 1.f);
+    float u = U(rng) * sum;
+    int action = A - 1;
+    float cum = 0.f;
+    for (int j = 0; j < A; ++j) {
+      cum += p[j];
+      if (u <= cum) { action = j; break; }
+    }
+    return action;
+  }
+
+  // LSTM batch forward: h/c (n, 256) fp32 updated IN PLACE
+  void forward_lstm(torch::Tensor frames, torch::Tensor reward,
+                    torch::Tensor done, torch::Tensor last_action,
+                    torch::Tensor h, torch::Tensor c,
+                    torch::Tensor out_logits, torch::Tensor out_action) {
+    TORCH_CHECK(has_lstm, "load_lstm first");
+    const int n = frames.size(0);
+    const unsigned char* fr = frames.data_ptr<unsigned char>();
+    const float* rew = reward.data_ptr<float>();
+    const bool* dn = done.data_ptr<bool>();
+    const int64_t* la = last_action.data_ptr<int64_t>();
+    float* hp = h.data_ptr<float>();
+    float* cp = c.data_ptr<float>();
+    float* lg = out_logits.data_ptr<float>();
+    int64_t* act = out_action.data_ptr<int64_t>();
+    for (int i = 0; i < n; ++i) {
+      stage_convert(fr + (size_t)i * C0 * H0 * W0);
+      stage_conv1();
+      stage_conv2();
+      stage_conv3();
+      stage_fc();
+      stage_lstm(rew[i], (int)la[i], dn[i] ? 0.f : 1.f, hp + (size_t)i * FC,
+                 cp + (size_t)i * FC, lg + (size_t)i * A);
+      act[i] = sample(lg + (size_t)i * A);
+    }
+  }
+
   // batch forward: frames (n, 4, 84, 84) u8; outputs logits (n, A) f32,
   // actions (n) i64
   void forward(torch::Tensor frames, torch::Tensor reward,
@@ -506,5 +642,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   py::class_<ShallowActorNet>(m, "ShallowActorNet")
       .def(py::init<int64_t>())
       .def("load_weights", &ShallowActorNet::load_weights)
-      .def("forward", &ShallowActorNet::forward);
+      .def("load_lstm", &ShallowActorNet::load_lstm)
+      .def("forward", &ShallowActorNet::forward)
+      .def("forward_lstm", &ShallowActorNet::forward_lstm);
 }

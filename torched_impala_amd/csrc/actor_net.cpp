@@ -529,6 +529,99 @@ struct ShallowActorNet {
 
   static float sigm(float x) { return 1.f / (1.f + std::exp(-x)); }
 
+  // ---- AVX-512 LSTM helpers ----
+
+  // exp(x) via range reduction + degree-5 polynomial; |rel err| ~1e-7
+  __attribute__((target("avx512f"))) static inline __m512 exp512(__m512 x) {
+    const __m512 hi = _mm512_set1_ps(88.3762626647950f);
+    const __m512 lo = _mm512_set1_ps(-87.3365478515625f);
+    x = _mm512_min_ps(_mm512_max_ps(x, lo), hi);
+    const __m512 log2e = _mm512_set1_ps(1.44269504088896341f);
+    const __m512 c1 = _mm512_set1_ps(0.693359375f);
+    const __m512 c2 = _mm512_set1_ps(-2.12194440e-4f);
+    __m512 fx = _mm512_roundscale_ps(_mm512_mul_ps(x, log2e),
+                                     _MM_FROUND_TO_NEAREST_INT);
+    x = _mm512_fnmadd_ps(fx, c1, x);
+    x = _mm512_fnmadd_ps(fx, c2, x);
+    const __m512 z = _mm512_mul_ps(x, x);
+    __m512 y = _mm512_set1_ps(1.9875691500e-4f);
+    y = _mm512_fmadd_ps(y, x, _mm512_set1_ps(1.3981999507e-3f));
+    y = _mm512_fmadd_ps(y, x, _mm512_set1_ps(8.3334519073e-3f));
+    y = _mm512_fmadd_ps(y, x, _mm512_set1_ps(4.1665795894e-2f));
+    y = _mm512_fmadd_ps(y, x, _mm512_set1_ps(1.6666665459e-1f));
+    y = _mm512_fmadd_ps(y, x, _mm512_set1_ps(5.0000001201e-1f));
+    y = _mm512_fmadd_ps(y, z, x);
+    y = _mm512_add_ps(y, _mm512_set1_ps(1.0f));
+    const __m512i n = _mm512_cvtps_epi32(fx);
+    const __m512i pow2n =
+        _mm512_slli_epi32(_mm512_add_epi32(n, _mm512_set1_epi32(127)), 23);
+    return _mm512_mul_ps(y, _mm512_castsi512_ps(pow2n));
+  }
+
+  __attribute__((target("avx512f"))) static inline __m512 sigm512(__m512 x) {
+    const __m512 one = _mm512_set1_ps(1.0f);
+    const __m512 e = exp512(_mm512_sub_ps(_mm512_setzero_ps(), x));
+    return _mm512_div_ps(one, _mm512_add_ps(one, e));
+  }
+
+  __attribute__((target("avx512f"))) static inline __m512 tanh512(__m512 x) {
+    // tanh(x) = (e2x - 1) / (e2x + 1), clamped
+    const __m512 one = _mm512_set1_ps(1.0f);
+    const __m512 e =
+        exp512(_mm512_mul_ps(x, _mm512_set1_ps(2.0f)));
+    return _mm512_div_ps(_mm512_sub_ps(e, one), _mm512_add_ps(e, one));
+  }
+
+  // blocked gates GEMV: 128 output cols in 8 zmm per pass
+  __attribute__((target("avx512f")))
+  void gates_gemv_512(const float* __restrict x1, int n1,
+                      const float* __restrict x2, int n2,
+                      const float* __restrict hm) {
+    for (int j0 = 0; j0 < 1024; j0 += 128) {
+      __m512 acc[8];
+      for (int q = 0; q < 8; ++q)
+        acc[q] = _mm512_loadu_ps(blstm.data() + j0 + 16 * q);
+      const float* __restrict W = wih.data() + j0;
+      for (int k = 0; k < n1; ++k) {
+        const __m512 x = _mm512_set1_ps(x1[k]);
+        const float* __restrict wv = W + (size_t)k * 1024;
+        for (int q = 0; q < 8; ++q)
+          acc[q] = _mm512_fmadd_ps(x, _mm512_loadu_ps(wv + 16 * q), acc[q]);
+      }
+      const float* __restrict W2 = wih.data() + (size_t)n1 * 1024 + j0;
+      for (int k = 0; k < n2; ++k) {
+        const __m512 x = _mm512_set1_ps(x2[k]);
+        const float* __restrict wv = W2 + (size_t)k * 1024;
+        for (int q = 0; q < 8; ++q)
+          acc[q] = _mm512_fmadd_ps(x, _mm512_loadu_ps(wv + 16 * q), acc[q]);
+      }
+      const float* __restrict W3 = whh.data() + j0;
+      for (int k = 0; k < FC; ++k) {
+        const __m512 x = _mm512_set1_ps(hm[k]);
+        const float* __restrict wv = W3 + (size_t)k * 1024;
+        for (int q = 0; q < 8; ++q)
+          acc[q] = _mm512_fmadd_ps(x, _mm512_loadu_ps(wv + 16 * q), acc[q]);
+      }
+      for (int q = 0; q < 8; ++q)
+        _mm512_storeu_ps(gates.data() + j0 + 16 * q, acc[q]);
+    }
+  }
+
+  __attribute__((target("avx512f")))
+  void lstm_cell_512(const float* __restrict cm, float* __restrict h,
+                     float* __restrict c) {
+    for (int k = 0; k < FC; k += 16) {
+      const __m512 gi = sigm512(_mm512_loadu_ps(gates.data() + k));
+      const __m512 gf = sigm512(_mm512_loadu_ps(gates.data() + FC + k));
+      const __m512 gg = tanh512(_mm512_loadu_ps(gates.data() + 2 * FC + k));
+      const __m512 go = sigm512(_mm512_loadu_ps(gates.data() + 3 * FC + k));
+      const __m512 cv = _mm512_fmadd_ps(gf, _mm512_loadu_ps(cm + k),
+                                        _mm512_mul_ps(gi, gg));
+      _mm512_storeu_ps(c + k, cv);
+      _mm512_storeu_ps(h + k, _mm512_mul_ps(go, tanh512(cv)));
+    }
+  }
+
   // one LSTM step for one frame; h/c updated in place (256 each)
   void stage_lstm(float reward, int last_action, float nd,
                   float* __restrict h, float* __restrict c,
@@ -539,20 +632,25 @@ struct ShallowActorNet {
     for (int i = 0; i < tail; ++i) ci_tail[i] = 0.f;
     ci_tail[0] = rc;
     ci_tail[1 + last_action] = 1.f;
-    for (int j = 0; j < 1024; ++j) gates[j] = blstm[j];
-    gemv_acc(wih.data(), fc_out.data(), 0, FC, gates.data());
-    gemv_acc(wih.data(), ci_tail, FC, FC + tail, gates.data());
     float hm[FC], cm[FC];
     for (int k = 0; k < FC; ++k) { hm[k] = h[k] * nd; cm[k] = c[k] * nd; }
-    gemv_acc(whh.data(), hm, 0, FC, gates.data());
-    for (int k = 0; k < FC; ++k) {
-      const float gi = sigm(gates[k]);
-      const float gf = sigm(gates[FC + k]);
-      const float gg = std::tanh(gates[2 * FC + k]);
-      const float go = sigm(gates[3 * FC + k]);
-      const float cv = gf * cm[k] + gi * gg;
-      c[k] = cv;
-      h[k] = go * std::tanh(cv);
+    if (use512) {
+      gates_gemv_512(fc_out.data(), FC, ci_tail, tail, hm);
+      lstm_cell_512(cm, h, c);
+    } else {
+      for (int j = 0; j < 1024; ++j) gates[j] = blstm[j];
+      gemv_acc(wih.data(), fc_out.data(), 0, FC, gates.data());
+      gemv_acc(wih.data(), ci_tail, FC, FC + tail, gates.data());
+      gemv_acc(whh.data(), hm, 0, FC, gates.data());
+      for (int k = 0; k < FC; ++k) {
+        const float gi = sigm(gates[k]);
+        const float gf = sigm(gates[FC + k]);
+        const float gg = std::tanh(gates[2 * FC + k]);
+        const float go = sigm(gates[3 * FC + k]);
+        const float cv = gf * cm[k] + gi * gg;
+        c[k] = cv;
+        h[k] = go * std::tanh(cv);
+      }
     }
     // policy over h
     float acc[64];
@@ -639,6 +737,8 @@ struct ShallowActorNet {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "native CPU actor inference (shallow IMPALA net)";
+  m.def("avx512_supported",
+        [] { return (bool)__builtin_cpu_supports("avx512f"); });
   py::class_<ShallowActorNet>(m, "ShallowActorNet")
       .def(py::init<int64_t>())
       .def("load_weights", &ShallowActorNet::load_weights)

@@ -186,15 +186,7 @@ class Rollout:
         use_lstm = getattr(model, "use_lstm", False)
         if jit_mode != "off" and hasattr(model, "encode"):
             try:
-                if (jit_mode == "native" and not use_lstm
-                        and getattr(model, "obs_shape", None) == (4, 84, 84)):
-                    # LSTM policies: the traced-frozen path batches the
-                    # 2.1MB gate GEMM across the env batch and measures
-                    # faster than the per-frame native LSTM step
-                    self.fast = NativeActorPolicy(
-                        model, model.num_actions,
-                        seed=flags.seed + 77777 * actor_index)
-                elif (jit_mode == "__native_lstm__"  # reserved: see ROADMAP
+                if (jit_mode == "native"
                         and getattr(model, "obs_shape", None) == (4, 84, 84)):
                     self.fast = NativeActorPolicy(
                         model, model.num_actions,

@@ -679,6 +679,66 @@ struct ShallowActorNet {
     return action;
   }
 
+  // batched gates GEMM: one j-block of 32 cols x up to 8 frames in
+  // registers; weight lines are loaded once per (k, j-block) and shared
+  // by the whole frame group => gate-weight traffic /8 per frame.
+  __attribute__((target("avx512f")))
+  void gates_gemv_batched_512(int nf, const float* __restrict fc_all,
+                              const float* __restrict tail_all, int tail,
+                              const float* __restrict hm_all,
+                              float* __restrict gates_all) {
+    for (int i0 = 0; i0 < nf; i0 += 8) {
+      const int gn = nf - i0 < 8 ? nf - i0 : 8;
+      for (int j0 = 0; j0 < 1024; j0 += 32) {
+        __m512 acc[8][2];
+        for (int g = 0; g < 8; ++g) {
+          acc[g][0] = _mm512_loadu_ps(blstm.data() + j0);
+          acc[g][1] = _mm512_loadu_ps(blstm.data() + j0 + 16);
+        }
+        const float* __restrict W = wih.data() + j0;
+        for (int k = 0; k < FC; ++k) {
+          const float* __restrict wv = W + (size_t)k * 1024;
+          const __m512 w0 = _mm512_loadu_ps(wv);
+          const __m512 w1 = _mm512_loadu_ps(wv + 16);
+          for (int g = 0; g < gn; ++g) {
+            const __m512 x = _mm512_set1_ps(fc_all[(size_t)(i0 + g) * FC + k]);
+            acc[g][0] = _mm512_fmadd_ps(x, w0, acc[g][0]);
+            acc[g][1] = _mm512_fmadd_ps(x, w1, acc[g][1]);
+          }
+        }
+        const float* __restrict Wt = wih.data() + (size_t)FC * 1024 + j0;
+        for (int k = 0; k < tail; ++k) {
+          const float* __restrict wv = Wt + (size_t)k * 1024;
+          const __m512 w0 = _mm512_loadu_ps(wv);
+          const __m512 w1 = _mm512_loadu_ps(wv + 16);
+          for (int g = 0; g < gn; ++g) {
+            const __m512 x =
+                _mm512_set1_ps(tail_all[(size_t)(i0 + g) * 64 + k]);
+            acc[g][0] = _mm512_fmadd_ps(x, w0, acc[g][0]);
+            acc[g][1] = _mm512_fmadd_ps(x, w1, acc[g][1]);
+          }
+        }
+        const float* __restrict Wh = whh.data() + j0;
+        for (int k = 0; k < FC; ++k) {
+          const float* __restrict wv = Wh + (size_t)k * 1024;
+          const __m512 w0 = _mm512_loadu_ps(wv);
+          const __m512 w1 = _mm512_loadu_ps(wv + 16);
+          for (int g = 0; g < gn; ++g) {
+            const __m512 x = _mm512_set1_ps(hm_all[(size_t)(i0 + g) * FC + k]);
+            acc[g][0] = _mm512_fmadd_ps(x, w0, acc[g][0]);
+            acc[g][1] = _mm512_fmadd_ps(x, w1, acc[g][1]);
+          }
+        }
+        for (int g = 0; g < gn; ++g) {
+          _mm512_storeu_ps(gates_all + (size_t)(i0 + g) * 1024 + j0,
+                           acc[g][0]);
+          _mm512_storeu_ps(gates_all + (size_t)(i0 + g) * 1024 + j0 + 16,
+                           acc[g][1]);
+        }
+      }
+    }
+  }
+
   // LSTM batch forward: h/c (n, 256) fp32 updated IN PLACE
   void forward_lstm(torch::Tensor frames, torch::Tensor reward,
                     torch::Tensor done, torch::Tensor last_action,
@@ -694,14 +754,59 @@ struct ShallowActorNet {
     float* cp = c.data_ptr<float>();
     float* lg = out_logits.data_ptr<float>();
     int64_t* act = out_action.data_ptr<int64_t>();
+    if (!use512) {
+      for (int i = 0; i < n; ++i) {
+        stage_convert(fr + (size_t)i * C0 * H0 * W0);
+        stage_conv1();
+        stage_conv2();
+        stage_conv3();
+        stage_fc();
+        stage_lstm(rew[i], (int)la[i], dn[i] ? 0.f : 1.f,
+                   hp + (size_t)i * FC, cp + (size_t)i * FC,
+                   lg + (size_t)i * A);
+        act[i] = sample(lg + (size_t)i * A);
+      }
+      return;
+    }
+    // AVX-512: encode all frames, then one frame-batched gates GEMM
+    const int tail = 1 + A;
+    std::vector<float> fc_all((size_t)n * FC), hm_all((size_t)n * FC),
+        cm_all((size_t)n * FC), tails((size_t)n * 64, 0.f),
+        gates_all((size_t)n * 1024);
     for (int i = 0; i < n; ++i) {
       stage_convert(fr + (size_t)i * C0 * H0 * W0);
       stage_conv1();
       stage_conv2();
       stage_conv3();
       stage_fc();
-      stage_lstm(rew[i], (int)la[i], dn[i] ? 0.f : 1.f, hp + (size_t)i * FC,
-                 cp + (size_t)i * FC, lg + (size_t)i * A);
+      std::memcpy(fc_all.data() + (size_t)i * FC, fc_out.data(),
+                  sizeof(float) * FC);
+      const float nd = dn[i] ? 0.f : 1.f;
+      for (int k = 0; k < FC; ++k) {
+        hm_all[(size_t)i * FC + k] = hp[(size_t)i * FC + k] * nd;
+        cm_all[(size_t)i * FC + k] = cp[(size_t)i * FC + k] * nd;
+      }
+      const float r = rew[i];
+      tails[(size_t)i * 64] = r > 1.f ? 1.f : (r < -1.f ? -1.f : r);
+      tails[(size_t)i * 64 + 1 + (int)la[i]] = 1.f;
+    }
+    gates_gemv_batched_512(n, fc_all.data(), tails.data(), tail,
+                           hm_all.data(), gates_all.data());
+    for (int i = 0; i < n; ++i) {
+      std::memcpy(gates.data(), gates_all.data() + (size_t)i * 1024,
+                  sizeof(float) * 1024);
+      lstm_cell_512(cm_all.data() + (size_t)i * FC, hp + (size_t)i * FC,
+                    cp + (size_t)i * FC);
+      // policy over h
+      float accp[64];
+      for (int j = 0; j < A_pad; ++j) accp[j] = bpol[j];
+      const float* __restrict hrow = hp + (size_t)i * FC;
+      for (int k = 0; k < FC; ++k) {
+        const float x = hrow[k];
+        const float* __restrict wv = wpol.data() + (size_t)k * A_pad;
+        for (int j = 0; j < A_pad; ++j) accp[j] += x * wv[j];
+      }
+      std::memcpy(lg + (size_t)i * A, accp, sizeof(float) * A);
       act[i] = sample(lg + (size_t)i * A);
     }
   }

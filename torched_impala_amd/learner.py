@@ -437,12 +437,33 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
             timers.start("rollout_wait")
             batch = batcher.next_batch(materialize=graphed is None)
             si = batch.get("_staging_index")
+            if graphed is not None and world > 1 and learner_steps == 0:
+                # multi-rank: agree on graph capture BEFORE any training
+                # collective — a rank that silently fell back to the
+                # bucketed eager path would issue a different collective
+                # pattern than the flat-all-reduce graph path and hang
+                # the job. One MIN-reduce of the capture outcome keeps
+                # every rank on the same path.
+                try:
+                    graphed._capture(si)
+                    ok = 1.0
+                except Exception:
+                    log.exception("graph capture failed on this rank")
+                    ok = 0.0
+                flag = torch.tensor([ok], device=device)
+                torch.distributed.all_reduce(
+                    flag, op=torch.distributed.ReduceOp.MIN)
+                if float(flag.item()) < 1.0:
+                    log.warning("graph capture unavailable on >=1 rank; "
+                                "all ranks using eager learn steps")
+                    graphed = None
+                    reducer = FlatGradReducer(model, optimizer)
             if graphed is not None:
                 timers.start("graph_step")
                 try:
                     stats = graphed.step(si, scheduler)
                 except Exception:
-                    log.exception("graph capture failed; falling back to "
+                    log.exception("graph step failed; falling back to "
                                   "eager learn steps")
                     graphed = None
                     if world > 1 and reducer is None:

@@ -146,3 +146,15 @@ def test_ring_core_tsan():
         r = subprocess.run([exe], capture_output=True, text=True, timeout=300)
         assert r.returncode == 0, f"tsan_ring failed:\n{r.stdout}\n{r.stderr}"
         assert "OK" in r.stdout
+
+
+def test_slotqueue_fallback_mode():
+    """The mp.Queue fallback (no C++ ext) exposes the same API."""
+    q = SlotQueue(8, force_fallback=True)
+    assert q.pop() == -1
+    assert q.push(3) and q.push(5)
+    assert q.pop_wait(timeout_ms=1000) == 3
+    assert q.pop() == 5
+    assert q.pop_wait(timeout_ms=50) == -1
+    q.abort()
+    assert q.pop_wait(timeout_ms=1000) == -2

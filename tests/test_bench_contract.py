@@ -47,3 +47,26 @@ def test_cli_train_entry():
          "--disable-checkpoint"],
         capture_output=True, text=True, cwd=ROOT, timeout=360)
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+@pytest.mark.timeout(420)
+def test_bench_torchrun_two_ranks_cpu():
+    """The driver's multi-GPU scaling invocation, on CPU/gloo: torchrun
+    2 ranks -> init_distributed -> per-rank actors -> eager learn steps
+    with FlatGradReducer -> MAX-reduced timing -> rank0 JSON line."""
+    import shutil
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29721", "bench.py", "--gpus", "2",
+         "--device", "cpu", "--steps", "3", "--warmup", "1",
+         "--num-actors", "1", "--batch-size", "2", "--unroll-length", "5",
+         "--envs-per-actor", "2"],
+        capture_output=True, text=True, cwd=ROOT, timeout=360)
+    assert r.returncode == 0, f"stdout:\n{r.stdout[-1500:]}\nstderr:\n{r.stderr[-2500:]}"
+    lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, f"expected exactly one JSON line (rank 0): {lines}"
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 4

@@ -36,7 +36,9 @@ class SyntheticFrameEnv:
                                  dtype=np.uint8)
         rng2 = np.random.default_rng(seed)
         self.reward_bank = (rng2.integers(0, 3, size=1024).astype(np.float32) - 1.0)
-        self.t = 0
+        # stagger the first episode boundary per env (same-seed envs stay
+        # identical) so batched envs don't all emit done on the same step
+        self.t = int(seed) % episode_len
         self.global_t = int(seed) * 977
 
     def _frame(self) -> np.ndarray:

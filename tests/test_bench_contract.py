@@ -70,3 +70,25 @@ def test_bench_torchrun_two_ranks_cpu():
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 4
+
+
+def test_flags_roundtrip_defaults():
+    """Every CLI flag maps onto a Flags field (no silent drops)."""
+    from torched_impala_amd.config import build_parser, Flags, flags_from_args
+    import dataclasses
+    ns = vars(build_parser().parse_args([]))
+    fields = {f.name for f in dataclasses.fields(Flags)}
+    unknown = set(ns) - fields
+    assert not unknown, f"parser dests without Flags fields: {unknown}"
+    f = flags_from_args([])
+    assert f.batch_size == Flags().batch_size
+
+
+def test_batch_envs_divisibility_error():
+    from torched_impala_amd.config import Flags
+    from torched_impala_amd.learner import create_buffers
+    import numpy as np
+    import pytest as _pt
+    flags = Flags(batch_size=6, envs_per_actor=4)
+    with _pt.raises(AssertionError):
+        create_buffers(flags, (4, 84, 84), np.uint8, 6)

@@ -53,7 +53,7 @@ def test_mirror_no_torn_weights_under_fast_writer():
         while True:
             alive = w.is_alive()
             last = mirror.pull_into(dst, last_seen=last)
-            vals = torch.cat([p.flatten() for p in dst.parameters()])
+            vals = torch.cat([p.detach().flatten() for p in dst.parameters()])
             uniq = torch.unique(vals)
             # THE invariant: a pulled model is never a mix of versions
             assert uniq.numel() == 1, f"torn weights: {uniq.tolist()[:5]}"

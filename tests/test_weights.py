@@ -63,7 +63,12 @@ def test_mirror_no_torn_weights_under_fast_writer():
             if not alive and int(mirror.version.item()) == last:
                 break
     finally:
-        w.join(timeout=30)
-    # the reader must at least observe the final published version
-    # (under CI load it may catch few intermediates — that's fine)
-    assert max(seen) >= n_pub, f"final version never observed: {sorted(seen)[-3:]}"
+        w.join(timeout=60)
+    assert w.exitcode == 0, f"writer died with exit code {w.exitcode}"
+    # after the writer exits the final publish is durable: one more pull
+    # must observe it (mid-flight observations depend on scheduler luck
+    # under CI load, so only the post-join state is asserted)
+    mirror.pull_into(dst, last_seen=-1)
+    vals = torch.unique(torch.cat([p.detach().flatten()
+                                   for p in dst.parameters()]))
+    assert vals.numel() == 1 and float(vals[0]) == n_pub, vals.tolist()[:5]

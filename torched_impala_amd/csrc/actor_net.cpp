@@ -136,9 +136,9 @@ struct ShallowActorNet {
     const __m512 zero = _mm512_setzero_ps();
     const __m512 bia = _mm512_loadu_ps(b1.data());  // F1 == 16 == one zmm
     for (int oy = 0; oy < H1; ++oy) {
-      for (int ox = 0; ox < W1; ox += 4) {
-        __m512 acc[4];
-        for (int p = 0; p < 4; ++p) acc[p] = zero;
+      for (int ox = 0; ox < W1; ox += 10) {  // 10-px block: one weight
+        __m512 acc[10];                      // load serves 10 FMAs
+        for (int p = 0; p < 10; ++p) acc[p] = zero;
         for (int c = 0; c < C0; ++c) {
           const float* __restrict ic = I + c * (H0 * W0);
           for (int ky = 0; ky < K1; ++ky) {
@@ -146,18 +146,18 @@ struct ShallowActorNet {
             const float* __restrict wr = W + ((c * K1 + ky) * K1) * F1;
             for (int kx = 0; kx < K1; ++kx) {
               const __m512 w = _mm512_loadu_ps(wr + kx * F1);
-              for (int p = 0; p < 4; ++p) {
+              for (int p = 0; p < 10; ++p) {
                 const __m512 x = _mm512_set1_ps(row[S1 * p + kx]);
                 acc[p] = _mm512_fmadd_ps(x, w, acc[p]);
               }
             }
           }
         }
-        float buf[4][F1];
-        for (int p = 0; p < 4; ++p)
+        float buf[10][F1];
+        for (int p = 0; p < 10; ++p)
           _mm512_storeu_ps(buf[p],
                            _mm512_max_ps(_mm512_add_ps(acc[p], bia), zero));
-        for (int p = 0; p < 4; ++p) {
+        for (int p = 0; p < 10; ++p) {
           const int px = oy * W1 + ox + p;
           for (int f = 0; f < F1; ++f) O[f * (H1 * W1) + px] = buf[p][f];
         }
@@ -172,37 +172,33 @@ struct ShallowActorNet {
     const __m512 zero = _mm512_setzero_ps();
     const __m512 bl = _mm512_loadu_ps(b2.data());
     const __m512 bh = _mm512_loadu_ps(b2.data() + 16);
-    for (int px0 = 0; px0 < H2 * W2; px0 += 3) {  // 81 = 27*3
-      __m512 acc[3][2];
-      for (int p = 0; p < 3; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
-      int base[3];
-      for (int p = 0; p < 3; ++p) {
-        const int oy = (px0 + p) / W2, ox = (px0 + p) - oy * W2;
-        base[p] = S2 * oy * W1 + S2 * ox;
-      }
+    for (int oy2 = 0; oy2 < H2; ++oy2) {  // one 9-px output row per
+      __m512 acc[9][2];                    // block (18 zmm of 32)
+      for (int p = 0; p < 9; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
+      const int rbase = S2 * oy2 * W1;
       for (int c = 0; c < F1; ++c) {
         const float* __restrict ic = I + c * (H1 * W1);
         for (int ky = 0; ky < K2; ++ky) {
           const float* __restrict wr = W + ((c * K2 + ky) * K2) * F2;
-          const int roff = ky * W1;
+          const int roff = rbase + ky * W1;
           for (int kx = 0; kx < K2; ++kx) {
             const __m512 w0 = _mm512_loadu_ps(wr + kx * F2);
             const __m512 w1v = _mm512_loadu_ps(wr + kx * F2 + 16);
-            for (int p = 0; p < 3; ++p) {
-              const __m512 x = _mm512_set1_ps(ic[base[p] + roff + kx]);
+            for (int p = 0; p < 9; ++p) {
+              const __m512 x = _mm512_set1_ps(ic[roff + S2 * p + kx]);
               acc[p][0] = _mm512_fmadd_ps(x, w0, acc[p][0]);
               acc[p][1] = _mm512_fmadd_ps(x, w1v, acc[p][1]);
             }
           }
         }
       }
-      for (int p = 0; p < 3; ++p) {
+      for (int p = 0; p < 9; ++p) {
         float buf[F2];
         _mm512_storeu_ps(buf,
                          _mm512_max_ps(_mm512_add_ps(acc[p][0], bl), zero));
         _mm512_storeu_ps(buf + 16,
                          _mm512_max_ps(_mm512_add_ps(acc[p][1], bh), zero));
-        const int px = px0 + p;
+        const int px = oy2 * W2 + p;
         for (int f = 0; f < F2; ++f) O[f * (H2 * W2) + px] = buf[f];
       }
     }
@@ -215,37 +211,32 @@ struct ShallowActorNet {
     const __m512 zero = _mm512_setzero_ps();
     const __m512 bl = _mm512_loadu_ps(b3.data());
     const __m512 bh = _mm512_loadu_ps(b3.data() + 16);
-    for (int px0 = 0; px0 < H3 * W3; px0 += 2) {  // 49 = 24*2 + 1
-      const int np = (px0 + 1 < H3 * W3) ? 2 : 1;
-      __m512 acc[2][2];
-      for (int p = 0; p < 2; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
-      int base[2] = {0, 0};
-      for (int p = 0; p < np; ++p) {
-        const int oy = (px0 + p) / W3, ox = (px0 + p) - oy * W3;
-        base[p] = oy * W2 + ox;
-      }
+    for (int oy3 = 0; oy3 < H3; ++oy3) {  // one 7-px output row per
+      __m512 acc[7][2];                    // block (14 zmm)
+      for (int p = 0; p < 7; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
       for (int c = 0; c < F2; ++c) {
         const float* __restrict ic = I + c * (H2 * W2);
         for (int ky = 0; ky < K3; ++ky) {
           const float* __restrict wr = W + ((c * K3 + ky) * K3) * F3;
+          const float* __restrict row = ic + (oy3 + ky) * W2;
           for (int kx = 0; kx < K3; ++kx) {
             const __m512 w0 = _mm512_loadu_ps(wr + kx * F3);
             const __m512 w1v = _mm512_loadu_ps(wr + kx * F3 + 16);
-            for (int p = 0; p < np; ++p) {
-              const __m512 x = _mm512_set1_ps(ic[base[p] + ky * W2 + kx]);
+            for (int p = 0; p < 7; ++p) {
+              const __m512 x = _mm512_set1_ps(row[p + kx]);
               acc[p][0] = _mm512_fmadd_ps(x, w0, acc[p][0]);
               acc[p][1] = _mm512_fmadd_ps(x, w1v, acc[p][1]);
             }
           }
         }
       }
-      for (int p = 0; p < np; ++p) {
+      for (int p = 0; p < 7; ++p) {
         float buf[F3];
         _mm512_storeu_ps(buf,
                          _mm512_max_ps(_mm512_add_ps(acc[p][0], bl), zero));
         _mm512_storeu_ps(buf + 16,
                          _mm512_max_ps(_mm512_add_ps(acc[p][1], bh), zero));
-        const int px = px0 + p;
+        const int px = oy3 * W3 + p;
         for (int f = 0; f < F3; ++f) O[f * (H3 * W3) + px] = buf[f];
       }
     }

@@ -1,0 +1,389 @@
+// Deep IMPALA ResNet encoder kernels — NHWC bf16 implicit-GEMM conv3x3 on
+// gfx950 MFMA matrix cores (BASELINE.json:11 deep model; design:
+// docs/deep_conv_design.md).
+//
+// Formulation: one workgroup computes a 16x16 spatial output tile of one
+// image as a 256(M=px) x COUT(N) x 9*CIN(K) GEMM on
+// v_mfma_f32_16x16x32_bf16.  The (16+2)x(16+2)xCIN input patch is staged
+// once in LDS (halo included, optional fused ReLU on load) and reused by
+// all 9 taps — no im2col materialisation, no halo re-reads from HBM.
+// Weights are pre-packed [COUT][3][3][CIN] so a B fragment is a
+// contiguous 8x bf16 read.  K is zero-padded to a multiple of 32 with
+// phantom taps (tap >= 9 reads zeros on both sides).
+//
+// Layout facts (cdna_hip_programming.md §3, verified for the f32 forms in
+// linear_kernels.hip): A lane l holds row = l&15, k = (l>>4)*8 + 0..7;
+// B lane holds col = l&15, same k; C/D col = lane&15,
+// row = (lane>>4)*4 + reg.  The standalone probe (deep_probe.cpp) checks
+// the bf16 A/B mapping on hardware before anything else.
+//
+// Correctness-first (round-2 ladder step 1-2): single-buffered LDS, no
+// XCD swizzle, no glds. Tuning happens after GPU parity + PMC evidence.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+namespace impala_deep {
+
+typedef __bf16 bf16_t;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+
+__device__ __forceinline__ float bf2f(bf16_t v) { return (float)v; }
+__device__ __forceinline__ bf16_t f2bf(float v) { return (bf16_t)v; }
+
+// ReLU directly on the bf16 bit pattern: sign bit set -> zero.
+__device__ __forceinline__ unsigned short relu_bits(unsigned short u) {
+  return (u & 0x8000u) ? (unsigned short)0 : u;
+}
+
+constexpr int TY = 16, TX = 16;        // output tile (256 px = 16 M-frags)
+constexpr int THREADS = 256;           // 4 waves
+
+// ---------------------------------------------------------------------------
+// conv3x3 pad1 stride1, NHWC bf16, fused: optional ReLU on input read,
+// bias add, optional residual add.  CIN in {4,16,32}, COUT in {16,32}.
+// x:   [N][H][W][CIN]  bf16
+// w:   [COUT][3][3][CIN] bf16 (pre-packed; a [n][k] row-major K-matrix)
+// b:   [COUT] float
+// res: [N][H][W][COUT] bf16 or nullptr
+// y:   [N][H][W][COUT] bf16
+// ---------------------------------------------------------------------------
+template <int CIN, int COUT, bool RELU_IN, bool HAS_RES>
+__global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+    const float* __restrict__ b, const bf16_t* __restrict__ res,
+    bf16_t* __restrict__ y, int N, int H, int W) {
+  constexpr int KTOT = 9 * CIN;
+  constexpr int KSTEPS = (KTOT + 31) / 32;       // MFMA k-steps (zero-padded)
+  constexpr int NFRAG = COUT / 16;
+  constexpr int PSTR = (TX + 2) * CIN;           // patch row stride (halfs)
+  constexpr int WSTR = 9 * CIN;                  // weight row stride (halfs)
+
+  __shared__ bf16_t s_patch[(TY + 2) * PSTR];
+  __shared__ bf16_t s_w[COUT * WSTR];
+
+  const int ntx = (W + TX - 1) / TX;
+  const int nty = (H + TY - 1) / TY;
+  const int tile = blockIdx.x % (ntx * nty);
+  const int img = blockIdx.x / (ntx * nty);
+  if (img >= N) return;
+  const int ty0 = (tile / ntx) * TY;             // tile origin (output px)
+  const int tx0 = (tile % ntx) * TX;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  // --- stage weights: straight copy, COUT*WSTR halfs over 256 threads ----
+  {
+    const unsigned short* ws = reinterpret_cast<const unsigned short*>(w);
+    unsigned short* wd = reinterpret_cast<unsigned short*>(s_w);
+    for (int i = tid; i < COUT * WSTR; i += THREADS) wd[i] = ws[i];
+  }
+
+  // --- stage input patch with halo (+ optional fused ReLU) ---------------
+  {
+    const unsigned short* xs = reinterpret_cast<const unsigned short*>(x) +
+                               (size_t)img * H * W * CIN;
+    unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
+    for (int p = tid; p < (TY + 2) * (TX + 2); p += THREADS) {
+      const int py = p / (TX + 2), px = p % (TX + 2);
+      const int iy = ty0 + py - 1, ix = tx0 + px - 1;
+      unsigned short* dst = pd + py * PSTR + px * CIN;
+      if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+        const unsigned short* src = xs + ((size_t)iy * W + ix) * CIN;
+        #pragma unroll
+        for (int c = 0; c < CIN; ++c)
+          dst[c] = RELU_IN ? relu_bits(src[c]) : src[c];
+      } else {
+        #pragma unroll
+        for (int c = 0; c < CIN; ++c) dst[c] = 0;
+      }
+    }
+  }
+  __syncthreads();
+
+  // --- MFMA main loop ----------------------------------------------------
+  // wave handles m-frags {wave*4 .. wave*4+3}; all NFRAG n-frags.
+  floatx4 acc[4][NFRAG];
+  #pragma unroll
+  for (int m = 0; m < 4; ++m)
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) acc[m][n] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+  const int arow = lane & 15;        // A fragment row (px within m-frag)
+  const int kbase = (lane >> 4) * 8; // lane's k offset within a 32-step
+
+  #pragma unroll
+  for (int s = 0; s < KSTEPS; ++s) {
+    const int k0 = s * 32 + kbase;   // this lane's first k
+    // B fragments: contiguous 8 halfs along k of row (col) of s_w.
+    bf16x8 bf[NFRAG];
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) {
+      const int col = n * 16 + (lane & 15);
+      bf16x8 v = {};
+      if (k0 < KTOT) {
+        const bf16_t* src = s_w + col * WSTR + k0;
+        if constexpr (CIN >= 8) {
+          v = *reinterpret_cast<const bf16x8*>(src);   // within one tap
+        } else {
+          #pragma unroll
+          for (int j = 0; j < 8; ++j)
+            v[j] = (k0 + j < KTOT) ? src[j] : f2bf(0.f);
+        }
+      }
+      bf[n] = v;
+    }
+    // A fragments: patch pixel (m-frag row) shifted by the tap offset.
+    #pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const int p = (wave * 4 + m) * 16 + arow;   // px index in 16x16 tile
+      const int oy = p >> 4, ox = p & 15;
+      bf16x8 af = {};
+      if (k0 < KTOT) {
+        if constexpr (CIN >= 8) {
+          const int tap = k0 / CIN, c0 = k0 % CIN;  // 8 elems, one tap
+          const int ky = tap / 3, kx = tap % 3;
+          af = *reinterpret_cast<const bf16x8*>(
+              s_patch + (oy + ky) * PSTR + (ox + kx) * CIN + c0);
+        } else {
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int k = k0 + j;
+            if (k < KTOT) {
+              const int tap = k / CIN, c = k % CIN;
+              af[j] = s_patch[(oy + tap / 3) * PSTR + (ox + tap % 3) * CIN + c];
+            }
+          }
+        }
+      }
+      #pragma unroll
+      for (int n = 0; n < NFRAG; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af, bf[n], acc[m][n], 0, 0, 0);
+    }
+  }
+
+  // --- epilogue: bias (+res) -> bf16 NHWC --------------------------------
+  const int ccol = lane & 15;
+  const int crow = (lane >> 4) * 4;
+  #pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) {
+      const int cout = n * 16 + ccol;
+      const float bias = b[cout];
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int p = (wave * 4 + m) * 16 + crow + reg;
+        const int oy = ty0 + (p >> 4), ox = tx0 + (p & 15);
+        if (oy < H && ox < W) {
+          const size_t off = (((size_t)img * H + oy) * W + ox) * COUT + cout;
+          float v = acc[m][n][reg] + bias;
+          if (HAS_RES) v += bf2f(res[off]);
+          y[off] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// maxpool 3x3 stride2 pad1, NHWC bf16: out (H+1)/2 x (W+1)/2.
+// One thread per (out px, channel). bf16 compare is exact via float.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void maxpool3x3s2_nhwc(
+    const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+    int N, int H, int W, int C) {
+  const int HO = (H + 1) / 2, WO = (W + 1) / 2;
+  const size_t total = (size_t)N * HO * WO * C;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (size_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    const size_t p = i / C;
+    const int ox = p % WO;
+    const int oy = (p / WO) % HO;
+    const int img = p / ((size_t)WO * HO);
+    const int iy0 = oy * 2 - 1, ix0 = ox * 2 - 1;
+    float best = -3.4e38f;
+    for (int dy = 0; dy < 3; ++dy) {
+      const int iy = iy0 + dy;
+      if (iy < 0 || iy >= H) continue;
+      for (int dx = 0; dx < 3; ++dx) {
+        const int ix = ix0 + dx;
+        if (ix < 0 || ix >= W) continue;
+        const float v =
+            bf2f(x[(((size_t)img * H + iy) * W + ix) * C + c]);
+        best = v > best ? v : best;
+      }
+    }
+    y[i] = f2bf(best);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// u8 NCHW [N][3][H][W] -> bf16 NHWC [N][H][W][4] with /255 and zero pad
+// channel 3 (CIN=4 feeds the conv1 kernel's scalar-gather path).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void u8_to_nhwc4(
+    const unsigned char* __restrict__ x, bf16_t* __restrict__ y,
+    int N, int H, int W) {
+  const size_t total = (size_t)N * H * W;
+  for (size_t p = (size_t)blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += (size_t)gridDim.x * blockDim.x) {
+    const int px = p % W;
+    const int py = (p / W) % H;
+    const int img = p / ((size_t)W * H);
+    const size_t cstr = (size_t)H * W;
+    const unsigned char* src = x + (size_t)img * 3 * cstr + py * W + px;
+    bf16_t* dst = y + p * 4;
+    dst[0] = f2bf(src[0] * (1.f / 255.f));
+    dst[1] = f2bf(src[cstr] * (1.f / 255.f));
+    dst[2] = f2bf(src[2 * cstr] * (1.f / 255.f));
+    dst[3] = f2bf(0.f);
+  }
+}
+
+// fp32 NCHW -> bf16 NHWC and back (bridges to the torch NCHW world).
+__global__ __launch_bounds__(256) void nchw_f32_to_nhwc_bf16(
+    const float* __restrict__ x, bf16_t* __restrict__ y,
+    int N, int C, int H, int W) {
+  const size_t total = (size_t)N * C * H * W;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (size_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    const size_t p = i / C;       // n*H*W + y*W + x
+    const int px = p % W;
+    const int py = (p / W) % H;
+    const int img = p / ((size_t)W * H);
+    y[i] = f2bf(x[(((size_t)img * C + c) * H + py) * W + px]);
+  }
+}
+
+__global__ __launch_bounds__(256) void nhwc_bf16_to_nchw_f32(
+    const bf16_t* __restrict__ x, float* __restrict__ y,
+    int N, int C, int H, int W) {
+  const size_t total = (size_t)N * C * H * W;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (size_t)gridDim.x * blockDim.x) {
+    const int px = i % W;
+    const int py = (i / W) % H;
+    const size_t rest = i / ((size_t)W * H);
+    const int c = rest % C;
+    const int img = rest / C;
+    y[i] = bf2f(x[(((size_t)img * H + py) * W + px) * C + c]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Layout probe: D[16][16] = A[16][32] @ W[16][32]^T on one MFMA, with the
+// assumed fragment mapping.  deep_probe.cpp diffs it against a CPU loop
+// BEFORE trusting any conv kernel (guide §3: asymmetric operands).
+// ---------------------------------------------------------------------------
+__global__ void mfma_probe(const bf16_t* __restrict__ A,
+                           const bf16_t* __restrict__ Wt,
+                           float* __restrict__ D) {
+  const int lane = threadIdx.x & 63;
+  bf16x8 a = *reinterpret_cast<const bf16x8*>(
+      A + (lane & 15) * 32 + (lane >> 4) * 8);
+  bf16x8 bv = *reinterpret_cast<const bf16x8*>(
+      Wt + (lane & 15) * 32 + (lane >> 4) * 8);
+  floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv, acc, 0, 0, 0);
+  for (int r = 0; r < 4; ++r)
+    D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+// ---------------------------------------------------------------------------
+// C-linkage launchers
+// ---------------------------------------------------------------------------
+static inline int conv_grid(int N, int H, int W) {
+  const int ntx = (W + TX - 1) / TX, nty = (H + TY - 1) / TY;
+  return N * ntx * nty;
+}
+
+#define CONV_LAUNCH(CIN, COUT)                                               \
+  if (relu_in) {                                                             \
+    if (res)                                                                 \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, true>), grid, blk,   \
+                         0, s, xb, wb, b, rb, yb, N, H, W);                  \
+    else                                                                     \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, false>), grid, blk,  \
+                         0, s, xb, wb, b, nullptr, yb, N, H, W);             \
+  } else {                                                                   \
+    if (res)                                                                 \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, true>), grid, blk,  \
+                         0, s, xb, wb, b, rb, yb, N, H, W);                  \
+    else                                                                     \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, false>), grid, blk, \
+                         0, s, xb, wb, b, nullptr, yb, N, H, W);             \
+  }
+
+extern "C" int launch_deep_conv3x3(const void* x, const void* w,
+                                   const float* b, const void* res, void* y,
+                                   int N, int H, int W, int cin, int cout,
+                                   int relu_in, hipStream_t s) {
+  const bf16_t* xb = static_cast<const bf16_t*>(x);
+  const bf16_t* wb = static_cast<const bf16_t*>(w);
+  const bf16_t* rb = static_cast<const bf16_t*>(res);
+  bf16_t* yb = static_cast<bf16_t*>(y);
+  dim3 grid(conv_grid(N, H, W)), blk(THREADS);
+  if (cin == 4 && cout == 16) { CONV_LAUNCH(4, 16) }
+  else if (cin == 16 && cout == 16) { CONV_LAUNCH(16, 16) }
+  else if (cin == 16 && cout == 32) { CONV_LAUNCH(16, 32) }
+  else if (cin == 32 && cout == 32) { CONV_LAUNCH(32, 32) }
+  else return -1;
+  return 0;
+}
+
+extern "C" void launch_deep_maxpool(const void* x, void* y, int N, int H,
+                                    int W, int C, hipStream_t s) {
+  const size_t total = (size_t)N * ((H + 1) / 2) * ((W + 1) / 2) * C;
+  int blocks = (int)((total + 255) / 256);
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL(maxpool3x3s2_nhwc, dim3(blocks), dim3(256), 0, s,
+                     static_cast<const bf16_t*>(x), static_cast<bf16_t*>(y),
+                     N, H, W, C);
+}
+
+extern "C" void launch_deep_u8_to_nhwc4(const void* x, void* y, int N, int H,
+                                        int W, hipStream_t s) {
+  const size_t total = (size_t)N * H * W;
+  int blocks = (int)((total + 255) / 256);
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL(u8_to_nhwc4, dim3(blocks), dim3(256), 0, s,
+                     static_cast<const unsigned char*>(x),
+                     static_cast<bf16_t*>(y), N, H, W);
+}
+
+extern "C" void launch_nchw_f32_to_nhwc_bf16(const void* x, void* y, int N,
+                                             int C, int H, int W,
+                                             hipStream_t s) {
+  const size_t total = (size_t)N * C * H * W;
+  int blocks = (int)((total + 255) / 256);
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL(nchw_f32_to_nhwc_bf16, dim3(blocks), dim3(256), 0, s,
+                     static_cast<const float*>(x), static_cast<bf16_t*>(y),
+                     N, C, H, W);
+}
+
+extern "C" void launch_nhwc_bf16_to_nchw_f32(const void* x, void* y, int N,
+                                             int C, int H, int W,
+                                             hipStream_t s) {
+  const size_t total = (size_t)N * C * H * W;
+  int blocks = (int)((total + 255) / 256);
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL(nhwc_bf16_to_nchw_f32, dim3(blocks), dim3(256), 0, s,
+                     static_cast<const bf16_t*>(x), static_cast<float*>(y),
+                     N, C, H, W);
+}
+
+extern "C" void launch_mfma_probe(const void* A, const void* Wt, float* D,
+                                  hipStream_t s) {
+  hipLaunchKernelGGL(mfma_probe, dim3(1), dim3(64), 0, s,
+                     static_cast<const bf16_t*>(A),
+                     static_cast<const bf16_t*>(Wt), D);
+}
+
+}  // namespace impala_deep

@@ -1,0 +1,266 @@
+// Standalone (torch-free) GPU validation ladder for deep_kernels.hip —
+// runs in seconds on an MI355X box, no python import cost:
+//   1. mfma_probe: the assumed v_mfma_f32_16x16x32_bf16 A/B/C fragment
+//      mapping vs a CPU loop (asymmetric operands — guide §3).
+//   2. conv3x3 parity for every (CIN,COUT) instantiation, with/without
+//      fused input-ReLU and residual add, on uneven H/W (partial tiles).
+//   3. maxpool / layout-converter parity.
+// Exit 0 = all PASS. Build: hipcc --offload-arch=gfx950 deep_probe.cpp
+// deep_kernels.hip -o deep_probe  (build_ext.py --probe does this).
+#include <hip/hip_runtime.h>
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+
+extern "C" int launch_deep_conv3x3(const void*, const void*, const float*,
+                                   const void*, void*, int, int, int, int,
+                                   int, int, hipStream_t);
+extern "C" void launch_deep_maxpool(const void*, void*, int, int, int, int,
+                                    hipStream_t);
+extern "C" void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int,
+                                        hipStream_t);
+extern "C" void launch_mfma_probe(const void*, const void*, float*,
+                                  hipStream_t);
+
+#define CHECK(x)                                                   \
+  do {                                                             \
+    hipError_t e = (x);                                            \
+    if (e != hipSuccess) {                                         \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(e),      \
+             __FILE__, __LINE__);                                  \
+      exit(2);                                                     \
+    }                                                              \
+  } while (0)
+
+typedef unsigned short u16;
+
+static u16 f2bf(float f) {
+  unsigned int u;
+  __builtin_memcpy(&u, &f, 4);
+  u += 0x7fff + ((u >> 16) & 1);  // round-nearest-even
+  return (u16)(u >> 16);
+}
+static float bf2f(u16 b) {
+  unsigned int u = (unsigned int)b << 16;
+  float f;
+  __builtin_memcpy(&f, &u, 4);
+  return f;
+}
+static float frand(unsigned& s) {  // deterministic LCG in [-1,1)
+  s = s * 1664525u + 1013904223u;
+  return (float)(int)s / 2147483648.0f;
+}
+
+static int n_fail = 0;
+static void report(const char* name, float maxerr, float tol) {
+  const bool ok = maxerr <= tol && std::isfinite(maxerr);
+  printf("%-44s max|err| %.3e  tol %.1e  %s\n", name, maxerr, tol,
+         ok ? "PASS" : "FAIL");
+  if (!ok) ++n_fail;
+}
+
+// CPU conv3x3 pad1 reference, NHWC, bf16-quantized inputs, f32 accumulate.
+static void ref_conv(const std::vector<u16>& x, const std::vector<u16>& w,
+                     const std::vector<float>& b, const u16* res,
+                     std::vector<float>& y, int N, int H, int W, int cin,
+                     int cout, bool relu_in) {
+  for (int n = 0; n < N; ++n)
+    for (int oy = 0; oy < H; ++oy)
+      for (int ox = 0; ox < W; ++ox)
+        for (int f = 0; f < cout; ++f) {
+          float acc = b[f];
+          for (int ky = 0; ky < 3; ++ky)
+            for (int kx = 0; kx < 3; ++kx) {
+              const int iy = oy + ky - 1, ix = ox + kx - 1;
+              if (iy < 0 || iy >= H || ix < 0 || ix >= W) continue;
+              for (int c = 0; c < cin; ++c) {
+                float v = bf2f(x[(((size_t)n * H + iy) * W + ix) * cin + c]);
+                if (relu_in && v < 0.f) v = 0.f;
+                // weights packed [cout][ky][kx][cin]
+                acc += v * bf2f(w[(((size_t)f * 3 + ky) * 3 + kx) * cin + c]);
+              }
+            }
+          const size_t off = (((size_t)n * H + oy) * W + ox) * cout + f;
+          if (res) acc += bf2f(res[off]);
+          y[off] = acc;
+        }
+}
+
+static float conv_case(int N, int H, int W, int cin, int cout, bool relu_in,
+                       bool has_res) {
+  unsigned seed = 12345u + cin * 7 + cout + H;
+  std::vector<u16> x((size_t)N * H * W * cin), w((size_t)cout * 9 * cin),
+      r((size_t)N * H * W * cout);
+  std::vector<float> b(cout);
+  for (auto& v : x) v = f2bf(frand(seed));
+  for (auto& v : w) v = f2bf(frand(seed) * 0.25f);
+  for (auto& v : r) v = f2bf(frand(seed));
+  for (auto& v : b) v = frand(seed);
+
+  std::vector<float> ref((size_t)N * H * W * cout);
+  ref_conv(x, w, b, has_res ? r.data() : nullptr, ref, N, H, W, cin, cout,
+           relu_in);
+
+  void *dx, *dw, *db, *dr, *dy;
+  CHECK(hipMalloc(&dx, x.size() * 2));
+  CHECK(hipMalloc(&dw, w.size() * 2));
+  CHECK(hipMalloc(&db, b.size() * 4));
+  CHECK(hipMalloc(&dr, r.size() * 2));
+  CHECK(hipMalloc(&dy, ref.size() * 2));
+  CHECK(hipMemcpy(dx, x.data(), x.size() * 2, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(dw, w.data(), w.size() * 2, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(db, b.data(), b.size() * 4, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(dr, r.data(), r.size() * 2, hipMemcpyHostToDevice));
+  CHECK(hipMemset(dy, 0, ref.size() * 2));
+  int rc = launch_deep_conv3x3(dx, dw, (const float*)db,
+                               has_res ? dr : nullptr, dy, N, H, W, cin,
+                               cout, relu_in ? 1 : 0, 0);
+  if (rc != 0) {
+    printf("unsupported cin/cout %d/%d\n", cin, cout);
+    exit(2);
+  }
+  CHECK(hipDeviceSynchronize());
+  std::vector<u16> out(ref.size());
+  CHECK(hipMemcpy(out.data(), dy, out.size() * 2, hipMemcpyDeviceToHost));
+  float maxerr = 0.f, maxmag = 1.f;
+  for (size_t i = 0; i < ref.size(); ++i) {
+    maxmag = std::max(maxmag, std::fabs(ref[i]));
+    maxerr = std::max(maxerr, std::fabs(bf2f(out[i]) - ref[i]));
+  }
+  hipFree(dx); hipFree(dw); hipFree(db); hipFree(dr); hipFree(dy);
+  return maxerr / maxmag;  // relative to the largest output
+}
+
+int main() {
+  int ndev = 0;
+  CHECK(hipGetDeviceCount(&ndev));
+  hipDeviceProp_t prop;
+  CHECK(hipGetDeviceProperties(&prop, 0));
+  printf("device: %s (%s)\n", prop.name, prop.gcnArchName);
+
+  // ---- 1. raw MFMA fragment-layout probe (asymmetric A and B) ----------
+  {
+    unsigned seed = 7u;
+    std::vector<u16> A(16 * 32), Wt(16 * 32);
+    for (int m = 0; m < 16; ++m)
+      for (int k = 0; k < 32; ++k) A[m * 32 + k] = f2bf(frand(seed));
+    for (int n = 0; n < 16; ++n)
+      for (int k = 0; k < 32; ++k) Wt[n * 32 + k] = f2bf(frand(seed) * 0.5f);
+    std::vector<float> ref(256, 0.f);
+    for (int m = 0; m < 16; ++m)
+      for (int n = 0; n < 16; ++n) {
+        float acc = 0.f;
+        for (int k = 0; k < 32; ++k)
+          acc += bf2f(A[m * 32 + k]) * bf2f(Wt[n * 32 + k]);
+        ref[m * 16 + n] = acc;
+      }
+    void *dA, *dW, *dD;
+    CHECK(hipMalloc(&dA, A.size() * 2));
+    CHECK(hipMalloc(&dW, Wt.size() * 2));
+    CHECK(hipMalloc(&dD, 256 * 4));
+    CHECK(hipMemcpy(dA, A.data(), A.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dW, Wt.data(), Wt.size() * 2, hipMemcpyHostToDevice));
+    launch_mfma_probe(dA, dW, (float*)dD, 0);
+    CHECK(hipDeviceSynchronize());
+    std::vector<float> D(256);
+    CHECK(hipMemcpy(D.data(), dD, 256 * 4, hipMemcpyDeviceToHost));
+    float maxerr = 0.f;
+    for (int i = 0; i < 256; ++i)
+      maxerr = std::max(maxerr, std::fabs(D[i] - ref[i]));
+    report("mfma_16x16x32_bf16 fragment layout", maxerr, 1e-1f);
+    if (n_fail) {  // everything else depends on this — dump a corner
+      printf("  D[0..3]=%g %g %g %g  ref=%g %g %g %g\n", D[0], D[1], D[2],
+             D[3], ref[0], ref[1], ref[2], ref[3]);
+      printf("  D[16]=%g ref[16]=%g (row1: row-major check)\n", D[16],
+             ref[16]);
+      return 1;
+    }
+    hipFree(dA); hipFree(dW); hipFree(dD);
+  }
+
+  // ---- 2. conv parity, every instantiation, partial tiles --------------
+  // DeepResNet shapes: 36x48 (sec1 res), 18x24 (sec2), 9x12 (sec3).
+  report("conv3x3  4->16 72x96 (sec1 head)",
+         conv_case(2, 72, 96, 4, 16, false, false), 2e-2f);
+  report("conv3x3 16->16 36x48 relu_in+res (res blk)",
+         conv_case(2, 36, 48, 16, 16, true, true), 2e-2f);
+  report("conv3x3 16->16 36x48 relu_in (res mid)",
+         conv_case(2, 36, 48, 16, 16, true, false), 2e-2f);
+  report("conv3x3 16->32 36x48 (sec2 head)",
+         conv_case(2, 36, 48, 16, 32, false, false), 2e-2f);
+  report("conv3x3 32->32 18x24 relu_in+res",
+         conv_case(2, 18, 24, 32, 32, true, true), 2e-2f);
+  report("conv3x3 32->32  9x12 relu_in+res (sec3)",
+         conv_case(3, 9, 12, 32, 32, true, true), 2e-2f);
+
+  // ---- 3. maxpool 3x3 s2 p1 -------------------------------------------
+  {
+    const int N = 2, H = 36, W = 48, C = 16, HO = 18, WO = 24;
+    unsigned seed = 99u;
+    std::vector<u16> x((size_t)N * H * W * C);
+    for (auto& v : x) v = f2bf(frand(seed));
+    std::vector<float> ref((size_t)N * HO * WO * C, -3.4e38f);
+    for (int n = 0; n < N; ++n)
+      for (int oy = 0; oy < HO; ++oy)
+        for (int ox = 0; ox < WO; ++ox)
+          for (int c = 0; c < C; ++c) {
+            float best = -3.4e38f;
+            for (int dy = 0; dy < 3; ++dy)
+              for (int dx = 0; dx < 3; ++dx) {
+                int iy = oy * 2 - 1 + dy, ix = ox * 2 - 1 + dx;
+                if (iy < 0 || iy >= H || ix < 0 || ix >= W) continue;
+                best = std::max(
+                    best, bf2f(x[(((size_t)n * H + iy) * W + ix) * C + c]));
+              }
+            ref[(((size_t)n * HO + oy) * WO + ox) * C + c] = best;
+          }
+    void *dx, *dy;
+    CHECK(hipMalloc(&dx, x.size() * 2));
+    CHECK(hipMalloc(&dy, ref.size() * 2));
+    CHECK(hipMemcpy(dx, x.data(), x.size() * 2, hipMemcpyHostToDevice));
+    launch_deep_maxpool(dx, dy, N, H, W, C, 0);
+    CHECK(hipDeviceSynchronize());
+    std::vector<u16> out(ref.size());
+    CHECK(hipMemcpy(out.data(), dy, out.size() * 2, hipMemcpyDeviceToHost));
+    float maxerr = 0.f;
+    for (size_t i = 0; i < ref.size(); ++i)
+      maxerr = std::max(maxerr, std::fabs(bf2f(out[i]) - ref[i]));
+    report("maxpool3x3s2p1 36x48->18x24", maxerr, 0.f);
+    hipFree(dx); hipFree(dy);
+  }
+
+  // ---- 4. u8 NCHW -> bf16 NHWC4 ----------------------------------------
+  {
+    const int N = 2, H = 8, W = 9;
+    std::vector<unsigned char> x((size_t)N * 3 * H * W);
+    unsigned seed = 5u;
+    for (auto& v : x) { seed = seed * 1664525u + 1013904223u; v = seed >> 24; }
+    void *dx, *dy;
+    CHECK(hipMalloc(&dx, x.size()));
+    CHECK(hipMalloc(&dy, (size_t)N * H * W * 4 * 2));
+    CHECK(hipMemcpy(dx, x.data(), x.size(), hipMemcpyHostToDevice));
+    launch_deep_u8_to_nhwc4(dx, dy, N, H, W, 0);
+    CHECK(hipDeviceSynchronize());
+    std::vector<u16> out((size_t)N * H * W * 4);
+    CHECK(hipMemcpy(out.data(), dy, out.size() * 2, hipMemcpyDeviceToHost));
+    float maxerr = 0.f;
+    for (int n = 0; n < N; ++n)
+      for (int py = 0; py < H; ++py)
+        for (int px = 0; px < W; ++px)
+          for (int c = 0; c < 4; ++c) {
+            float want =
+                c < 3 ? bf2f(f2bf(x[((size_t)(n * 3 + c) * H + py) * W + px] /
+                                  255.0f))
+                      : 0.f;
+            float got =
+                bf2f(out[(((size_t)n * H + py) * W + px) * 4 + c]);
+            maxerr = std::max(maxerr, std::fabs(got - want));
+          }
+    report("u8 nchw -> bf16 nhwc4 (/255)", maxerr, 1e-6f);
+    hipFree(dx); hipFree(dy);
+  }
+
+  printf(n_fail ? "DEEP_PROBE FAIL (%d)\n" : "DEEP_PROBE ALL PASS\n", n_fail);
+  return n_fail ? 1 : 0;
+}

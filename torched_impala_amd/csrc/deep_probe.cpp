@@ -132,7 +132,59 @@ static float conv_case(int N, int H, int W, int cin, int cout, bool relu_in,
   return maxerr / maxmag;  // relative to the largest output
 }
 
-int main() {
+// Throughput bench: one conv shape at production-ish N, hipEvent-timed.
+static void bench_case(const char* name, int N, int H, int W, int cin,
+                       int cout, bool relu_in, bool has_res) {
+  void *dx, *dw, *db, *dr, *dy;
+  CHECK(hipMalloc(&dx, (size_t)N * H * W * cin * 2));
+  CHECK(hipMalloc(&dw, (size_t)cout * 9 * cin * 2));
+  CHECK(hipMalloc(&db, cout * 4));
+  CHECK(hipMalloc(&dr, (size_t)N * H * W * cout * 2));
+  CHECK(hipMalloc(&dy, (size_t)N * H * W * cout * 2));
+  CHECK(hipMemset(dx, 0x3c, (size_t)N * H * W * cin * 2));   // ~0.018 bf16
+  CHECK(hipMemset(dw, 0x3c, (size_t)cout * 9 * cin * 2));
+  CHECK(hipMemset(db, 0, cout * 4));
+  CHECK(hipMemset(dr, 0x3c, (size_t)N * H * W * cout * 2));
+  hipEvent_t e0, e1;
+  CHECK(hipEventCreate(&e0));
+  CHECK(hipEventCreate(&e1));
+  const int iters = 20;
+  for (int i = 0; i < 3; ++i)
+    launch_deep_conv3x3(dx, dw, (const float*)db, has_res ? dr : nullptr,
+                        dy, N, H, W, cin, cout, relu_in, 0);
+  CHECK(hipEventRecord(e0, 0));
+  for (int i = 0; i < iters; ++i)
+    launch_deep_conv3x3(dx, dw, (const float*)db, has_res ? dr : nullptr,
+                        dy, N, H, W, cin, cout, relu_in, 0);
+  CHECK(hipEventRecord(e1, 0));
+  CHECK(hipDeviceSynchronize());
+  float ms = 0.f;
+  CHECK(hipEventElapsedTime(&ms, e0, e1));
+  ms /= iters;
+  const double flops = 2.0 * N * H * W * (double)cout * 9.0 * cin;
+  const double bytes = 2.0 * N * H * W *
+                       (cin + (double)cout * (has_res ? 2 : 1));
+  printf("%-40s N=%-6d %8.3f ms  %7.1f TF  %6.1f GB/s\n", name, N, ms,
+         flops / ms / 1e9, bytes / ms / 1e6);
+  hipFree(dx); hipFree(dw); hipFree(db); hipFree(dr); hipFree(dy);
+  CHECK(hipEventDestroy(e0));
+  CHECK(hipEventDestroy(e1));
+}
+
+int main(int argc, char** argv) {
+  if (argc > 1 && argv[1][0] == 'b') {
+    hipDeviceProp_t prop;
+    CHECK(hipGetDeviceProperties(&prop, 0));
+    printf("bench on %s — per-conv times at N frames (BASELINE cfg5 step "
+           "N=51712; scale accordingly)\n", prop.gcnArchName);
+    const int N = 12928;  // quarter step: fits fast alloc, same regime
+    bench_case("conv  4->16 72x96 (sec1 head)", N, 72, 96, 4, 16, 0, 0);
+    bench_case("conv 16->16 36x48 relu+res", N, 36, 48, 16, 16, 1, 1);
+    bench_case("conv 16->32 36x48 (sec2 head)", N, 36, 48, 16, 32, 0, 0);
+    bench_case("conv 32->32 18x24 relu+res", N, 18, 24, 32, 32, 1, 1);
+    bench_case("conv 32->32  9x12 relu+res", N, 9, 12, 32, 32, 1, 1);
+    return 0;
+  }
   int ndev = 0;
   CHECK(hipGetDeviceCount(&ndev));
   hipDeviceProp_t prop;

@@ -76,7 +76,21 @@ def build_actor(force: bool = False) -> str:
 
 
 HIP_SOURCES = ["hip_ext.cpp", "vtrace_kernels.hip", "lstm_kernels.hip",
-               "optim_kernels.hip", "conv_kernels.hip", "linear_kernels.hip"]
+               "optim_kernels.hip", "conv_kernels.hip", "linear_kernels.hip",
+               "deep_kernels.hip"]
+
+
+def build_probe(force: bool = False) -> str:
+    """Standalone torch-free GPU validator/bench (deep_probe.cpp) — runs in
+    ~1 s of box time, used for kernel bring-up without the torch import
+    cost."""
+    out = os.path.join(CSRC, "deep_probe")
+    sources = [os.path.join(CSRC, "deep_probe.cpp"),
+               os.path.join(CSRC, "deep_kernels.hip")]
+    if force or _needs_build(out, sources):
+        _run(["hipcc", "--offload-arch=gfx950", "-O2", "-Wno-unused-value",
+              "-x", "hip"] + sources + ["-o", out])
+    return out
 
 
 def build_hip(force: bool = False) -> str:
@@ -106,6 +120,7 @@ def build_all(force: bool = False):
         build_hip(force)
     except FileNotFoundError:
         print("[build_ext] HIP sources not present yet; skipped")
+    build_probe(force)
 
 
 if __name__ == "__main__":

@@ -54,6 +54,15 @@ int shallow_encoder_slab_elems();
 void launch_linear_mfma(const float*, const float*, const float*,
                         const float*, int, int, int, int, float*,
                         hipStream_t);
+int launch_deep_conv3x3(const void*, const void*, const float*, const void*,
+                        void*, int, int, int, int, int, int, hipStream_t);
+void launch_deep_maxpool(const void*, void*, int, int, int, int,
+                         hipStream_t);
+void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int, hipStream_t);
+void launch_nchw_f32_to_nhwc_bf16(const void*, void*, int, int, int, int,
+                                  hipStream_t);
+void launch_nhwc_bf16_to_nchw_f32(const void*, void*, int, int, int, int,
+                                  hipStream_t);
 }
 
 namespace {
@@ -387,6 +396,92 @@ torch::Tensor linear_mfma(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
+// ---------------------------------------------------------------------------
+// deep ResNet NHWC bf16 kernels (deep_kernels.hip; validated by deep_probe)
+// ---------------------------------------------------------------------------
+inline void check_bf16_nhwc(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// x [N][H][W][CIN] bf16, w [COUT][3][3][CIN] bf16, b [COUT] f32,
+// res optional [N][H][W][COUT] bf16 -> y [N][H][W][COUT] bf16
+torch::Tensor deep_conv3x3(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                           c10::optional<torch::Tensor> res, bool relu_in) {
+  check_bf16_nhwc(x, "x");
+  check_bf16_nhwc(w, "w");
+  check_f32(b, "b");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), cin = x.size(3);
+  const int cout = w.size(0);
+  TORCH_CHECK(w.dim() == 4 && w.size(1) == 3 && w.size(2) == 3 &&
+                  w.size(3) == cin, "w must be [cout][3][3][cin]");
+  TORCH_CHECK(b.size(0) == cout, "bias size");
+  DeviceScope guard(x.get_device());
+  auto y = torch::empty({N, H, W, cout}, x.options());
+  const void* res_p = nullptr;
+  if (res.has_value()) {
+    check_bf16_nhwc(*res, "res");
+    TORCH_CHECK(res->size(3) == cout, "res channels");
+    res_p = res->data_ptr();
+  }
+  int rc = launch_deep_conv3x3(x.data_ptr(), w.data_ptr(),
+                               b.data_ptr<float>(), res_p, y.data_ptr(), N,
+                               H, W, cin, cout, relu_in ? 1 : 0,
+                               current_stream());
+  TORCH_CHECK(rc == 0, "unsupported cin/cout ", cin, "/", cout);
+  HIP_CHECK(hipGetLastError());
+  return y;
+}
+
+torch::Tensor deep_maxpool(torch::Tensor x) {
+  check_bf16_nhwc(x, "x");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  DeviceScope guard(x.get_device());
+  auto y = torch::empty({N, (H + 1) / 2, (W + 1) / 2, C}, x.options());
+  launch_deep_maxpool(x.data_ptr(), y.data_ptr(), N, H, W, C,
+                      current_stream());
+  HIP_CHECK(hipGetLastError());
+  return y;
+}
+
+// u8 [N][3][H][W] -> bf16 [N][H][W][4] with /255 and a zero pad channel
+torch::Tensor deep_u8_to_nhwc4(torch::Tensor frames) {
+  TORCH_CHECK(frames.is_cuda() && frames.scalar_type() == torch::kUInt8 &&
+                  frames.is_contiguous() && frames.size(1) == 3,
+              "frames must be cuda u8 [N][3][H][W] contiguous");
+  const int N = frames.size(0), H = frames.size(2), W = frames.size(3);
+  DeviceScope guard(frames.get_device());
+  auto y = torch::empty({N, H, W, 4},
+                        frames.options().dtype(torch::kBFloat16));
+  launch_deep_u8_to_nhwc4(frames.data_ptr(), y.data_ptr(), N, H, W,
+                          current_stream());
+  HIP_CHECK(hipGetLastError());
+  return y;
+}
+
+torch::Tensor nchw_to_nhwc_bf16(torch::Tensor x) {
+  check_f32(x, "x");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  DeviceScope guard(x.get_device());
+  auto y = torch::empty({N, H, W, C}, x.options().dtype(torch::kBFloat16));
+  launch_nchw_f32_to_nhwc_bf16(x.data_ptr(), y.data_ptr(), N, C, H, W,
+                               current_stream());
+  HIP_CHECK(hipGetLastError());
+  return y;
+}
+
+torch::Tensor nhwc_bf16_to_nchw(torch::Tensor x) {
+  check_bf16_nhwc(x, "x");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  DeviceScope guard(x.get_device());
+  auto y = torch::empty({N, C, H, W}, x.options().dtype(torch::kFloat32));
+  launch_nhwc_bf16_to_nchw_f32(x.data_ptr(), y.data_ptr(), N, C, H, W,
+                               current_stream());
+  HIP_CHECK(hipGetLastError());
+  return y;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -404,4 +499,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_mfma", &linear_mfma, py::arg("A"), py::arg("B"),
         py::arg("bias") = py::none(), py::arg("ymask") = py::none(),
         py::arg("mode") = 0);
+  m.def("deep_conv3x3", &deep_conv3x3, py::arg("x"), py::arg("w"),
+        py::arg("b"), py::arg("res") = py::none(),
+        py::arg("relu_in") = false);
+  m.def("deep_maxpool", &deep_maxpool);
+  m.def("deep_u8_to_nhwc4", &deep_u8_to_nhwc4);
+  m.def("nchw_to_nhwc_bf16", &nchw_to_nhwc_bf16);
+  m.def("nhwc_bf16_to_nchw", &nhwc_bf16_to_nchw);
 }

@@ -190,12 +190,245 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
 }
 
 // ---------------------------------------------------------------------------
+// conv3x3 backward-to-input: dX = conv3x3(dY, W~) with
+// W~[ci][ky][kx][f] = W[f][2-ky][2-kx][ci] (repacked host-side — the
+// forward kernel template is reused verbatim for the GEMM), optionally
+// epilogue-masked by sign(x_fwd) (fused ReLU backward).  This variant
+// only adds the mask read; cin/cout roles are swapped by the caller.
+// ---------------------------------------------------------------------------
+template <int CIN, int COUT>
+__global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
+    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ wt,
+    const bf16_t* __restrict__ xmask, bf16_t* __restrict__ dx, int N, int H,
+    int W) {
+  constexpr int KTOT = 9 * CIN;
+  constexpr int KSTEPS = (KTOT + 31) / 32;
+  constexpr int NFRAG = COUT / 16;
+  constexpr int PSTR = (TX + 2) * CIN;
+  constexpr int WSTR = 9 * CIN;
+
+  __shared__ bf16_t s_patch[(TY + 2) * PSTR];
+  __shared__ bf16_t s_w[COUT * WSTR];
+
+  const int ntx = (W + TX - 1) / TX;
+  const int nty = (H + TY - 1) / TY;
+  const int tile = blockIdx.x % (ntx * nty);
+  const int img = blockIdx.x / (ntx * nty);
+  if (img >= N) return;
+  const int ty0 = (tile / ntx) * TY;
+  const int tx0 = (tile % ntx) * TX;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  {
+    const unsigned short* ws = reinterpret_cast<const unsigned short*>(wt);
+    unsigned short* wd = reinterpret_cast<unsigned short*>(s_w);
+    for (int i = tid; i < COUT * WSTR; i += THREADS) wd[i] = ws[i];
+  }
+  {
+    const unsigned short* xs = reinterpret_cast<const unsigned short*>(dy) +
+                               (size_t)img * H * W * CIN;
+    unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
+    for (int p = tid; p < (TY + 2) * (TX + 2); p += THREADS) {
+      const int py = p / (TX + 2), px = p % (TX + 2);
+      const int iy = ty0 + py - 1, ix = tx0 + px - 1;
+      unsigned short* dst = pd + py * PSTR + px * CIN;
+      if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+        const unsigned short* src = xs + ((size_t)iy * W + ix) * CIN;
+        #pragma unroll
+        for (int c = 0; c < CIN; ++c) dst[c] = src[c];
+      } else {
+        #pragma unroll
+        for (int c = 0; c < CIN; ++c) dst[c] = 0;
+      }
+    }
+  }
+  __syncthreads();
+
+  floatx4 acc[4][NFRAG];
+  #pragma unroll
+  for (int m = 0; m < 4; ++m)
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) acc[m][n] = floatx4{0.f, 0.f, 0.f, 0.f};
+  const int arow = lane & 15;
+  const int kbase = (lane >> 4) * 8;
+  #pragma unroll
+  for (int s = 0; s < KSTEPS; ++s) {
+    const int k0 = s * 32 + kbase;
+    bf16x8 bf[NFRAG];
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) {
+      const int col = n * 16 + (lane & 15);
+      bf16x8 v = {};
+      if (k0 < KTOT) v = *reinterpret_cast<const bf16x8*>(
+          s_w + col * WSTR + k0);
+      bf[n] = v;
+    }
+    #pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const int p = (wave * 4 + m) * 16 + arow;
+      const int oy = p >> 4, ox = p & 15;
+      bf16x8 af = {};
+      if (k0 < KTOT) {
+        const int tap = k0 / CIN, c0 = k0 % CIN;
+        af = *reinterpret_cast<const bf16x8*>(
+            s_patch + (oy + tap / 3) * PSTR + (ox + tap % 3) * CIN + c0);
+      }
+      #pragma unroll
+      for (int n = 0; n < NFRAG; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af, bf[n], acc[m][n], 0, 0, 0);
+    }
+  }
+
+  const int ccol = lane & 15;
+  const int crow = (lane >> 4) * 4;
+  #pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) {
+      const int cout = n * 16 + ccol;
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int p = (wave * 4 + m) * 16 + crow + reg;
+        const int oy = ty0 + (p >> 4), ox = tx0 + (p & 15);
+        if (oy < H && ox < W) {
+          const size_t off = (((size_t)img * H + oy) * W + ox) * COUT + cout;
+          float v = acc[m][n][reg];
+          if (xmask != nullptr && bf2f(xmask[off]) <= 0.f) v = 0.f;
+          dx[off] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv3x3 backward-to-weights: dW[f][ky][kx][c] += sum over px of
+// act(x)[px + tap] * dY[px].  Per WG: one 16x16 px tile; MFMA with
+// M = 9*CIN (taps x cin), N = COUT, K = 256 px; fp32 atomicAdd into the
+// global dW (fp32 rounding-order nondeterminism only — ROADMAP notes a
+// slab+reduce variant for bitwise determinism).  A/B gathers are
+// k(px)-strided scalar reads — correctness-first; ds_read_b64_tr_b16 is
+// the round-2 lever.
+// ---------------------------------------------------------------------------
+template <int CIN, int COUT, bool RELU_IN>
+__global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
+    float* __restrict__ dw, int N, int H, int W) {
+  constexpr int MTOT = 9 * CIN;                  // 144 or 288
+  constexpr int MFRAGS = (MTOT + 15) / 16;       // 9 or 18
+  constexpr int NFRAG = COUT / 16;
+  constexpr int PSTR = (TX + 2) * CIN;
+
+  __shared__ bf16_t s_patch[(TY + 2) * PSTR];
+  __shared__ bf16_t s_dy[TY * TX * COUT];
+
+  const int ntx = (W + TX - 1) / TX;
+  const int nty = (H + TY - 1) / TY;
+  const int tile = blockIdx.x % (ntx * nty);
+  const int img = blockIdx.x / (ntx * nty);
+  if (img >= N) return;
+  const int ty0 = (tile / ntx) * TY;
+  const int tx0 = (tile % ntx) * TX;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  {  // x patch with halo (+ReLU)
+    const unsigned short* xs = reinterpret_cast<const unsigned short*>(x) +
+                               (size_t)img * H * W * CIN;
+    unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
+    for (int p = tid; p < (TY + 2) * (TX + 2); p += THREADS) {
+      const int py = p / (TX + 2), px = p % (TX + 2);
+      const int iy = ty0 + py - 1, ix = tx0 + px - 1;
+      unsigned short* dst = pd + py * PSTR + px * CIN;
+      if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+        const unsigned short* src = xs + ((size_t)iy * W + ix) * CIN;
+        #pragma unroll
+        for (int c = 0; c < CIN; ++c)
+          dst[c] = RELU_IN ? relu_bits(src[c]) : src[c];
+      } else {
+        #pragma unroll
+        for (int c = 0; c < CIN; ++c) dst[c] = 0;
+      }
+    }
+  }
+  {  // dY tile [px][cout], zero where out of range
+    const unsigned short* ds = reinterpret_cast<const unsigned short*>(dy) +
+                               (size_t)img * H * W * COUT;
+    unsigned short* dd = reinterpret_cast<unsigned short*>(s_dy);
+    for (int p = tid; p < TY * TX; p += THREADS) {
+      const int oy = ty0 + (p >> 4), ox = tx0 + (p & 15);
+      unsigned short* dst = dd + p * COUT;
+      if (oy < H && ox < W) {
+        const unsigned short* src = ds + ((size_t)oy * W + ox) * COUT;
+        #pragma unroll
+        for (int c = 0; c < COUT; ++c) dst[c] = src[c];
+      } else {
+        #pragma unroll
+        for (int c = 0; c < COUT; ++c) dst[c] = 0;
+      }
+    }
+  }
+  __syncthreads();
+
+  // each wave owns m-frags wave, wave+4, wave+8, ... (MFRAGS 9/18)
+  const int kbase = (lane >> 4) * 8;
+  for (int mf = wave; mf < MFRAGS; mf += 4) {
+    floatx4 acc[NFRAG];
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) acc[n] = floatx4{0.f, 0.f, 0.f, 0.f};
+    const int m0 = mf * 16 + (lane & 15);        // this lane's A row
+    const int tap = m0 / CIN, c = m0 % CIN;      // may be >=9*CIN (pad)
+    const int ky = tap / 3, kx = tap % 3;
+    #pragma unroll 2
+    for (int s = 0; s < (TY * TX) / 32; ++s) {   // 8 k-steps of 32 px
+      bf16x8 af = {}, bfv[NFRAG] = {};
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int px = s * 32 + kbase + j;
+        if (m0 < MTOT)
+          af[j] = s_patch[((px >> 4) + ky) * PSTR + ((px & 15) + kx) * CIN
+                          + c];
+      }
+      #pragma unroll
+      for (int n = 0; n < NFRAG; ++n) {
+        const int col = n * 16 + (lane & 15);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bfv[n][j] = s_dy[(s * 32 + kbase + j) * COUT + col];
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfv[n], acc[n],
+                                                         0, 0, 0);
+      }
+    }
+    // epilogue: D[m=9cin-row][n=cout] -> dw[f][ky][kx][c], atomic f32
+    const int dcol = lane & 15;
+    const int drow = (lane >> 4) * 4;
+    #pragma unroll
+    for (int n = 0; n < NFRAG; ++n) {
+      const int f = n * 16 + dcol;
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = mf * 16 + drow + reg;
+        if (m < MTOT) {
+          const int t = m / CIN, cc = m % CIN;
+          atomicAdd(&dw[(((size_t)f * 3 + t / 3) * 3 + t % 3) * CIN + cc],
+                    acc[n][reg]);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // maxpool 3x3 stride2 pad1, NHWC bf16: out (H+1)/2 x (W+1)/2.
 // One thread per (out px, channel). bf16 compare is exact via float.
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void maxpool3x3s2_nhwc(
     const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
-    int N, int H, int W, int C) {
+    unsigned char* __restrict__ argmax, int N, int H, int W, int C) {
   const int HO = (H + 1) / 2, WO = (W + 1) / 2;
   const size_t total = (size_t)N * HO * WO * C;
   for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -207,6 +440,7 @@ __global__ __launch_bounds__(256) void maxpool3x3s2_nhwc(
     const int img = p / ((size_t)WO * HO);
     const int iy0 = oy * 2 - 1, ix0 = ox * 2 - 1;
     float best = -3.4e38f;
+    int bidx = 0;
     for (int dy = 0; dy < 3; ++dy) {
       const int iy = iy0 + dy;
       if (iy < 0 || iy >= H) continue;
@@ -215,10 +449,39 @@ __global__ __launch_bounds__(256) void maxpool3x3s2_nhwc(
         if (ix < 0 || ix >= W) continue;
         const float v =
             bf2f(x[(((size_t)img * H + iy) * W + ix) * C + c]);
-        best = v > best ? v : best;
+        if (v > best) { best = v; bidx = dy * 3 + dx; }
       }
     }
     y[i] = f2bf(best);
+    if (argmax != nullptr) argmax[i] = (unsigned char)bidx;
+  }
+}
+
+// Pool backward — deterministic gather: each input px scans the <=4
+// output windows that cover it and claims dY where argmax points here.
+__global__ __launch_bounds__(256) void maxpool3x3s2_bwd_nhwc(
+    const bf16_t* __restrict__ dy, const unsigned char* __restrict__ argmax,
+    bf16_t* __restrict__ dx, int N, int H, int W, int C) {
+  const int HO = (H + 1) / 2, WO = (W + 1) / 2;
+  const size_t total = (size_t)N * H * W * C;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (size_t)gridDim.x * blockDim.x) {
+    const int c = i % C;
+    const size_t p = i / C;
+    const int ix = p % W;
+    const int iy = (p / W) % H;
+    const int img = p / ((size_t)W * H);
+    float g = 0.f;
+    // windows: oy*2-1 <= iy <= oy*2+1  =>  oy in [(iy-1+1)/2, (iy+1)/2]
+    const int oy_lo = (iy - 1 + 1) / 2, oy_hi = (iy + 1) / 2;
+    const int ox_lo = (ix - 1 + 1) / 2, ox_hi = (ix + 1) / 2;
+    for (int oy = oy_lo; oy <= oy_hi && oy < HO; ++oy)
+      for (int ox = ox_lo; ox <= ox_hi && ox < WO; ++ox) {
+        const size_t o = (((size_t)img * HO + oy) * WO + ox) * C + c;
+        const int want = (iy - (oy * 2 - 1)) * 3 + (ix - (ox * 2 - 1));
+        if (argmax[o] == (unsigned char)want) g += bf2f(dy[o]);
+      }
+    dx[i] = f2bf(g);
   }
 }
 
@@ -337,14 +600,78 @@ extern "C" int launch_deep_conv3x3(const void* x, const void* w,
   return 0;
 }
 
-extern "C" void launch_deep_maxpool(const void* x, void* y, int N, int H,
-                                    int W, int C, hipStream_t s) {
+extern "C" void launch_deep_maxpool(const void* x, void* y, void* argmax,
+                                    int N, int H, int W, int C,
+                                    hipStream_t s) {
   const size_t total = (size_t)N * ((H + 1) / 2) * ((W + 1) / 2) * C;
   int blocks = (int)((total + 255) / 256);
   if (blocks > 16384) blocks = 16384;
   hipLaunchKernelGGL(maxpool3x3s2_nhwc, dim3(blocks), dim3(256), 0, s,
                      static_cast<const bf16_t*>(x), static_cast<bf16_t*>(y),
-                     N, H, W, C);
+                     static_cast<unsigned char*>(argmax), N, H, W, C);
+}
+
+extern "C" void launch_deep_maxpool_bwd(const void* dy, const void* argmax,
+                                        void* dx, int N, int H, int W, int C,
+                                        hipStream_t s) {
+  const size_t total = (size_t)N * H * W * C;
+  int blocks = (int)((total + 255) / 256);
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL(maxpool3x3s2_bwd_nhwc, dim3(blocks), dim3(256), 0, s,
+                     static_cast<const bf16_t*>(dy),
+                     static_cast<const unsigned char*>(argmax),
+                     static_cast<bf16_t*>(dx), N, H, W, C);
+}
+
+extern "C" int launch_deep_conv3x3_dx(const void* dy, const void* wt,
+                                      const void* xmask, void* dx, int N,
+                                      int H, int W, int cout_fwd,
+                                      int cin_fwd, hipStream_t s) {
+  // roles swapped: the GEMM's CIN = forward cout, COUT = forward cin.
+  const bf16_t* d = static_cast<const bf16_t*>(dy);
+  const bf16_t* w = static_cast<const bf16_t*>(wt);
+  const bf16_t* m = static_cast<const bf16_t*>(xmask);
+  bf16_t* o = static_cast<bf16_t*>(dx);
+  dim3 grid(conv_grid(N, H, W)), blk(THREADS);
+  if (cout_fwd == 16 && cin_fwd == 16)
+    hipLaunchKernelGGL((conv3x3_dx_nhwc<16, 16>), grid, blk, 0, s, d, w, m,
+                       o, N, H, W);
+  else if (cout_fwd == 32 && cin_fwd == 16)
+    hipLaunchKernelGGL((conv3x3_dx_nhwc<32, 16>), grid, blk, 0, s, d, w, m,
+                       o, N, H, W);
+  else if (cout_fwd == 32 && cin_fwd == 32)
+    hipLaunchKernelGGL((conv3x3_dx_nhwc<32, 32>), grid, blk, 0, s, d, w, m,
+                       o, N, H, W);
+  else if (cout_fwd == 16 && cin_fwd == 4)
+    // conv1 dX (rarely needed: frames have no grad) — COUT=4 < 16 not
+    // representable; caller should skip conv1 dX entirely.
+    return -1;
+  else
+    return -1;
+  return 0;
+}
+
+extern "C" int launch_deep_conv3x3_dw(const void* x, const void* dy,
+                                      float* dw, int N, int H, int W,
+                                      int cin, int cout, int relu_in,
+                                      hipStream_t s) {
+  const bf16_t* xb = static_cast<const bf16_t*>(x);
+  const bf16_t* db = static_cast<const bf16_t*>(dy);
+  dim3 grid(conv_grid(N, H, W)), blk(THREADS);
+#define DW_CASE(CI, CO)                                                     \
+  if (relu_in)                                                              \
+    hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, true>), grid, blk, 0, s,    \
+                       xb, db, dw, N, H, W);                                \
+  else                                                                      \
+    hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, false>), grid, blk, 0, s,   \
+                       xb, db, dw, N, H, W);
+  if (cin == 4 && cout == 16) { DW_CASE(4, 16) }
+  else if (cin == 16 && cout == 16) { DW_CASE(16, 16) }
+  else if (cin == 16 && cout == 32) { DW_CASE(16, 32) }
+  else if (cin == 32 && cout == 32) { DW_CASE(32, 32) }
+  else return -1;
+#undef DW_CASE
+  return 0;
 }
 
 extern "C" void launch_deep_u8_to_nhwc4(const void* x, void* y, int N, int H,

@@ -16,8 +16,15 @@
 extern "C" int launch_deep_conv3x3(const void*, const void*, const float*,
                                    const void*, void*, int, int, int, int,
                                    int, int, hipStream_t);
-extern "C" void launch_deep_maxpool(const void*, void*, int, int, int, int,
-                                    hipStream_t);
+extern "C" void launch_deep_maxpool(const void*, void*, void*, int, int, int,
+                                    int, hipStream_t);
+extern "C" void launch_deep_maxpool_bwd(const void*, const void*, void*, int,
+                                        int, int, int, hipStream_t);
+extern "C" int launch_deep_conv3x3_dx(const void*, const void*, const void*,
+                                      void*, int, int, int, int, int,
+                                      hipStream_t);
+extern "C" int launch_deep_conv3x3_dw(const void*, const void*, float*, int,
+                                      int, int, int, int, int, hipStream_t);
 extern "C" void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int,
                                         hipStream_t);
 extern "C" void launch_mfma_probe(const void*, const void*, float*,
@@ -271,7 +278,7 @@ int main(int argc, char** argv) {
     CHECK(hipMalloc(&dx, x.size() * 2));
     CHECK(hipMalloc(&dy, ref.size() * 2));
     CHECK(hipMemcpy(dx, x.data(), x.size() * 2, hipMemcpyHostToDevice));
-    launch_deep_maxpool(dx, dy, N, H, W, C, 0);
+    launch_deep_maxpool(dx, dy, nullptr, N, H, W, C, 0);
     CHECK(hipDeviceSynchronize());
     std::vector<u16> out(ref.size());
     CHECK(hipMemcpy(out.data(), dy, out.size() * 2, hipMemcpyDeviceToHost));
@@ -311,6 +318,183 @@ int main(int argc, char** argv) {
           }
     report("u8 nchw -> bf16 nhwc4 (/255)", maxerr, 1e-6f);
     hipFree(dx); hipFree(dy);
+  }
+
+  // ---- 5. conv backward: dX (transposed conv + ReLU mask) --------------
+  // dX[ci] = sum_f sum_taps dY[y+1-ky][x+1-kx][f] * W[f][ky][kx][ci],
+  // masked by x>0.  Kernel input: repacked Wt[ci][ky'][kx'][f] with
+  // (ky',kx') = (2-ky, 2-kx) — same repack the python side does.
+  for (int cfg = 0; cfg < 3; ++cfg) {
+    const int cin = cfg == 2 ? 32 : 16, cout = cfg == 0 ? 16 : 32;
+    const int N = 2, H = cfg == 2 ? 18 : 36, W = cfg == 2 ? 24 : 48;
+    unsigned seed = 31u + cfg;
+    std::vector<u16> xv((size_t)N * H * W * cin), wv((size_t)cout * 9 * cin),
+        dyv((size_t)N * H * W * cout);
+    for (auto& v : xv) v = f2bf(frand(seed));
+    for (auto& v : wv) v = f2bf(frand(seed) * 0.25f);
+    for (auto& v : dyv) v = f2bf(frand(seed));
+    // CPU dX reference
+    std::vector<float> ref((size_t)N * H * W * cin, 0.f);
+    for (int n = 0; n < N; ++n)
+      for (int oy = 0; oy < H; ++oy)
+        for (int ox = 0; ox < W; ++ox)
+          for (int f = 0; f < cout; ++f) {
+            const float g =
+                bf2f(dyv[(((size_t)n * H + oy) * W + ox) * cout + f]);
+            for (int ky = 0; ky < 3; ++ky)
+              for (int kx = 0; kx < 3; ++kx) {
+                const int iy = oy + ky - 1, ix = ox + kx - 1;
+                if (iy < 0 || iy >= H || ix < 0 || ix >= W) continue;
+                for (int c = 0; c < cin; ++c)
+                  ref[(((size_t)n * H + iy) * W + ix) * cin + c] +=
+                      g * bf2f(wv[(((size_t)f * 3 + ky) * 3 + kx) * cin + c]);
+              }
+          }
+    for (size_t i = 0; i < ref.size(); ++i)
+      if (bf2f(xv[i]) <= 0.f) ref[i] = 0.f;   // fused ReLU mask
+    // repack: wt[c][2-ky][2-kx][f] = w[f][ky][kx][c], bf16-rounded
+    std::vector<u16> wt((size_t)cin * 9 * cout);
+    for (int f = 0; f < cout; ++f)
+      for (int ky = 0; ky < 3; ++ky)
+        for (int kx = 0; kx < 3; ++kx)
+          for (int c = 0; c < cin; ++c)
+            wt[(((size_t)c * 3 + (2 - ky)) * 3 + (2 - kx)) * cout + f] =
+                wv[(((size_t)f * 3 + ky) * 3 + kx) * cin + c];
+    void *dX, *dW_, *dM, *dO;
+    CHECK(hipMalloc(&dX, dyv.size() * 2));
+    CHECK(hipMalloc(&dW_, wt.size() * 2));
+    CHECK(hipMalloc(&dM, xv.size() * 2));
+    CHECK(hipMalloc(&dO, xv.size() * 2));
+    CHECK(hipMemcpy(dX, dyv.data(), dyv.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dW_, wt.data(), wt.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dM, xv.data(), xv.size() * 2, hipMemcpyHostToDevice));
+    if (launch_deep_conv3x3_dx(dX, dW_, dM, dO, N, H, W, cout, cin, 0)) {
+      printf("dx unsupported %d/%d\n", cout, cin);
+      exit(2);
+    }
+    CHECK(hipDeviceSynchronize());
+    std::vector<u16> out(xv.size());
+    CHECK(hipMemcpy(out.data(), dO, out.size() * 2, hipMemcpyDeviceToHost));
+    float maxerr = 0.f, maxmag = 1.f;
+    for (size_t i = 0; i < ref.size(); ++i) {
+      maxmag = std::max(maxmag, std::fabs(ref[i]));
+      maxerr = std::max(maxerr, std::fabs(bf2f(out[i]) - ref[i]));
+    }
+    char nm[64];
+    snprintf(nm, sizeof nm, "conv dX %d->%d %dx%d relu-mask", cin, cout, H,
+             W);
+    report(nm, maxerr / maxmag, 2e-2f);
+    hipFree(dX); hipFree(dW_); hipFree(dM); hipFree(dO);
+  }
+
+  // ---- 6. conv backward: dW --------------------------------------------
+  for (int cfg = 0; cfg < 3; ++cfg) {
+    const int cin = cfg == 0 ? 4 : (cfg == 1 ? 16 : 32);
+    const int cout = cfg == 0 ? 16 : (cfg == 1 ? 32 : 32);
+    const int N = 2, H = cfg == 0 ? 72 : (cfg == 1 ? 36 : 18);
+    const int W = cfg == 0 ? 96 : (cfg == 1 ? 48 : 24);
+    const bool relu_in = cfg != 0;
+    unsigned seed = 77u + cfg;
+    std::vector<u16> xv((size_t)N * H * W * cin),
+        dyv((size_t)N * H * W * cout);
+    for (auto& v : xv) v = f2bf(frand(seed));
+    for (auto& v : dyv) v = f2bf(frand(seed) * 0.5f);
+    std::vector<float> ref((size_t)cout * 9 * cin, 0.f);
+    for (int n = 0; n < N; ++n)
+      for (int oy = 0; oy < H; ++oy)
+        for (int ox = 0; ox < W; ++ox)
+          for (int f = 0; f < cout; ++f) {
+            const float g =
+                bf2f(dyv[(((size_t)n * H + oy) * W + ox) * cout + f]);
+            for (int ky = 0; ky < 3; ++ky)
+              for (int kx = 0; kx < 3; ++kx) {
+                const int iy = oy + ky - 1, ix = ox + kx - 1;
+                if (iy < 0 || iy >= H || ix < 0 || ix >= W) continue;
+                for (int c = 0; c < cin; ++c) {
+                  float v =
+                      bf2f(xv[(((size_t)n * H + iy) * W + ix) * cin + c]);
+                  if (relu_in && v < 0.f) v = 0.f;
+                  ref[(((size_t)f * 3 + ky) * 3 + kx) * cin + c] += g * v;
+                }
+              }
+          }
+    void *dXp, *dYp, *dWp;
+    CHECK(hipMalloc(&dXp, xv.size() * 2));
+    CHECK(hipMalloc(&dYp, dyv.size() * 2));
+    CHECK(hipMalloc(&dWp, ref.size() * 4));
+    CHECK(hipMemcpy(dXp, xv.data(), xv.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dYp, dyv.data(), dyv.size() * 2,
+                    hipMemcpyHostToDevice));
+    CHECK(hipMemset(dWp, 0, ref.size() * 4));
+    if (launch_deep_conv3x3_dw(dXp, dYp, (float*)dWp, N, H, W, cin, cout,
+                               relu_in, 0)) {
+      printf("dw unsupported %d/%d\n", cin, cout);
+      exit(2);
+    }
+    CHECK(hipDeviceSynchronize());
+    std::vector<float> out(ref.size());
+    CHECK(hipMemcpy(out.data(), dWp, out.size() * 4,
+                    hipMemcpyDeviceToHost));
+    float maxerr = 0.f, maxmag = 1.f;
+    for (size_t i = 0; i < ref.size(); ++i) {
+      maxmag = std::max(maxmag, std::fabs(ref[i]));
+      maxerr = std::max(maxerr, std::fabs(out[i] - ref[i]));
+    }
+    char nm[64];
+    snprintf(nm, sizeof nm, "conv dW %d->%d %dx%d%s", cin, cout, H, W,
+             relu_in ? " relu" : "");
+    // bf16 products accumulated over N*H*W px: tolerance scales with
+    // sqrt(count); 2e-2 relative holds comfortably at these sizes.
+    report(nm, maxerr / maxmag, 2e-2f);
+    hipFree(dXp); hipFree(dYp); hipFree(dWp);
+  }
+
+  // ---- 7. maxpool argmax + backward ------------------------------------
+  {
+    const int N = 2, H = 36, W = 48, C = 16, HO = 18, WO = 24;
+    unsigned seed = 13u;
+    std::vector<u16> xv((size_t)N * H * W * C), dyv((size_t)N * HO * WO * C);
+    for (auto& v : xv) v = f2bf(frand(seed));
+    for (auto& v : dyv) v = f2bf(frand(seed));
+    // CPU: fwd argmax then scatter
+    std::vector<float> ref((size_t)N * H * W * C, 0.f);
+    for (int n = 0; n < N; ++n)
+      for (int oy = 0; oy < HO; ++oy)
+        for (int ox = 0; ox < WO; ++ox)
+          for (int c = 0; c < C; ++c) {
+            float best = -3.4e38f;
+            int by = 0, bx = 0;
+            for (int dy2 = 0; dy2 < 3; ++dy2)
+              for (int dx2 = 0; dx2 < 3; ++dx2) {
+                int iy = oy * 2 - 1 + dy2, ix = ox * 2 - 1 + dx2;
+                if (iy < 0 || iy >= H || ix < 0 || ix >= W) continue;
+                float v = bf2f(xv[(((size_t)n * H + iy) * W + ix) * C + c]);
+                if (v > best) { best = v; by = iy; bx = ix; }
+              }
+            ref[(((size_t)n * H + by) * W + bx) * C + c] +=
+                bf2f(dyv[(((size_t)n * HO + oy) * WO + ox) * C + c]);
+          }
+    void *dxp, *dyp2, *dip, *dgp, *dop;
+    CHECK(hipMalloc(&dxp, xv.size() * 2));
+    CHECK(hipMalloc(&dyp2, dyv.size() * 2));
+    CHECK(hipMalloc(&dip, dyv.size()));
+    CHECK(hipMalloc(&dgp, dyv.size() * 2));
+    CHECK(hipMalloc(&dop, xv.size() * 2));
+    CHECK(hipMemcpy(dxp, xv.data(), xv.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dgp, dyv.data(), dyv.size() * 2,
+                    hipMemcpyHostToDevice));
+    launch_deep_maxpool(dxp, dyp2, dip, N, H, W, C, 0);
+    launch_deep_maxpool_bwd(dgp, dip, dop, N, H, W, C, 0);
+    CHECK(hipDeviceSynchronize());
+    std::vector<u16> out(xv.size());
+    CHECK(hipMemcpy(out.data(), dop, out.size() * 2,
+                    hipMemcpyDeviceToHost));
+    float maxerr = 0.f;
+    for (size_t i = 0; i < ref.size(); ++i)
+      maxerr = std::max(maxerr,
+                        std::fabs(bf2f(out[i]) - bf2f(f2bf(ref[i]))));
+    report("maxpool bwd (argmax scatter==gather)", maxerr, 1e-6f);
+    hipFree(dxp); hipFree(dyp2); hipFree(dip); hipFree(dgp); hipFree(dop);
   }
 
   printf(n_fail ? "DEEP_PROBE FAIL (%d)\n" : "DEEP_PROBE ALL PASS\n", n_fail);

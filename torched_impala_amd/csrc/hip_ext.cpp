@@ -56,8 +56,14 @@ void launch_linear_mfma(const float*, const float*, const float*,
                         hipStream_t);
 int launch_deep_conv3x3(const void*, const void*, const float*, const void*,
                         void*, int, int, int, int, int, int, hipStream_t);
-void launch_deep_maxpool(const void*, void*, int, int, int, int,
+void launch_deep_maxpool(const void*, void*, void*, int, int, int, int,
                          hipStream_t);
+void launch_deep_maxpool_bwd(const void*, const void*, void*, int, int, int,
+                             int, hipStream_t);
+int launch_deep_conv3x3_dx(const void*, const void*, const void*, void*,
+                           int, int, int, int, int, hipStream_t);
+int launch_deep_conv3x3_dw(const void*, const void*, float*, int, int, int,
+                           int, int, int, hipStream_t);
 void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int, hipStream_t);
 void launch_nchw_f32_to_nhwc_bf16(const void*, void*, int, int, int, int,
                                   hipStream_t);
@@ -434,15 +440,80 @@ torch::Tensor deep_conv3x3(torch::Tensor x, torch::Tensor w, torch::Tensor b,
   return y;
 }
 
-torch::Tensor deep_maxpool(torch::Tensor x) {
+std::vector<torch::Tensor> deep_maxpool(torch::Tensor x, bool want_argmax) {
   check_bf16_nhwc(x, "x");
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   DeviceScope guard(x.get_device());
   auto y = torch::empty({N, (H + 1) / 2, (W + 1) / 2, C}, x.options());
-  launch_deep_maxpool(x.data_ptr(), y.data_ptr(), N, H, W, C,
+  torch::Tensor am;
+  void* am_p = nullptr;
+  if (want_argmax) {
+    am = torch::empty({N, (H + 1) / 2, (W + 1) / 2, C},
+                      x.options().dtype(torch::kUInt8));
+    am_p = am.data_ptr();
+  } else {
+    am = torch::empty({0}, x.options().dtype(torch::kUInt8));
+  }
+  launch_deep_maxpool(x.data_ptr(), y.data_ptr(), am_p, N, H, W, C,
                       current_stream());
   HIP_CHECK(hipGetLastError());
-  return y;
+  return {y, am};
+}
+
+torch::Tensor deep_maxpool_bwd(torch::Tensor dy, torch::Tensor argmax,
+                               int64_t H, int64_t W) {
+  check_bf16_nhwc(dy, "dy");
+  TORCH_CHECK(argmax.scalar_type() == torch::kUInt8 &&
+                  argmax.is_contiguous(), "argmax must be u8 contiguous");
+  const int N = dy.size(0), C = dy.size(3);
+  DeviceScope guard(dy.get_device());
+  auto dx = torch::empty({N, H, W, C}, dy.options());
+  launch_deep_maxpool_bwd(dy.data_ptr(), argmax.data_ptr(), dx.data_ptr(),
+                          N, (int)H, (int)W, C, current_stream());
+  HIP_CHECK(hipGetLastError());
+  return dx;
+}
+
+// dX: wt pre-repacked [cin_fwd][3][3][cout_fwd] (rot180 + transpose);
+// xmask (optional) = the forward input, for the fused ReLU backward.
+torch::Tensor deep_conv3x3_dx(torch::Tensor dy, torch::Tensor wt,
+                              c10::optional<torch::Tensor> xmask) {
+  check_bf16_nhwc(dy, "dy");
+  check_bf16_nhwc(wt, "wt");
+  const int N = dy.size(0), H = dy.size(1), W = dy.size(2);
+  const int cout_fwd = dy.size(3), cin_fwd = wt.size(0);
+  TORCH_CHECK(wt.size(3) == cout_fwd, "wt must be [cin][3][3][cout]");
+  DeviceScope guard(dy.get_device());
+  auto dx = torch::empty({N, H, W, cin_fwd}, dy.options());
+  const void* mask_p = nullptr;
+  if (xmask.has_value()) {
+    check_bf16_nhwc(*xmask, "xmask");
+    mask_p = xmask->data_ptr();
+  }
+  int rc = launch_deep_conv3x3_dx(dy.data_ptr(), wt.data_ptr(), mask_p,
+                                  dx.data_ptr(), N, H, W, cout_fwd, cin_fwd,
+                                  current_stream());
+  TORCH_CHECK(rc == 0, "unsupported dx cout/cin ", cout_fwd, "/", cin_fwd);
+  HIP_CHECK(hipGetLastError());
+  return dx;
+}
+
+// dW in the kernel's [cout][3][3][cin] layout, fp32.
+torch::Tensor deep_conv3x3_dw(torch::Tensor x, torch::Tensor dy,
+                              bool relu_in) {
+  check_bf16_nhwc(x, "x");
+  check_bf16_nhwc(dy, "dy");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), cin = x.size(3);
+  const int cout = dy.size(3);
+  DeviceScope guard(x.get_device());
+  auto dw = torch::zeros({cout, 3, 3, cin},
+                         x.options().dtype(torch::kFloat32));
+  int rc = launch_deep_conv3x3_dw(x.data_ptr(), dy.data_ptr(),
+                                  dw.data_ptr<float>(), N, H, W, cin, cout,
+                                  relu_in ? 1 : 0, current_stream());
+  TORCH_CHECK(rc == 0, "unsupported dw cin/cout ", cin, "/", cout);
+  HIP_CHECK(hipGetLastError());
+  return dw;
 }
 
 // u8 [N][3][H][W] -> bf16 [N][H][W][4] with /255 and a zero pad channel
@@ -502,7 +573,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deep_conv3x3", &deep_conv3x3, py::arg("x"), py::arg("w"),
         py::arg("b"), py::arg("res") = py::none(),
         py::arg("relu_in") = false);
-  m.def("deep_maxpool", &deep_maxpool);
+  m.def("deep_maxpool", &deep_maxpool, py::arg("x"),
+        py::arg("want_argmax") = false);
+  m.def("deep_maxpool_bwd", &deep_maxpool_bwd);
+  m.def("deep_conv3x3_dx", &deep_conv3x3_dx, py::arg("dy"), py::arg("wt"),
+        py::arg("xmask") = py::none());
+  m.def("deep_conv3x3_dw", &deep_conv3x3_dw, py::arg("x"), py::arg("dy"),
+        py::arg("relu_in") = false);
   m.def("deep_u8_to_nhwc4", &deep_u8_to_nhwc4);
   m.def("nchw_to_nhwc_bf16", &nchw_to_nhwc_bf16);
   m.def("nhwc_bf16_to_nchw", &nhwc_bf16_to_nchw);

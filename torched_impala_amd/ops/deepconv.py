@@ -78,7 +78,7 @@ class DeepEncoderWeights:
         for sec_w in self._convs:
             (wh, bh), (w00, b00), (w01, b01), (w10, b10), (w11, b11) = sec_w
             x = ext.deep_conv3x3(x, wh, bh)
-            x = ext.deep_maxpool(x)
+            x = ext.deep_maxpool(x)[0]
             y = ext.deep_conv3x3(x, w00, b00, relu_in=True)
             x = ext.deep_conv3x3(y, w01, b01, res=x, relu_in=True)
             y = ext.deep_conv3x3(x, w10, b10, relu_in=True)

@@ -249,6 +249,16 @@ class DeepResNet(_CoreMixin, _HeadsMixin):
         self._init_heads(self.core_output_size, num_actions)
 
     def encode(self, x: torch.Tensor) -> torch.Tensor:
+        """(N, 3, H, W) uint8/float -> (N, 256).  On GPU the whole conv
+        stack runs on the NHWC bf16 MFMA kernels (ops/deepconv.py,
+        design docs/deep_conv_design.md); eager fp32 NCHW otherwise."""
+        if (x.is_cuda and x.dtype == torch.uint8
+                and self.obs_shape[0] == 3
+                and not getattr(self, "force_eager_core", False)):
+            from torched_impala_amd import ops
+            if ops.hip_available():
+                from torched_impala_amd.ops.deepconv import deep_encode_train
+                return deep_encode_train(self, x)
         x = x.float() / 255.0
         for sec in self.sections:
             x = sec(x)

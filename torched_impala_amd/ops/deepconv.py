@@ -25,6 +25,90 @@ import torch
 from . import ext_hip
 
 
+class _DeepConv3x3(torch.autograd.Function):
+    """Fused conv3x3(pad1) NHWC bf16 with optional pre-conv ReLU and
+    residual add; grads to x/res (bf16 NHWC) and to the model's fp32
+    [cout][cin][3][3] weight + bias.  Kernels: deep_kernels.hip,
+    hardware-validated in profiles/deep_probe_bwd_r01.log."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, res, relu_in):
+        ext = ext_hip()
+        cin = x.shape[3]
+        packed = pack_conv_weight(w, cin)          # [cout][3][3][cin] bf16
+        y = ext.deep_conv3x3(x.contiguous(), packed,
+                             b.detach().float().contiguous(),
+                             res=res.contiguous() if res is not None
+                             else None,
+                             relu_in=relu_in)
+        ctx.relu_in = relu_in
+        ctx.has_res = res is not None
+        ctx.cin_model = w.shape[1]
+        ctx.save_for_backward(x, packed)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ext_hip()
+        x, packed = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            # repack: wt[cin][3][3][cout] = packed rot180 + transposed
+            wt = packed.flip(1, 2).permute(3, 1, 2, 0).contiguous()
+            dx = ext.deep_conv3x3_dx(dy, wt,
+                                     xmask=x if ctx.relu_in else None)
+        if ctx.needs_input_grad[1]:
+            dwk = ext.deep_conv3x3_dw(x.contiguous(), dy,
+                                      relu_in=ctx.relu_in)
+            dw = dwk.permute(0, 3, 1, 2)[:, :ctx.cin_model].contiguous()
+        if ctx.needs_input_grad[2]:
+            db = dy.float().sum(dim=(0, 1, 2))
+        dres = dy if ctx.has_res else None
+        return dx, dw, db, dres, None
+
+
+class _DeepMaxpool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = ext_hip()
+        y, am = ext.deep_maxpool(x.contiguous(), want_argmax=True)
+        ctx.hw = (x.shape[1], x.shape[2])
+        ctx.save_for_backward(am)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ext_hip()
+        (am,) = ctx.saved_tensors
+        return ext.deep_maxpool_bwd(dy.contiguous(), am, *ctx.hw)
+
+
+def deep_encode_train(model, frames_u8: torch.Tensor) -> torch.Tensor:
+    """Differentiable DeepResNet.encode on the NHWC bf16 kernels:
+    u8 [N][3][H][W] -> fp32 [N, 256] (sections + ReLU + fc + ReLU).
+    Weight/bias grads land on the model's fp32 params; activations are
+    bf16 NHWC throughout the conv stack (docs/deep_conv_design.md §5)."""
+    ext = ext_hip()
+    x = ext.deep_u8_to_nhwc4(frames_u8.contiguous())
+    for sec in model.sections:
+        x = _DeepConv3x3.apply(x, sec.conv.weight, sec.conv.bias, None,
+                               False)
+        x = _DeepMaxpool.apply(x)
+        y = _DeepConv3x3.apply(x, sec.res0.conv0.weight, sec.res0.conv0.bias,
+                               None, True)
+        x = _DeepConv3x3.apply(y, sec.res0.conv1.weight, sec.res0.conv1.bias,
+                               x, True)
+        y = _DeepConv3x3.apply(x, sec.res1.conv0.weight, sec.res1.conv0.bias,
+                               None, True)
+        x = _DeepConv3x3.apply(y, sec.res1.conv1.weight, sec.res1.conv1.bias,
+                               x, True)
+    n = x.shape[0]
+    flat = torch.relu(x.float()).permute(0, 3, 1, 2).reshape(n, -1)
+    return torch.relu(torch.nn.functional.linear(
+        flat, model.fc.weight, model.fc.bias))
+
+
 def pack_conv_weight(w: torch.Tensor, cin_pad: int | None = None
                      ) -> torch.Tensor:
     """[cout][cin][3][3] fp32 -> contiguous [cout][3][3][cin_pad] bf16."""

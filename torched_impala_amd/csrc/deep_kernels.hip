@@ -313,10 +313,18 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
 // k(px)-strided scalar reads — correctness-first; ds_read_b64_tr_b16 is
 // the round-2 lever.
 // ---------------------------------------------------------------------------
+// dw_slab: [DW_NSLAB][cout*9*cin] zero-initialised fp32; workgroup i
+// accumulates into slab i % DW_NSLAB (the caller sums the slabs).  With
+// a single copy ~310k workgroups serialise on each of the ~2-5k words
+// (measured 9 TF); 256 slabs cut the per-word contention 256x.
+constexpr int DW_NSLAB = 256;
+
 template <int CIN, int COUT, bool RELU_IN>
 __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
-    float* __restrict__ dw, int N, int H, int W) {
+    float* __restrict__ dw_slab, int N, int H, int W) {
+  float* __restrict__ dw =
+      dw_slab + (size_t)(blockIdx.x % DW_NSLAB) * (COUT * 9 * CIN);
   constexpr int MTOT = 9 * CIN;                  // 144 or 288
   constexpr int MFRAGS = (MTOT + 15) / 16;       // 9 or 18
   constexpr int NFRAG = COUT / 16;
@@ -651,6 +659,10 @@ extern "C" int launch_deep_conv3x3_dx(const void* dy, const void* wt,
   return 0;
 }
 
+extern "C" int deep_dw_nslab() { return DW_NSLAB; }
+
+// dw must point at DW_NSLAB zeroed copies of [cout][3][3][cin] fp32; the
+// caller sums them (deep_dw_nslab() exposes the count).
 extern "C" int launch_deep_conv3x3_dw(const void* x, const void* dy,
                                       float* dw, int N, int H, int W,
                                       int cin, int cout, int relu_in,

@@ -25,6 +25,7 @@ extern "C" int launch_deep_conv3x3_dx(const void*, const void*, const void*,
                                       hipStream_t);
 extern "C" int launch_deep_conv3x3_dw(const void*, const void*, float*, int,
                                       int, int, int, int, int, hipStream_t);
+extern "C" int deep_dw_nslab();
 extern "C" void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int,
                                         hipStream_t);
 extern "C" void launch_mfma_probe(const void*, const void*, float*,
@@ -190,6 +191,46 @@ int main(int argc, char** argv) {
     bench_case("conv 16->32 36x48 (sec2 head)", N, 36, 48, 16, 32, 0, 0);
     bench_case("conv 32->32 18x24 relu+res", N, 18, 24, 32, 32, 1, 1);
     bench_case("conv 32->32  9x12 relu+res", N, 9, 12, 32, 32, 1, 1);
+    // backward kernels at the dominant shapes
+    {
+      const int H = 36, W = 48, cin = 16, cout = 16;
+      void *dyp, *wtp, *mp, *op, *dwp;
+      CHECK(hipMalloc(&dyp, (size_t)N * H * W * cout * 2));
+      CHECK(hipMalloc(&wtp, (size_t)cin * 9 * cout * 2));
+      CHECK(hipMalloc(&mp, (size_t)N * H * W * cin * 2));
+      CHECK(hipMalloc(&op, (size_t)N * H * W * cin * 2));
+      CHECK(hipMalloc(&dwp, (size_t)deep_dw_nslab() * cout * 9 * cin * 4));
+      CHECK(hipMemset(dwp, 0, (size_t)deep_dw_nslab() * cout * 9 * cin * 4));
+      CHECK(hipMemset(dyp, 0x3c, (size_t)N * H * W * cout * 2));
+      CHECK(hipMemset(wtp, 0x3c, (size_t)cin * 9 * cout * 2));
+      CHECK(hipMemset(mp, 0x3c, (size_t)N * H * W * cin * 2));
+      hipEvent_t e0, e1;
+      CHECK(hipEventCreate(&e0)); CHECK(hipEventCreate(&e1));
+      const int iters = 20;
+      for (int i = 0; i < 3; ++i)
+        launch_deep_conv3x3_dx(dyp, wtp, mp, op, N, H, W, cout, cin, 0);
+      CHECK(hipEventRecord(e0, 0));
+      for (int i = 0; i < iters; ++i)
+        launch_deep_conv3x3_dx(dyp, wtp, mp, op, N, H, W, cout, cin, 0);
+      CHECK(hipEventRecord(e1, 0));
+      CHECK(hipDeviceSynchronize());
+      float ms; CHECK(hipEventElapsedTime(&ms, e0, e1)); ms /= iters;
+      printf("%-40s N=%-6d %8.3f ms  %7.1f TF\n", "dX 16->16 36x48+mask", N,
+             ms, 2.0 * N * H * W * cout * 9.0 * cin / ms / 1e9);
+      for (int i = 0; i < 3; ++i)
+        launch_deep_conv3x3_dw(mp, dyp, (float*)dwp, N, H, W, cin, cout, 1,
+                               0);
+      CHECK(hipEventRecord(e0, 0));
+      for (int i = 0; i < iters; ++i)
+        launch_deep_conv3x3_dw(mp, dyp, (float*)dwp, N, H, W, cin, cout, 1,
+                               0);
+      CHECK(hipEventRecord(e1, 0));
+      CHECK(hipDeviceSynchronize());
+      CHECK(hipEventElapsedTime(&ms, e0, e1)); ms /= iters;
+      printf("%-40s N=%-6d %8.3f ms  %7.1f TF\n", "dW 16->16 36x48 relu", N,
+             ms, 2.0 * N * H * W * cout * 9.0 * cin / ms / 1e9);
+      hipFree(dyp); hipFree(wtp); hipFree(mp); hipFree(op); hipFree(dwp);
+    }
     return 0;
   }
   int ndev = 0;
@@ -418,23 +459,28 @@ int main(int argc, char** argv) {
                 }
               }
           }
+    const int nslab = deep_dw_nslab();
     void *dXp, *dYp, *dWp;
     CHECK(hipMalloc(&dXp, xv.size() * 2));
     CHECK(hipMalloc(&dYp, dyv.size() * 2));
-    CHECK(hipMalloc(&dWp, ref.size() * 4));
+    CHECK(hipMalloc(&dWp, (size_t)nslab * ref.size() * 4));
     CHECK(hipMemcpy(dXp, xv.data(), xv.size() * 2, hipMemcpyHostToDevice));
     CHECK(hipMemcpy(dYp, dyv.data(), dyv.size() * 2,
                     hipMemcpyHostToDevice));
-    CHECK(hipMemset(dWp, 0, ref.size() * 4));
+    CHECK(hipMemset(dWp, 0, (size_t)nslab * ref.size() * 4));
     if (launch_deep_conv3x3_dw(dXp, dYp, (float*)dWp, N, H, W, cin, cout,
                                relu_in, 0)) {
       printf("dw unsupported %d/%d\n", cin, cout);
       exit(2);
     }
     CHECK(hipDeviceSynchronize());
-    std::vector<float> out(ref.size());
-    CHECK(hipMemcpy(out.data(), dWp, out.size() * 4,
+    std::vector<float> slabs((size_t)nslab * ref.size());
+    CHECK(hipMemcpy(slabs.data(), dWp, slabs.size() * 4,
                     hipMemcpyDeviceToHost));
+    std::vector<float> out(ref.size(), 0.f);
+    for (int sl = 0; sl < nslab; ++sl)
+      for (size_t i = 0; i < ref.size(); ++i)
+        out[i] += slabs[(size_t)sl * ref.size() + i];
     float maxerr = 0.f, maxmag = 1.f;
     for (size_t i = 0; i < ref.size(); ++i) {
       maxmag = std::max(maxmag, std::fabs(ref[i]));

@@ -64,6 +64,7 @@ int launch_deep_conv3x3_dx(const void*, const void*, const void*, void*,
                            int, int, int, int, int, hipStream_t);
 int launch_deep_conv3x3_dw(const void*, const void*, float*, int, int, int,
                            int, int, int, hipStream_t);
+int deep_dw_nslab();
 void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int, hipStream_t);
 void launch_nchw_f32_to_nhwc_bf16(const void*, void*, int, int, int, int,
                                   hipStream_t);
@@ -498,7 +499,9 @@ torch::Tensor deep_conv3x3_dx(torch::Tensor dy, torch::Tensor wt,
   return dx;
 }
 
-// dW in the kernel's [cout][3][3][cin] layout, fp32.
+// dW in the kernel's [cout][3][3][cin] layout, fp32.  The kernel
+// scatters into deep_dw_nslab() slab copies (atomic-contention split);
+// the sum over slabs happens here as one torch reduction.
 torch::Tensor deep_conv3x3_dw(torch::Tensor x, torch::Tensor dy,
                               bool relu_in) {
   check_bf16_nhwc(x, "x");
@@ -506,14 +509,14 @@ torch::Tensor deep_conv3x3_dw(torch::Tensor x, torch::Tensor dy,
   const int N = x.size(0), H = x.size(1), W = x.size(2), cin = x.size(3);
   const int cout = dy.size(3);
   DeviceScope guard(x.get_device());
-  auto dw = torch::zeros({cout, 3, 3, cin},
-                         x.options().dtype(torch::kFloat32));
+  auto slab = torch::zeros({deep_dw_nslab(), cout, 3, 3, cin},
+                           x.options().dtype(torch::kFloat32));
   int rc = launch_deep_conv3x3_dw(x.data_ptr(), dy.data_ptr(),
-                                  dw.data_ptr<float>(), N, H, W, cin, cout,
-                                  relu_in ? 1 : 0, current_stream());
+                                  slab.data_ptr<float>(), N, H, W, cin,
+                                  cout, relu_in ? 1 : 0, current_stream());
   TORCH_CHECK(rc == 0, "unsupported dw cin/cout ", cin, "/", cout);
   HIP_CHECK(hipGetLastError());
-  return dw;
+  return slab.sum(0);
 }
 
 // u8 [N][3][H][W] -> bf16 [N][H][W][4] with /255 and a zero pad channel

@@ -50,6 +50,21 @@ def test_cli_train_entry():
 
 
 @pytest.mark.timeout(420)
+def test_cli_train_deep_dmlab_cpu():
+    """End-to-end train loop with the deep ResNet on the DMLab-shaped
+    synthetic env (eager fp32 path on CPU; the NHWC bf16 kernel path is
+    covered by test_gpu_deep.py)."""
+    r = subprocess.run(
+        [sys.executable, "-m", "torched_impala_amd.cli", "--env",
+         "synthetic-dmlab", "--model", "deep", "--device", "cpu",
+         "--num-actors", "1", "--envs-per-actor", "2", "--batch-size", "2",
+         "--unroll-length", "5", "--total-steps", "20",
+         "--disable-checkpoint"],
+        capture_output=True, text=True, cwd=ROOT, timeout=360)
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+@pytest.mark.timeout(420)
 def test_bench_torchrun_two_ranks_cpu():
     """The driver's multi-GPU scaling invocation, on CPU/gloo: torchrun
     2 ranks -> init_distributed -> per-rank actors -> eager learn steps

@@ -177,8 +177,10 @@ class Rollout:
         self.done = torch.ones(n, dtype=torch.bool)  # first frame: episode start
         self.last_action = torch.zeros(n, dtype=torch.int64)
         self.core_state = model.initial_state(n)
-        self.episode_return = torch.zeros(n)
-        self.episode_step = torch.zeros(n, dtype=torch.int32)
+        # episode stats live in numpy: the hot loop updates them
+        # vectorised (torch scalar indexing costs ~10 us/env/step)
+        self.episode_return = np.zeros(n, dtype=np.float32)
+        self.episode_step = np.zeros(n, dtype=np.int32)
         self.core_before = tuple(t.clone() for t in self.core_state)
         self.carry = None
         self.fast = None
@@ -248,15 +250,14 @@ class Rollout:
             done=self.done.clone(), action=action,
             behavior_logits=out["policy_logits"][0],
             last_action=self.last_action,  # the action INPUT to this forward
-            episode_return=self.episode_return.clone(),
-            episode_step=self.episode_step.clone(),
+            episode_return=torch.from_numpy(self.episode_return.copy()),
+            episode_step=torch.from_numpy(self.episode_step.copy()),
         )
         finished = self.done
         if finished.any():  # reset stats after the done row was recorded
-            self.episode_return = self.episode_return.clone()
-            self.episode_step = self.episode_step.clone()
-            self.episode_return[finished] = 0.0
-            self.episode_step[finished] = 0
+            fin = finished.numpy()
+            self.episode_return[fin] = 0.0
+            self.episode_step[fin] = 0
         self.core_state = new_core
         self.last_action = action
         # step all envs with the chosen actions (auto-reset on done);
@@ -276,13 +277,13 @@ class Rollout:
         act = action.tolist()
         for i, env in enumerate(self.envs):
             o, r, d, _ = env.step(act[i])
-            self.episode_return[i] += r
-            self.episode_step[i] += 1
             if d:
                 o = env.reset()
             obs_buf[i] = o
             rew[i] = r
             done[i] = d
+        self.episode_return += rew
+        self.episode_step += 1
         self.frame = torch.from_numpy(obs_buf)
         self.reward = torch.from_numpy(rew)
         self.done = torch.from_numpy(done)

@@ -59,6 +59,36 @@ def test_behavior_logits_match_replay(jit):
                 msg=lambda m: f"row {t} logits drifted: {m}")
 
 
+def test_native_bf16_mode_self_consistent_and_close():
+    """actor_jit=native-bf16: behavior logits are the bf16 policy's own
+    (self-consistent with its sampling — what V-trace needs) and close
+    to the fp32 replay. On hosts without avx512bf16 the mode silently
+    runs the fp32 path, so this test is valid everywhere."""
+    flags, model, roll = _setup(seed=3, actor_jit="native-bf16")
+    T = 5
+    bufs = TrajectoryBuffers(T, (4, 84, 84), torch.uint8, 6, 2,
+                             envs_per_slot=4)
+    roll.fill_slot(bufs.slot(0), T)
+    s = bufs.slot(0)
+    # (a) deterministic self-consistency: the same native policy on the
+    # stored rows reproduces the stored logits bitwise
+    nat = roll.fast
+    assert nat is not None and nat.mode == "native"
+    for t in range(T + 1):
+        logits, _ = nat(s["frame"][t], s["reward"][t], s["last_action"][t])
+        assert torch.equal(logits, s["behavior_logits"][t]), f"row {t}"
+    # (b) proximity to the fp32 policy (bf16 rounding only)
+    with torch.inference_mode():
+        for t in range(T + 1):
+            inp = dict(frame=s["frame"][t:t + 1], reward=s["reward"][t:t + 1],
+                       done=s["done"][t:t + 1],
+                       last_action=s["last_action"][t:t + 1])
+            out, _ = model(inp, ())
+            err = (out["policy_logits"][0]
+                   - s["behavior_logits"][t]).abs().max().item()
+            assert err < 5e-3, f"row {t}: bf16 drift {err}"
+
+
 def test_lstm_initial_state_replay():
     """initial_core_state lets the learner replay rows 0..T exactly."""
     flags, model, roll = _setup(use_lstm=True, seed=7)

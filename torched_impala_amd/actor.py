@@ -78,13 +78,15 @@ class NativeActorPolicy:
 
     mode = "native"
 
-    def __init__(self, model, num_actions: int, seed: int):
+    def __init__(self, model, num_actions: int, seed: int,
+                 bf16: bool = False):
         import torched_impala_amd._impala_actor as ext
         assert model.obs_shape == (4, 84, 84)
         self.use_lstm = bool(getattr(model, "use_lstm", False))
         if self.use_lstm:
             assert model.core.hidden_size == 256
-        self._ext = ext.ShallowActorNet(seed)
+        self._ext = ext.ShallowActorNet(seed, bf16)
+        self.bf16_active = bool(getattr(self._ext, "bf16_active", False))
         self._model = model
         self._A = num_actions
         self.refresh()
@@ -188,12 +190,13 @@ class Rollout:
         use_lstm = getattr(model, "use_lstm", False)
         if jit_mode != "off" and hasattr(model, "encode"):
             try:
-                if (jit_mode == "native"
+                if (jit_mode in ("native", "native-bf16")
                         and getattr(model, "obs_shape", None) == (4, 84, 84)):
                     self.fast = NativeActorPolicy(
                         model, model.num_actions,
-                        seed=flags.seed + 77777 * actor_index)
-                elif jit_mode == "native":
+                        seed=flags.seed + 77777 * actor_index,
+                        bf16=jit_mode == "native-bf16")
+                elif jit_mode in ("native", "native-bf16"):
                     # native covers the shallow feedforward net only;
                     # other policies take the traced-frozen path
                     example = ((self.frame, self.reward, self.done,

@@ -33,10 +33,12 @@ class Flags:
     envs_per_actor: int = 1            # batched envs inside one actor process
     actor_threads: int = 0             # CPU threads per actor; 0 = auto
                                        # (cores/num_actors, capped at 8)
-    actor_jit: str = "native"          # off | safe | frozen |
-                                       # frozen (weights folded; fastest,
-                                       # refreshed every actor_jit_refresh
-                                       # slot claims)
+    actor_jit: str = "native"          # off | safe | frozen | native |
+                                       # native-bf16 (AVX512-BF16 conv/fc
+                                       # stages, ~2x MAC rate; behavior
+                                       # logits are the bf16 policy's --
+                                       # V-trace corrects staleness and
+                                       # precision alike)
     actor_jit_refresh: int = 4
 
     # --- model ---
@@ -106,7 +108,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--envs-per-actor", type=int, default=f.envs_per_actor)
     p.add_argument("--actor-threads", type=int, default=f.actor_threads)
     p.add_argument("--actor-jit", default=f.actor_jit,
-                   choices=["off", "safe", "frozen", "native"])
+                   choices=["off", "safe", "frozen", "native",
+                            "native-bf16"])
     p.add_argument("--actor-jit-refresh", type=int, default=f.actor_jit_refresh)
     _add_bool(p, "use-lstm", f.use_lstm)
     p.add_argument("--model", default=f.model,

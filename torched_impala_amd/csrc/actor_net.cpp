@@ -50,13 +50,29 @@ struct ShallowActorNet {
   // scratch
   std::vector<float> img, a1, a2, a3, fc_out, ci, logits_pad;
 
-  explicit ShallowActorNet(int64_t seed) : rng((uint64_t)seed) {
+  explicit ShallowActorNet(int64_t seed, bool want_bf16 = false)
+      : rng((uint64_t)seed) {
     use512 = __builtin_cpu_supports("avx512f");
+    use_bf16 = want_bf16 && __builtin_cpu_supports("avx512bf16");
     img.resize(C0 * H0 * W0);
     a1.resize(F1 * H1 * W1);
     a2.resize(F2 * H2 * W2);
     a3.resize(FLAT);
     fc_out.resize(FC);
+    if (use_bf16) {
+      img_h.resize(C0 * H0 * W0);
+      a1_h.resize(F1 * H1 * W1 + 2);
+      a2_h.resize(F2 * H2 * W2 + 2);   // +2: conv3's padded-pair overread
+      a3_h.resize(FLAT);
+      for (int v = 0; v < 256; ++v) u8lut[v] = f2bf((float)v * (1.f / 255.f));
+    }
+  }
+
+  static uint16_t f2bf(float f) {  // round-nearest-even fp32 -> bf16
+    uint32_t u;
+    std::memcpy(&u, &f, 4);
+    u += 0x7fff + ((u >> 16) & 1);
+    return (uint16_t)(u >> 16);
   }
 
   static void check(const torch::Tensor& t, std::initializer_list<int64_t> s,
@@ -95,6 +111,38 @@ struct ShallowActorNet {
     pack_conv(w1, cw1, F1, C0, K1);
     pack_conv(w2, cw2, F2, F1, K2);
     pack_conv(w3, cw3, F3, F2, K3);
+    if (use_bf16) {
+      // kx-PAIR packed bf16 weights for vdpbf16ps: per (c,ky,kxp) a run
+      // of F bf16 pairs, lane f = (w[kx=2kp][f], w[kx=2kp+1][f]); odd K
+      // (conv3) pads the last pair's high half with zero.
+      auto pack_pairs = [](std::vector<uint16_t>& dst,
+                           const torch::Tensor& src, int F, int C, int K) {
+        const int KP = (K + 1) / 2;
+        dst.assign((size_t)C * K * KP * 2 * F, 0);  // [c][ky][kxp][f*2+j]
+        auto s = src.accessor<float, 4>();
+        for (int c = 0; c < C; ++c)
+          for (int ky = 0; ky < K; ++ky)
+            for (int kp = 0; kp < KP; ++kp)
+              for (int f = 0; f < F; ++f)
+                for (int j = 0; j < 2; ++j) {
+                  const int kx = 2 * kp + j;
+                  if (kx < K)
+                    dst[((((size_t)c * K + ky) * KP + kp) * F + f) * 2 + j] =
+                        f2bf(s[f][c][ky][kx]);
+                }
+      };
+      pack_pairs(w1p, cw1, F1, C0, K1);
+      pack_pairs(w2p, cw2, F2, F1, K2);
+      pack_pairs(w3p, cw3, F3, F2, K3);
+      // fc: k-PAIRS, lane f = (w[2kp][f], w[2kp+1][f]); FLAT is even.
+      wfcp.assign((size_t)(FLAT / 2) * 2 * FC, 0);
+      auto s = fw.accessor<float, 2>();
+      for (int kp = 0; kp < FLAT / 2; ++kp)
+        for (int f = 0; f < FC; ++f) {
+          wfcp[((size_t)kp * FC + f) * 2 + 0] = f2bf(s[f][2 * kp]);
+          wfcp[((size_t)kp * FC + f) * 2 + 1] = f2bf(s[f][2 * kp + 1]);
+        }
+    }
     b1.assign(cb1.data_ptr<float>(), cb1.data_ptr<float>() + F1);
     b2.assign(cb2.data_ptr<float>(), cb2.data_ptr<float>() + F2);
     b3.assign(cb3.data_ptr<float>(), cb3.data_ptr<float>() + F3);
@@ -123,11 +171,190 @@ struct ShallowActorNet {
   // FMA contraction order per output element) ---
 
   bool use512 = false;
+  bool use_bf16 = false;
+  // bf16 path state: planar bf16 activations + kx-pair-packed weights
+  std::vector<uint16_t> img_h, a1_h, a2_h, a3_h;
+  std::vector<uint16_t> w1p, w2p, w3p, wfcp;
+  uint16_t u8lut[256];
 
-  void stage_conv1() { use512 ? stage_conv1_512() : stage_conv1_avx2(); }
-  void stage_conv2() { use512 ? stage_conv2_512() : stage_conv2_avx2(); }
-  void stage_conv3() { use512 ? stage_conv3_512() : stage_conv3_avx2(); }
-  void stage_fc() { use512 ? stage_fc_512() : stage_fc_avx2(); }
+  void stage_conv1() {
+    use_bf16 ? stage_conv1_bf16()
+             : (use512 ? stage_conv1_512() : stage_conv1_avx2());
+  }
+  void stage_conv2() {
+    use_bf16 ? stage_conv2_bf16()
+             : (use512 ? stage_conv2_512() : stage_conv2_avx2());
+  }
+  void stage_conv3() {
+    use_bf16 ? stage_conv3_bf16()
+             : (use512 ? stage_conv3_512() : stage_conv3_avx2());
+  }
+  void stage_fc() {
+    use_bf16 ? stage_fc_bf16()
+             : (use512 ? stage_fc_512() : stage_fc_avx2());
+  }
+
+  // --- bf16 stages (AVX512-BF16 vdpbf16ps: 2 MACs per lane per instr,
+  // fp32 accumulate; activations stored planar bf16). x-operands are
+  // kx-adjacent PAIRS broadcast as one 32-bit word; weights pre-packed
+  // in matching pair-interleave by load_weights. ~2x the fp32 MAC rate
+  // on full-512 cores (Zen5 / SPR). ---
+
+  __attribute__((target("avx512f,avx512bf16"))) static __m512bh
+  bcast_pair(const uint16_t* p) {
+    uint32_t w;
+    std::memcpy(&w, p, 4);
+    return (__m512bh)_mm512_set1_epi32((int)w);
+  }
+
+  __attribute__((target("avx512f,avx512bf16"))) void stage_conv1_bf16() {
+    const uint16_t* __restrict W = w1p.data();
+    const uint16_t* __restrict I = img_h.data();
+    uint16_t* __restrict O = a1_h.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bia = _mm512_loadu_ps(b1.data());
+    for (int oy = 0; oy < H1; ++oy) {
+      for (int ox = 0; ox < W1; ox += 10) {
+        __m512 acc[10];
+        for (int p = 0; p < 10; ++p) acc[p] = zero;
+        for (int c = 0; c < C0; ++c) {
+          const uint16_t* __restrict ic = I + c * (H0 * W0);
+          for (int ky = 0; ky < K1; ++ky) {
+            const uint16_t* __restrict row =
+                ic + (S1 * oy + ky) * W0 + S1 * ox;
+            const uint16_t* __restrict wr =
+                W + (((size_t)c * K1 + ky) * (K1 / 2)) * (2 * F1);
+            for (int kp = 0; kp < K1 / 2; ++kp) {
+              const __m512bh wv = (__m512bh)_mm512_loadu_si512(
+                  (const void*)(wr + kp * 2 * F1));
+              for (int p = 0; p < 10; ++p)
+                acc[p] = _mm512_dpbf16_ps(acc[p],
+                                          bcast_pair(row + S1 * p + 2 * kp),
+                                          wv);
+            }
+          }
+        }
+        for (int p = 0; p < 10; ++p) {
+          const __m512 v = _mm512_max_ps(_mm512_add_ps(acc[p], bia), zero);
+          uint16_t hb[F1];
+          _mm256_storeu_si256((__m256i*)hb,
+                              (__m256i)_mm512_cvtneps_pbh(v));
+          const int px = oy * W1 + ox + p;
+          for (int f = 0; f < F1; ++f) O[f * (H1 * W1) + px] = hb[f];
+        }
+      }
+    }
+  }
+
+  __attribute__((target("avx512f,avx512bf16"))) void stage_conv2_bf16() {
+    const uint16_t* __restrict W = w2p.data();
+    const uint16_t* __restrict I = a1_h.data();
+    uint16_t* __restrict O = a2_h.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bl = _mm512_loadu_ps(b2.data());
+    const __m512 bh = _mm512_loadu_ps(b2.data() + 16);
+    for (int oy2 = 0; oy2 < H2; ++oy2) {
+      __m512 acc[9][2];
+      for (int p = 0; p < 9; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
+      const int rbase = S2 * oy2 * W1;
+      for (int c = 0; c < F1; ++c) {
+        const uint16_t* __restrict ic = I + c * (H1 * W1);
+        for (int ky = 0; ky < K2; ++ky) {
+          const uint16_t* __restrict wr =
+              W + (((size_t)c * K2 + ky) * (K2 / 2)) * (2 * F2);
+          const int roff = rbase + ky * W1;
+          for (int kp = 0; kp < K2 / 2; ++kp) {
+            const __m512bh w0 = (__m512bh)_mm512_loadu_si512(
+                (const void*)(wr + kp * 2 * F2));
+            const __m512bh w1v = (__m512bh)_mm512_loadu_si512(
+                (const void*)(wr + kp * 2 * F2 + 32));
+            for (int p = 0; p < 9; ++p) {
+              const __m512bh x =
+                  bcast_pair(ic + roff + S2 * p + 2 * kp);
+              acc[p][0] = _mm512_dpbf16_ps(acc[p][0], x, w0);
+              acc[p][1] = _mm512_dpbf16_ps(acc[p][1], x, w1v);
+            }
+          }
+        }
+      }
+      for (int p = 0; p < 9; ++p) {
+        const __m512 v0 = _mm512_max_ps(_mm512_add_ps(acc[p][0], bl), zero);
+        const __m512 v1 = _mm512_max_ps(_mm512_add_ps(acc[p][1], bh), zero);
+        uint16_t hb[F2];
+        _mm256_storeu_si256((__m256i*)hb, (__m256i)_mm512_cvtneps_pbh(v0));
+        _mm256_storeu_si256((__m256i*)(hb + 16),
+                            (__m256i)_mm512_cvtneps_pbh(v1));
+        const int px = oy2 * W2 + p;
+        for (int f = 0; f < F2; ++f) O[f * (H2 * W2) + px] = hb[f];
+      }
+    }
+  }
+
+  __attribute__((target("avx512f,avx512bf16"))) void stage_conv3_bf16() {
+    const uint16_t* __restrict W = w3p.data();
+    const uint16_t* __restrict I = a2_h.data();
+    uint16_t* __restrict O = a3_h.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bl = _mm512_loadu_ps(b3.data());
+    const __m512 bh = _mm512_loadu_ps(b3.data() + 16);
+    constexpr int KP3 = (K3 + 1) / 2;  // 2: pair (0,1) + (2, zero-pad)
+    for (int oy3 = 0; oy3 < H3; ++oy3) {
+      __m512 acc[7][2];
+      for (int p = 0; p < 7; ++p) { acc[p][0] = zero; acc[p][1] = zero; }
+      for (int c = 0; c < F2; ++c) {
+        const uint16_t* __restrict ic = I + c * (H2 * W2);
+        for (int ky = 0; ky < K3; ++ky) {
+          const uint16_t* __restrict wr =
+              W + (((size_t)c * K3 + ky) * KP3) * (2 * F3);
+          const uint16_t* __restrict row = ic + (oy3 + ky) * W2;
+          for (int kp = 0; kp < KP3; ++kp) {
+            const __m512bh w0 = (__m512bh)_mm512_loadu_si512(
+                (const void*)(wr + kp * 2 * F3));
+            const __m512bh w1v = (__m512bh)_mm512_loadu_si512(
+                (const void*)(wr + kp * 2 * F3 + 32));
+            for (int p = 0; p < 7; ++p) {
+              // kp=1 high half is zero weight: the x overread is padded
+              const __m512bh x = bcast_pair(row + p + 2 * kp);
+              acc[p][0] = _mm512_dpbf16_ps(acc[p][0], x, w0);
+              acc[p][1] = _mm512_dpbf16_ps(acc[p][1], x, w1v);
+            }
+          }
+        }
+      }
+      for (int p = 0; p < 7; ++p) {
+        const __m512 v0 = _mm512_max_ps(_mm512_add_ps(acc[p][0], bl), zero);
+        const __m512 v1 = _mm512_max_ps(_mm512_add_ps(acc[p][1], bh), zero);
+        uint16_t hb[F3];
+        _mm256_storeu_si256((__m256i*)hb, (__m256i)_mm512_cvtneps_pbh(v0));
+        _mm256_storeu_si256((__m256i*)(hb + 16),
+                            (__m256i)_mm512_cvtneps_pbh(v1));
+        const int px = oy3 * W3 + p;
+        for (int f = 0; f < F3; ++f) O[f * (H3 * W3) + px] = hb[f];
+      }
+    }
+  }
+
+  __attribute__((target("avx512f,avx512bf16"))) void stage_fc_bf16() {
+    const uint16_t* __restrict X = a3_h.data();
+    const uint16_t* __restrict W = wfcp.data();
+    float* __restrict O = fc_out.data();
+    const __m512 zero = _mm512_setzero_ps();
+    for (int f0 = 0; f0 < FC; f0 += 128) {
+      __m512 acc[8];
+      for (int q = 0; q < 8; ++q)
+        acc[q] = _mm512_loadu_ps(bfc.data() + f0 + 16 * q);
+      for (int kp = 0; kp < FLAT / 2; ++kp) {
+        const __m512bh x = bcast_pair(X + 2 * kp);
+        const uint16_t* __restrict wv = W + ((size_t)kp * FC + f0) * 2;
+        for (int q = 0; q < 8; ++q)
+          acc[q] = _mm512_dpbf16_ps(
+              acc[q],
+              x, (__m512bh)_mm512_loadu_si512((const void*)(wv + 32 * q)));
+      }
+      for (int q = 0; q < 8; ++q)
+        _mm512_storeu_ps(O + f0 + 16 * q, _mm512_max_ps(acc[q], zero));
+    }
+  }
 
   __attribute__((target("avx512f"))) void stage_conv1_512() {
     const float* __restrict W = w1.data();
@@ -265,6 +492,11 @@ struct ShallowActorNet {
   // --- per-frame stages (AVX2 forms) ---
 
   void stage_convert(const unsigned char* __restrict fr) {
+    if (use_bf16) {  // exact: 256-entry LUT of bf16(v/255)
+      uint16_t* __restrict o = img_h.data();
+      for (int i = 0; i < C0 * H0 * W0; ++i) o[i] = u8lut[fr[i]];
+      return;
+    }
     constexpr float inv = 1.0f / 255.0f;
     float* __restrict o = img.data();
     for (int i = 0; i < C0 * H0 * W0; ++i) o[i] = (float)fr[i] * inv;
@@ -835,8 +1067,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "native CPU actor inference (shallow IMPALA net)";
   m.def("avx512_supported",
         [] { return (bool)__builtin_cpu_supports("avx512f"); });
+  m.def("avx512bf16_supported",
+        [] { return (bool)__builtin_cpu_supports("avx512bf16"); });
   py::class_<ShallowActorNet>(m, "ShallowActorNet")
       .def(py::init<int64_t>())
+      .def(py::init<int64_t, bool>())
+      .def_readonly("bf16_active", &ShallowActorNet::use_bf16)
       .def("load_weights", &ShallowActorNet::load_weights)
       .def("load_lstm", &ShallowActorNet::load_lstm)
       .def("forward", &ShallowActorNet::forward)

@@ -72,8 +72,36 @@ def test_deep_maxpool_parity():
     ext = ops.ext_hip()
     x = torch.randn(4, 16, 36, 48, device=_dev()).to(torch.bfloat16)
     want = F.max_pool2d(x.float(), 3, stride=2, padding=1)
-    got = ext.deep_maxpool(x.permute(0, 2, 3, 1).contiguous())
+    got, _ = ext.deep_maxpool(x.permute(0, 2, 3, 1).contiguous())
     assert torch.equal(got.permute(0, 3, 1, 2).float(), want)
+
+
+def test_deep_maxpool_bwd_scatters_by_argmax():
+    """Pool backward == deterministic scatter of dY by the kernel's own
+    argmax (not torch's backward, whose tie-breaking under bf16
+    quantisation ties can legitimately differ)."""
+    torch.manual_seed(12)
+    ext = ops.ext_hip()
+    N, C, H, W = 3, 16, 36, 48
+    HO, WO = (H + 1) // 2, (W + 1) // 2
+    x = torch.randn(N, H, W, C, device=_dev()).to(torch.bfloat16)
+    g = torch.randn(N, HO, WO, C, device=_dev()).to(torch.bfloat16)
+    y, am = ext.deep_maxpool(x.contiguous(), want_argmax=True)
+    dx = ext.deep_maxpool_bwd(g.contiguous(), am, H, W)
+    ref = torch.zeros(N, H, W, C, device=_dev())
+    gf = g.float()
+    for dyy in range(3):
+        for dxx in range(3):
+            k = dyy * 3 + dxx
+            iy = torch.arange(HO, device=_dev()) * 2 - 1 + dyy
+            ix = torch.arange(WO, device=_dev()) * 2 - 1 + dxx
+            oky = (iy >= 0) & (iy < H)
+            okx = (ix >= 0) & (ix < W)
+            sel = (am == k)[:, oky][:, :, okx]
+            ref[:, iy[oky][:, None], ix[okx][None, :], :] += torch.where(
+                sel, gf[:, oky][:, :, okx], torch.zeros((), device=_dev()))
+    torch.testing.assert_close(dx.float(), ref.to(torch.bfloat16).float(),
+                               rtol=1e-2, atol=1e-2)
 
 
 def test_deep_u8_entry_parity():

@@ -702,6 +702,8 @@ struct ShallowActorNet {
   bool has_lstm = false;
   std::vector<float> wih, whh, blstm;  // [263][1024], [256][1024], [1024]
   std::vector<float> gates;            // [1024]
+  // bf16 k-pair packs: [kp][j*2+s] (row stride 2048), fc/tail/hh parts
+  std::vector<uint16_t> wihp_fc, wihp_tail, whhp;
   int lstm_in = 0;
 
   void load_lstm(torch::Tensor w_ih, torch::Tensor w_hh, torch::Tensor b_ih,
@@ -737,6 +739,31 @@ struct ShallowActorNet {
     bpol.assign(A_pad, 0.f);
     std::memcpy(bpol.data(), pb.data_ptr<float>(), sizeof(float) * A);
     gates.assign(1024, 0.f);
+    if (use_bf16) {
+      auto sacc = w_ih.accessor<float, 2>();
+      const int tail = lstm_in - FC;
+      const int tp = (tail + 1) / 2;
+      wihp_fc.assign((size_t)(FC / 2) * 2048, 0);
+      for (int kp = 0; kp < FC / 2; ++kp)
+        for (int j = 0; j < 1024; ++j)
+          for (int s = 0; s < 2; ++s)
+            wihp_fc[((size_t)kp * 1024 + j) * 2 + s] =
+                f2bf(sacc[j][2 * kp + s]);
+      wihp_tail.assign((size_t)tp * 2048, 0);
+      for (int kp = 0; kp < tp; ++kp)
+        for (int j = 0; j < 1024; ++j)
+          for (int s = 0; s < 2; ++s)
+            if (2 * kp + s < tail)
+              wihp_tail[((size_t)kp * 1024 + j) * 2 + s] =
+                  f2bf(sacc[j][FC + 2 * kp + s]);
+      auto hacc = w_hh.accessor<float, 2>();
+      whhp.assign((size_t)(FC / 2) * 2048, 0);
+      for (int kp = 0; kp < FC / 2; ++kp)
+        for (int j = 0; j < 1024; ++j)
+          for (int s = 0; s < 2; ++s)
+            whhp[((size_t)kp * 1024 + j) * 2 + s] =
+                f2bf(hacc[j][2 * kp + s]);
+    }
     has_lstm = true;
   }
 
@@ -905,6 +932,67 @@ struct ShallowActorNet {
   // batched gates GEMM: one j-block of 32 cols x up to 8 frames in
   // registers; weight lines are loaded once per (k, j-block) and shared
   // by the whole frame group => gate-weight traffic /8 per frame.
+  __attribute__((target("avx512f,avx512bf16"))) void gates_gemv_batched_bf16(
+      int nf, const uint16_t* __restrict fc_h,
+      const uint16_t* __restrict tail_h, int tailp,
+      const uint16_t* __restrict hm_h, float* __restrict gates_all) {
+    for (int i0 = 0; i0 < nf; i0 += 8) {
+      const int gn = nf - i0 < 8 ? nf - i0 : 8;
+      for (int j0 = 0; j0 < 1024; j0 += 32) {
+        __m512 acc[8][2];
+        for (int g = 0; g < 8; ++g) {
+          acc[g][0] = _mm512_loadu_ps(blstm.data() + j0);
+          acc[g][1] = _mm512_loadu_ps(blstm.data() + j0 + 16);
+        }
+        const uint16_t* __restrict W = wihp_fc.data() + j0 * 2;
+        for (int kp = 0; kp < FC / 2; ++kp) {
+          const uint16_t* __restrict wv = W + (size_t)kp * 2048;
+          const __m512bh w0 = (__m512bh)_mm512_loadu_si512((const void*)wv);
+          const __m512bh w1 =
+              (__m512bh)_mm512_loadu_si512((const void*)(wv + 32));
+          for (int g = 0; g < gn; ++g) {
+            const __m512bh x =
+                bcast_pair(fc_h + (size_t)(i0 + g) * FC + 2 * kp);
+            acc[g][0] = _mm512_dpbf16_ps(acc[g][0], x, w0);
+            acc[g][1] = _mm512_dpbf16_ps(acc[g][1], x, w1);
+          }
+        }
+        const uint16_t* __restrict Wt = wihp_tail.data() + j0 * 2;
+        for (int kp = 0; kp < tailp; ++kp) {
+          const uint16_t* __restrict wv = Wt + (size_t)kp * 2048;
+          const __m512bh w0 = (__m512bh)_mm512_loadu_si512((const void*)wv);
+          const __m512bh w1 =
+              (__m512bh)_mm512_loadu_si512((const void*)(wv + 32));
+          for (int g = 0; g < gn; ++g) {
+            const __m512bh x =
+                bcast_pair(tail_h + (size_t)(i0 + g) * 64 + 2 * kp);
+            acc[g][0] = _mm512_dpbf16_ps(acc[g][0], x, w0);
+            acc[g][1] = _mm512_dpbf16_ps(acc[g][1], x, w1);
+          }
+        }
+        const uint16_t* __restrict Wh = whhp.data() + j0 * 2;
+        for (int kp = 0; kp < FC / 2; ++kp) {
+          const uint16_t* __restrict wv = Wh + (size_t)kp * 2048;
+          const __m512bh w0 = (__m512bh)_mm512_loadu_si512((const void*)wv);
+          const __m512bh w1 =
+              (__m512bh)_mm512_loadu_si512((const void*)(wv + 32));
+          for (int g = 0; g < gn; ++g) {
+            const __m512bh x =
+                bcast_pair(hm_h + (size_t)(i0 + g) * FC + 2 * kp);
+            acc[g][0] = _mm512_dpbf16_ps(acc[g][0], x, w0);
+            acc[g][1] = _mm512_dpbf16_ps(acc[g][1], x, w1);
+          }
+        }
+        for (int g = 0; g < gn; ++g) {
+          _mm512_storeu_ps(gates_all + (size_t)(i0 + g) * 1024 + j0,
+                           acc[g][0]);
+          _mm512_storeu_ps(gates_all + (size_t)(i0 + g) * 1024 + j0 + 16,
+                           acc[g][1]);
+        }
+      }
+    }
+  }
+
   __attribute__((target("avx512f")))
   void gates_gemv_batched_512(int nf, const float* __restrict fc_all,
                               const float* __restrict tail_all, int tail,
@@ -1013,8 +1101,20 @@ struct ShallowActorNet {
       tails[(size_t)i * 64] = r > 1.f ? 1.f : (r < -1.f ? -1.f : r);
       tails[(size_t)i * 64 + 1 + (int)la[i]] = 1.f;
     }
-    gates_gemv_batched_512(n, fc_all.data(), tails.data(), tail,
-                           hm_all.data(), gates_all.data());
+    if (use_bf16) {
+      std::vector<uint16_t> fc_h((size_t)n * FC), hm_h((size_t)n * FC),
+          tails_h((size_t)n * 64, 0);
+      for (size_t k = 0; k < fc_h.size(); ++k) fc_h[k] = f2bf(fc_all[k]);
+      for (size_t k = 0; k < hm_h.size(); ++k) hm_h[k] = f2bf(hm_all[k]);
+      for (size_t k = 0; k < tails_h.size(); ++k)
+        tails_h[k] = f2bf(tails[k]);
+      gates_gemv_batched_bf16(n, fc_h.data(), tails_h.data(),
+                              (tail + 1) / 2, hm_h.data(),
+                              gates_all.data());
+    } else {
+      gates_gemv_batched_512(n, fc_all.data(), tails.data(), tail,
+                             hm_all.data(), gates_all.data());
+    }
     for (int i = 0; i < n; ++i) {
       std::memcpy(gates.data(), gates_all.data() + (size_t)i * 1024,
                   sizeof(float) * 1024);

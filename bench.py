@@ -36,7 +36,9 @@ def main():
     p.add_argument("--num-actors", type=int, default=32)
     p.add_argument("--envs-per-actor", type=int, default=16)
     p.add_argument("--actor-threads", type=int, default=0)
-    p.add_argument("--actor-jit", default="native", choices=["off", "safe", "frozen", "native"])
+    p.add_argument("--actor-jit", default="native-bf16",
+                   choices=["off", "safe", "frozen", "native",
+                            "native-bf16"])
     p.add_argument("--use-lstm", action="store_true", default=False)
     p.add_argument("--model", default="auto")
     p.add_argument("--no-hip-kernels", action="store_true")

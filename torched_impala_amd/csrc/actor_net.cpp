@@ -54,6 +54,7 @@ struct ShallowActorNet {
       : rng((uint64_t)seed) {
     use512 = __builtin_cpu_supports("avx512f");
     use_bf16 = want_bf16 && __builtin_cpu_supports("avx512bf16");
+    use_vnni = use_bf16 && __builtin_cpu_supports("avx512vnni");
     img.resize(C0 * H0 * W0);
     a1.resize(F1 * H1 * W1);
     a2.resize(F2 * H2 * W2);
@@ -134,6 +135,27 @@ struct ShallowActorNet {
       pack_pairs(w1p, cw1, F1, C0, K1);
       pack_pairs(w2p, cw2, F2, F1, K2);
       pack_pairs(w3p, cw3, F3, F2, K3);
+      if (use_vnni) {
+        auto s = cw1.accessor<float, 4>();
+        w1scale.assign(F1, 0.f);
+        float m[F1] = {0};
+        for (int f = 0; f < F1; ++f)
+          for (int c = 0; c < C0; ++c)
+            for (int ky = 0; ky < K1; ++ky)
+              for (int kx = 0; kx < K1; ++kx)
+                m[f] = std::max(m[f], std::fabs(s[f][c][ky][kx]));
+        w1q.assign((size_t)C0 * K1 * 2 * F1 * 4, 0);
+        for (int f = 0; f < F1; ++f) {
+          const float inv = m[f] > 0 ? 127.f / m[f] : 0.f;
+          w1scale[f] = m[f] / (127.f * 255.f);
+          for (int c = 0; c < C0; ++c)
+            for (int ky = 0; ky < K1; ++ky)
+              for (int kg = 0; kg < 2; ++kg)
+                for (int j = 0; j < 4; ++j)
+                  w1q[((((size_t)c * K1 + ky) * 2 + kg) * F1 + f) * 4 + j] =
+                      (int8_t)std::lround(s[f][c][ky][4 * kg + j] * inv);
+        }
+      }
       // fc: k-PAIRS, lane f = (w[2kp][f], w[2kp+1][f]); FLAT is even.
       wfcp.assign((size_t)(FLAT / 2) * 2 * FC, 0);
       auto s = fw.accessor<float, 2>();
@@ -176,10 +198,63 @@ struct ShallowActorNet {
   std::vector<uint16_t> img_h, a1_h, a2_h, a3_h;
   std::vector<uint16_t> w1p, w2p, w3p, wfcp;
   uint16_t u8lut[256];
+  // int8 conv1 (VNNI): s8 weights in kx-QUADS, per-filter scales; the
+  // u8 frame is consumed raw (exact), so only weights are quantised
+  bool use_vnni = false;
+  std::vector<int8_t> w1q;        // [c][ky][kxg][f*4+j]
+  std::vector<float> w1scale;     // [F1] = max|w|/(127*255)
+  const unsigned char* cur_frame = nullptr;
 
   void stage_conv1() {
-    use_bf16 ? stage_conv1_bf16()
-             : (use512 ? stage_conv1_512() : stage_conv1_avx2());
+    use_vnni ? stage_conv1_vnni()
+             : use_bf16 ? stage_conv1_bf16()
+                        : (use512 ? stage_conv1_512() : stage_conv1_avx2());
+  }
+
+  // conv1 on vpdpbusd: raw u8 frame x s8 weights, 4 kx-taps per instr;
+  // per-filter dequant scale folds the /255 normalisation.
+  __attribute__((target("avx512f,avx512vnni,avx512bf16")))
+  void stage_conv1_vnni() {
+    const int8_t* __restrict W = w1q.data();
+    const unsigned char* __restrict I = cur_frame;
+    uint16_t* __restrict O = a1_h.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bia = _mm512_loadu_ps(b1.data());
+    const __m512 scl = _mm512_loadu_ps(w1scale.data());
+    for (int oy = 0; oy < H1; ++oy) {
+      for (int ox = 0; ox < W1; ox += 10) {
+        __m512i acc[10];
+        for (int p = 0; p < 10; ++p) acc[p] = _mm512_setzero_si512();
+        for (int c = 0; c < C0; ++c) {
+          const unsigned char* __restrict ic = I + c * (H0 * W0);
+          for (int ky = 0; ky < K1; ++ky) {
+            const unsigned char* __restrict row =
+                ic + (S1 * oy + ky) * W0 + S1 * ox;
+            const int8_t* __restrict wr =
+                W + (((size_t)c * K1 + ky) * 2) * (F1 * 4);
+            for (int kg = 0; kg < 2; ++kg) {
+              const __m512i wv =
+                  _mm512_loadu_si512((const void*)(wr + kg * F1 * 4));
+              for (int p = 0; p < 10; ++p) {
+                uint32_t xw;
+                std::memcpy(&xw, row + S1 * p + 4 * kg, 4);
+                acc[p] = _mm512_dpbusd_epi32(acc[p], _mm512_set1_epi32(xw),
+                                             wv);
+              }
+            }
+          }
+        }
+        for (int p = 0; p < 10; ++p) {
+          const __m512 v = _mm512_max_ps(
+              _mm512_fmadd_ps(_mm512_cvtepi32_ps(acc[p]), scl, bia), zero);
+          uint16_t hb[F1];
+          _mm256_storeu_si256((__m256i*)hb,
+                              (__m256i)_mm512_cvtneps_pbh(v));
+          const int px = oy * W1 + ox + p;
+          for (int f = 0; f < F1; ++f) O[f * (H1 * W1) + px] = hb[f];
+        }
+      }
+    }
   }
   void stage_conv2() {
     use_bf16 ? stage_conv2_bf16()
@@ -492,6 +567,8 @@ struct ShallowActorNet {
   // --- per-frame stages (AVX2 forms) ---
 
   void stage_convert(const unsigned char* __restrict fr) {
+    cur_frame = fr;
+    if (use_vnni) return;  // conv1 consumes the raw u8 frame directly
     if (use_bf16) {  // exact: 256-entry LUT of bf16(v/255)
       uint16_t* __restrict o = img_h.data();
       for (int i = 0; i < C0 * H0 * W0; ++i) o[i] = u8lut[fr[i]];

@@ -67,6 +67,11 @@ struct ShallowActorNet {
       a3_h.resize(FLAT);
       for (int v = 0; v < 256; ++v) u8lut[v] = f2bf((float)v * (1.f / 255.f));
     }
+    if (use_vnni) {
+      a1_u8.resize(F1 * H1 * W1 + 4);
+      a2_u8.resize(F2 * H2 * W2 + 4);  // +4: conv3's quad overread
+      a3_u8.resize(FLAT);
+    }
   }
 
   static uint16_t f2bf(float f) {  // round-nearest-even fp32 -> bf16
@@ -136,6 +141,55 @@ struct ShallowActorNet {
       pack_pairs(w2p, cw2, F2, F1, K2);
       pack_pairs(w3p, cw3, F3, F2, K3);
       if (use_vnni) {
+        // kx-QUAD s8 packs for conv2 (K=4: one quad/row), conv3 (K=3:
+        // quad with zero 4th tap) and fc (k quads); per-filter scales.
+        auto quant_conv = [](std::vector<int8_t>& dst,
+                             std::vector<float>& scale,
+                             const torch::Tensor& src, int F, int C,
+                             int K) {
+          const int KG = (K + 3) / 4;
+          auto s = src.accessor<float, 4>();
+          scale.assign(F, 0.f);
+          std::vector<float> m(F, 0.f);
+          for (int f = 0; f < F; ++f)
+            for (int c = 0; c < C; ++c)
+              for (int ky = 0; ky < K; ++ky)
+                for (int kx = 0; kx < K; ++kx)
+                  m[f] = std::max(m[f], std::fabs(s[f][c][ky][kx]));
+          dst.assign((size_t)C * K * KG * F * 4, 0);
+          for (int f = 0; f < F; ++f) {
+            const float inv = m[f] > 0 ? 127.f / m[f] : 0.f;
+            scale[f] = m[f] / 127.f;
+            for (int c = 0; c < C; ++c)
+              for (int ky = 0; ky < K; ++ky)
+                for (int kg = 0; kg < KG; ++kg)
+                  for (int j = 0; j < 4; ++j) {
+                    const int kx = 4 * kg + j;
+                    if (kx < K)
+                      dst[((((size_t)c * K + ky) * KG + kg) * F + f) * 4 +
+                          j] = (int8_t)std::lround(s[f][c][ky][kx] * inv);
+                  }
+          }
+        };
+        quant_conv(w2q, w2scale, cw2, F2, F1, K2);
+        quant_conv(w3q, w3scale, cw3, F3, F2, K3);
+        {
+          auto s = fw.accessor<float, 2>();
+          wfcscale.assign(FC, 0.f);
+          std::vector<float> m(FC, 0.f);
+          for (int f = 0; f < FC; ++f)
+            for (int k = 0; k < FLAT; ++k)
+              m[f] = std::max(m[f], std::fabs(s[f][k]));
+          wfcq.assign((size_t)(FLAT / 4) * FC * 4, 0);
+          for (int f = 0; f < FC; ++f) {
+            const float inv = m[f] > 0 ? 127.f / m[f] : 0.f;
+            wfcscale[f] = m[f] / 127.f;
+            for (int kg = 0; kg < FLAT / 4; ++kg)
+              for (int j = 0; j < 4; ++j)
+                wfcq[((size_t)kg * FC + f) * 4 + j] =
+                    (int8_t)std::lround(s[f][4 * kg + j] * inv);
+          }
+        }
         auto s = cw1.accessor<float, 4>();
         w1scale.assign(F1, 0.f);
         float m[F1] = {0};
@@ -204,6 +258,12 @@ struct ShallowActorNet {
   std::vector<int8_t> w1q;        // [c][ky][kxg][f*4+j]
   std::vector<float> w1scale;     // [F1] = max|w|/(127*255)
   const unsigned char* cur_frame = nullptr;
+  // int8 chain for conv2/3/fc: ReLU outputs requantised to u8 with a
+  // per-frame dynamic scale (zero-point 0, exact for non-negatives)
+  std::vector<int8_t> w2q, w3q, wfcq;   // kx-quad packs [..][f*4+j]
+  std::vector<float> w2scale, w3scale, wfcscale;  // max|w|/127 per f
+  std::vector<uint8_t> a1_u8, a2_u8, a3_u8;
+  float a1_m = 0.f, a2_m = 0.f, a3_m = 0.f;  // per-frame act maxima
 
   void stage_conv1() {
     use_vnni ? stage_conv1_vnni()
@@ -217,10 +277,11 @@ struct ShallowActorNet {
   void stage_conv1_vnni() {
     const int8_t* __restrict W = w1q.data();
     const unsigned char* __restrict I = cur_frame;
-    uint16_t* __restrict O = a1_h.data();
+    float* __restrict O = a1.data();
     const __m512 zero = _mm512_setzero_ps();
     const __m512 bia = _mm512_loadu_ps(b1.data());
     const __m512 scl = _mm512_loadu_ps(w1scale.data());
+    __m512 mx = zero;
     for (int oy = 0; oy < H1; ++oy) {
       for (int ox = 0; ox < W1; ox += 10) {
         __m512i acc[10];
@@ -247,26 +308,186 @@ struct ShallowActorNet {
         for (int p = 0; p < 10; ++p) {
           const __m512 v = _mm512_max_ps(
               _mm512_fmadd_ps(_mm512_cvtepi32_ps(acc[p]), scl, bia), zero);
-          uint16_t hb[F1];
-          _mm256_storeu_si256((__m256i*)hb,
-                              (__m256i)_mm512_cvtneps_pbh(v));
+          mx = _mm512_max_ps(mx, v);
+          float buf[F1];
+          _mm512_storeu_ps(buf, v);
           const int px = oy * W1 + ox + p;
-          for (int f = 0; f < F1; ++f) O[f * (H1 * W1) + px] = hb[f];
+          for (int f = 0; f < F1; ++f) O[f * (H1 * W1) + px] = buf[f];
         }
+      }
+    }
+    a1_m = _mm512_reduce_max_ps(mx);
+    requant(a1.data(), F1 * H1 * W1, a1_m, a1_u8.data());
+  }
+
+  // u8 requant of a non-negative activation plane: u = round(v*255/m)
+  __attribute__((target("avx512f")))
+  static void requant(const float* __restrict a, int n, float m,
+                      uint8_t* __restrict o) {
+    const float s = m > 0.f ? 255.f / m : 0.f;
+    const __m512 sv = _mm512_set1_ps(s);
+    int i = 0;
+    for (; i + 16 <= n; i += 16) {
+      const __m512i q = _mm512_cvtps_epi32(
+          _mm512_mul_ps(_mm512_loadu_ps(a + i), sv));
+      _mm_storeu_si128((__m128i*)(o + i), _mm512_cvtusepi32_epi8(q));
+    }
+    for (; i < n; ++i)
+      o[i] = (uint8_t)std::lround(std::min(a[i] * s, 255.f));
+  }
+
+  __attribute__((target("avx512f,avx512vnni"))) void stage_conv2_vnni() {
+    const int8_t* __restrict W = w2q.data();
+    const uint8_t* __restrict I = a1_u8.data();
+    float* __restrict O = a2.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bl = _mm512_loadu_ps(b2.data());
+    const __m512 bh = _mm512_loadu_ps(b2.data() + 16);
+    const __m512 aml = _mm512_set1_ps(a1_m * (1.f / 255.f));
+    const __m512 s0 = _mm512_mul_ps(_mm512_loadu_ps(w2scale.data()), aml);
+    const __m512 s1 = _mm512_mul_ps(_mm512_loadu_ps(w2scale.data() + 16),
+                                    aml);
+    __m512 mx = zero;
+    for (int oy2 = 0; oy2 < H2; ++oy2) {
+      __m512i acc[9][2];
+      for (int p = 0; p < 9; ++p) {
+        acc[p][0] = _mm512_setzero_si512();
+        acc[p][1] = _mm512_setzero_si512();
+      }
+      const int rbase = S2 * oy2 * W1;
+      for (int c = 0; c < F1; ++c) {
+        const uint8_t* __restrict ic = I + c * (H1 * W1);
+        for (int ky = 0; ky < K2; ++ky) {
+          // K2 = 4: exactly one quad per (c,ky)
+          const int8_t* __restrict wr =
+              W + (((size_t)c * K2 + ky) * 1) * (F2 * 4);
+          const __m512i w0 = _mm512_loadu_si512((const void*)wr);
+          const __m512i w1v =
+              _mm512_loadu_si512((const void*)(wr + 16 * 4));
+          const int roff = rbase + ky * W1;
+          for (int p = 0; p < 9; ++p) {
+            uint32_t xw;
+            std::memcpy(&xw, ic + roff + S2 * p, 4);
+            const __m512i x = _mm512_set1_epi32(xw);
+            acc[p][0] = _mm512_dpbusd_epi32(acc[p][0], x, w0);
+            acc[p][1] = _mm512_dpbusd_epi32(acc[p][1], x, w1v);
+          }
+        }
+      }
+      for (int p = 0; p < 9; ++p) {
+        const __m512 v0 = _mm512_max_ps(
+            _mm512_fmadd_ps(_mm512_cvtepi32_ps(acc[p][0]), s0, bl), zero);
+        const __m512 v1 = _mm512_max_ps(
+            _mm512_fmadd_ps(_mm512_cvtepi32_ps(acc[p][1]), s1, bh), zero);
+        mx = _mm512_max_ps(mx, _mm512_max_ps(v0, v1));
+        float buf[F2];
+        _mm512_storeu_ps(buf, v0);
+        _mm512_storeu_ps(buf + 16, v1);
+        const int px = oy2 * W2 + p;
+        for (int f = 0; f < F2; ++f) O[f * (H2 * W2) + px] = buf[f];
+      }
+    }
+    a2_m = _mm512_reduce_max_ps(mx);
+    requant(a2.data(), F2 * H2 * W2, a2_m, a2_u8.data());
+  }
+
+  __attribute__((target("avx512f,avx512vnni"))) void stage_conv3_vnni() {
+    const int8_t* __restrict W = w3q.data();
+    const uint8_t* __restrict I = a2_u8.data();
+    float* __restrict O = a3.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 bl = _mm512_loadu_ps(b3.data());
+    const __m512 bh = _mm512_loadu_ps(b3.data() + 16);
+    const __m512 aml = _mm512_set1_ps(a2_m * (1.f / 255.f));
+    const __m512 s0 = _mm512_mul_ps(_mm512_loadu_ps(w3scale.data()), aml);
+    const __m512 s1 = _mm512_mul_ps(_mm512_loadu_ps(w3scale.data() + 16),
+                                    aml);
+    __m512 mx = zero;
+    for (int oy3 = 0; oy3 < H3; ++oy3) {
+      __m512i acc[7][2];
+      for (int p = 0; p < 7; ++p) {
+        acc[p][0] = _mm512_setzero_si512();
+        acc[p][1] = _mm512_setzero_si512();
+      }
+      for (int c = 0; c < F2; ++c) {
+        const uint8_t* __restrict ic = I + c * (H2 * W2);
+        for (int ky = 0; ky < K3; ++ky) {
+          // K3 = 3: one quad with a zero 4th tap (x overread padded)
+          const int8_t* __restrict wr =
+              W + (((size_t)c * K3 + ky) * 1) * (F3 * 4);
+          const __m512i w0 = _mm512_loadu_si512((const void*)wr);
+          const __m512i w1v =
+              _mm512_loadu_si512((const void*)(wr + 16 * 4));
+          const uint8_t* __restrict row = ic + (oy3 + ky) * W2;
+          for (int p = 0; p < 7; ++p) {
+            uint32_t xw;
+            std::memcpy(&xw, row + p, 4);
+            const __m512i x = _mm512_set1_epi32(xw);
+            acc[p][0] = _mm512_dpbusd_epi32(acc[p][0], x, w0);
+            acc[p][1] = _mm512_dpbusd_epi32(acc[p][1], x, w1v);
+          }
+        }
+      }
+      for (int p = 0; p < 7; ++p) {
+        const __m512 v0 = _mm512_max_ps(
+            _mm512_fmadd_ps(_mm512_cvtepi32_ps(acc[p][0]), s0, bl), zero);
+        const __m512 v1 = _mm512_max_ps(
+            _mm512_fmadd_ps(_mm512_cvtepi32_ps(acc[p][1]), s1, bh), zero);
+        mx = _mm512_max_ps(mx, _mm512_max_ps(v0, v1));
+        float buf[F3];
+        _mm512_storeu_ps(buf, v0);
+        _mm512_storeu_ps(buf + 16, v1);
+        const int px = oy3 * W3 + p;
+        for (int f = 0; f < F3; ++f) O[f * (H3 * W3) + px] = buf[f];
+      }
+    }
+    a3_m = _mm512_reduce_max_ps(mx);
+    requant(a3.data(), FLAT, a3_m, a3_u8.data());
+  }
+
+  __attribute__((target("avx512f,avx512vnni"))) void stage_fc_vnni() {
+    const uint8_t* __restrict X = a3_u8.data();
+    const int8_t* __restrict W = wfcq.data();
+    float* __restrict O = fc_out.data();
+    const __m512 zero = _mm512_setzero_ps();
+    const __m512 aml = _mm512_set1_ps(a3_m * (1.f / 255.f));
+    for (int f0 = 0; f0 < FC; f0 += 128) {
+      __m512i acc[8];
+      for (int q = 0; q < 8; ++q) acc[q] = _mm512_setzero_si512();
+      for (int kg = 0; kg < FLAT / 4; ++kg) {
+        uint32_t xw;
+        std::memcpy(&xw, X + 4 * kg, 4);
+        const __m512i x = _mm512_set1_epi32(xw);
+        const int8_t* __restrict wv = W + ((size_t)kg * FC + f0) * 4;
+        for (int q = 0; q < 8; ++q)
+          acc[q] = _mm512_dpbusd_epi32(
+              acc[q], x, _mm512_loadu_si512((const void*)(wv + 64 * q)));
+      }
+      for (int q = 0; q < 8; ++q) {
+        const __m512 scl = _mm512_mul_ps(
+            _mm512_loadu_ps(wfcscale.data() + f0 + 16 * q), aml);
+        const __m512 v = _mm512_max_ps(
+            _mm512_fmadd_ps(_mm512_cvtepi32_ps(acc[q]), scl,
+                            _mm512_loadu_ps(bfc.data() + f0 + 16 * q)),
+            zero);
+        _mm512_storeu_ps(O + f0 + 16 * q, v);
       }
     }
   }
   void stage_conv2() {
-    use_bf16 ? stage_conv2_bf16()
-             : (use512 ? stage_conv2_512() : stage_conv2_avx2());
+    use_vnni ? stage_conv2_vnni()
+             : use_bf16 ? stage_conv2_bf16()
+                        : (use512 ? stage_conv2_512() : stage_conv2_avx2());
   }
   void stage_conv3() {
-    use_bf16 ? stage_conv3_bf16()
-             : (use512 ? stage_conv3_512() : stage_conv3_avx2());
+    use_vnni ? stage_conv3_vnni()
+             : use_bf16 ? stage_conv3_bf16()
+                        : (use512 ? stage_conv3_512() : stage_conv3_avx2());
   }
   void stage_fc() {
-    use_bf16 ? stage_fc_bf16()
-             : (use512 ? stage_fc_512() : stage_fc_avx2());
+    use_vnni ? stage_fc_vnni()
+             : use_bf16 ? stage_fc_bf16()
+                        : (use512 ? stage_fc_512() : stage_fc_avx2());
   }
 
   // --- bf16 stages (AVX512-BF16 vdpbf16ps: 2 MACs per lane per instr,

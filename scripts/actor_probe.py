@@ -44,7 +44,8 @@ def main():
     p.add_argument("--seconds", type=float, default=8.0)
     p.add_argument("--use-lstm", action="store_true")
     p.add_argument("--threads", type=int, default=1)
-    p.add_argument("--jit", default="native", choices=["off", "safe", "frozen", "native"])
+    p.add_argument("--jit", default="native-bf16",
+                   choices=["off", "safe", "frozen", "native", "native-bf16"])
     args = p.parse_args()
     import torch.multiprocessing as mp
     ctx = mp.get_context("spawn")

@@ -30,6 +30,11 @@ extern "C" void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int,
                                         hipStream_t);
 extern "C" void launch_mfma_probe(const void*, const void*, float*,
                                   hipStream_t);
+extern "C" void launch_shallow_mfma_fwd(const unsigned char*, const void*,
+                                        const float*, const void*,
+                                        const float*, const void*,
+                                        const float*, void*, void*, void*,
+                                        int, hipStream_t);
 
 #define CHECK(x)                                                   \
   do {                                                             \
@@ -541,6 +546,124 @@ int main(int argc, char** argv) {
                         std::fabs(bf2f(out[i]) - bf2f(f2bf(ref[i]))));
     report("maxpool bwd (argmax scatter==gather)", maxerr, 1e-6f);
     hipFree(dxp); hipFree(dyp2); hipFree(dip); hipFree(dgp); hipFree(dop);
+  }
+
+  // ---- 8. shallow encoder on MFMA (shallow_mfma.hip) -------------------
+  {
+    const int N = 3;
+    const int C0 = 4, H0 = 84, W0 = 84;
+    const int F1 = 16, K1 = 8, S1 = 4, H1 = 20, W1 = 20;
+    const int F2 = 32, K2 = 4, S2 = 2, H2 = 9, W2 = 9;
+    const int F3 = 32, K3 = 3, H3 = 7, W3 = 7;
+    unsigned seed = 1234u;
+    std::vector<unsigned char> fr((size_t)N * C0 * H0 * W0);
+    for (auto& v : fr) { seed = seed * 1664525u + 1013904223u; v = seed >> 24; }
+    // weights: packed layouts (k per header comment), bf16
+    std::vector<u16> w1((size_t)F1 * 256), w2((size_t)F2 * 256),
+        w3((size_t)F3 * 288);
+    std::vector<float> b1(F1), b2(F2), b3(F3);
+    for (auto& v : w1) v = f2bf(frand(seed) * 0.1f);
+    for (auto& v : w2) v = f2bf(frand(seed) * 0.1f);
+    for (auto& v : w3) v = f2bf(frand(seed) * 0.1f);
+    for (auto& v : b1) v = frand(seed) * 0.1f;
+    for (auto& v : b2) v = frand(seed) * 0.1f;
+    for (auto& v : b3) v = frand(seed) * 0.1f;
+    // CPU reference chain (bf16 weights/activations, fp32 accumulate)
+    std::vector<float> r1((size_t)N * F1 * H1 * W1),
+        r2((size_t)N * F2 * H2 * W2), r3((size_t)N * F3 * H3 * W3);
+    for (int n = 0; n < N; ++n) {
+      for (int f = 0; f < F1; ++f)
+        for (int oy = 0; oy < H1; ++oy)
+          for (int ox = 0; ox < W1; ++ox) {
+            float acc = b1[f];
+            for (int c = 0; c < C0; ++c)
+              for (int ky = 0; ky < K1; ++ky)
+                for (int kx = 0; kx < K1; ++kx) {
+                  const float x = bf2f(f2bf(
+                      fr[((size_t)(n * C0 + c) * H0 + S1 * oy + ky) * W0 +
+                         S1 * ox + kx] * (1.f / 255.f)));
+                  acc += x * bf2f(w1[f * 256 + c * 64 + ky * 8 + kx]);
+                }
+            r1[((size_t)(n * F1 + f) * H1 + oy) * W1 + ox] =
+                bf2f(f2bf(acc > 0 ? acc : 0));
+          }
+      for (int f = 0; f < F2; ++f)
+        for (int oy = 0; oy < H2; ++oy)
+          for (int ox = 0; ox < W2; ++ox) {
+            float acc = b2[f];
+            for (int c = 0; c < F1; ++c)
+              for (int ky = 0; ky < K2; ++ky)
+                for (int kx = 0; kx < K2; ++kx)
+                  acc += r1[((size_t)(n * F1 + c) * H1 + S2 * oy + ky) * W1 +
+                            S2 * ox + kx] *
+                         bf2f(w2[f * 256 + c * 16 + ky * 4 + kx]);
+            r2[((size_t)(n * F2 + f) * H2 + oy) * W2 + ox] =
+                bf2f(f2bf(acc > 0 ? acc : 0));
+          }
+      for (int f = 0; f < F3; ++f)
+        for (int oy = 0; oy < H3; ++oy)
+          for (int ox = 0; ox < W3; ++ox) {
+            float acc = b3[f];
+            for (int c = 0; c < F2; ++c)
+              for (int ky = 0; ky < K3; ++ky)
+                for (int kx = 0; kx < K3; ++kx)
+                  acc += r2[((size_t)(n * F2 + c) * H2 + oy + ky) * W2 + ox +
+                            kx] *
+                         bf2f(w3[f * 288 + (ky * 3 + kx) * 32 + c]);
+            r3[(size_t)n * (F3 * H3 * W3) + f * (H3 * W3) + oy * W3 + ox] =
+                acc > 0 ? acc : 0;
+          }
+    }
+    void *dfr, *dw1, *db1, *dw2, *db2, *dw3, *db3, *da1, *da2, *da3;
+    CHECK(hipMalloc(&dfr, fr.size()));
+    CHECK(hipMalloc(&dw1, w1.size() * 2));
+    CHECK(hipMalloc(&db1, F1 * 4));
+    CHECK(hipMalloc(&dw2, w2.size() * 2));
+    CHECK(hipMalloc(&db2, F2 * 4));
+    CHECK(hipMalloc(&dw3, w3.size() * 2));
+    CHECK(hipMalloc(&db3, F3 * 4));
+    CHECK(hipMalloc(&da1, (size_t)N * H1 * W1 * F1 * 2));
+    CHECK(hipMalloc(&da2, (size_t)N * H2 * W2 * F2 * 2));
+    CHECK(hipMalloc(&da3, (size_t)N * F3 * H3 * W3 * 2));
+    CHECK(hipMemcpy(dfr, fr.data(), fr.size(), hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dw1, w1.data(), w1.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(db1, b1.data(), F1 * 4, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dw2, w2.data(), w2.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(db2, b2.data(), F2 * 4, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dw3, w3.data(), w3.size() * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(db3, b3.data(), F3 * 4, hipMemcpyHostToDevice));
+    launch_shallow_mfma_fwd((const unsigned char*)dfr, dw1, (float*)db1,
+                            dw2, (float*)db2, dw3, (float*)db3, da1, da2,
+                            da3, N, 0);
+    CHECK(hipDeviceSynchronize());
+    // a1 NHWC vs reference NCHW
+    std::vector<u16> o1((size_t)N * H1 * W1 * F1),
+        o2((size_t)N * H2 * W2 * F2), o3((size_t)N * F3 * H3 * W3);
+    CHECK(hipMemcpy(o1.data(), da1, o1.size() * 2, hipMemcpyDeviceToHost));
+    CHECK(hipMemcpy(o2.data(), da2, o2.size() * 2, hipMemcpyDeviceToHost));
+    CHECK(hipMemcpy(o3.data(), da3, o3.size() * 2, hipMemcpyDeviceToHost));
+    float e1 = 0, e2 = 0, e3 = 0, m3 = 1;
+    for (int n = 0; n < N; ++n)
+      for (int f = 0; f < F1; ++f)
+        for (int p = 0; p < H1 * W1; ++p)
+          e1 = std::max(e1, std::fabs(
+              bf2f(o1[((size_t)n * H1 * W1 + p) * F1 + f]) -
+              r1[(size_t)(n * F1 + f) * H1 * W1 + p]));
+    for (int n = 0; n < N; ++n)
+      for (int f = 0; f < F2; ++f)
+        for (int p = 0; p < H2 * W2; ++p)
+          e2 = std::max(e2, std::fabs(
+              bf2f(o2[((size_t)n * H2 * W2 + p) * F2 + f]) -
+              r2[(size_t)(n * F2 + f) * H2 * W2 + p]));
+    for (size_t i = 0; i < o3.size(); ++i) {
+      m3 = std::max(m3, std::fabs(r3[i]));
+      e3 = std::max(e3, std::fabs(bf2f(o3[i]) - r3[i]));
+    }
+    report("shallow mfma conv1 (u8 -> a1 nhwc)", e1, 2e-2f);
+    report("shallow mfma conv2 (a1 -> a2 nhwc)", e2, 2e-2f);
+    report("shallow mfma conv3 (a2 -> a3 flat)", e3 / m3, 2e-2f);
+    hipFree(dfr); hipFree(dw1); hipFree(db1); hipFree(dw2); hipFree(db2);
+    hipFree(dw3); hipFree(db3); hipFree(da1); hipFree(da2); hipFree(da3);
   }
 
   printf(n_fail ? "DEEP_PROBE FAIL (%d)\n" : "DEEP_PROBE ALL PASS\n", n_fail);

@@ -35,6 +35,14 @@ extern "C" void launch_shallow_mfma_fwd(const unsigned char*, const void*,
                                         const float*, const void*,
                                         const float*, void*, void*, void*,
                                         int, hipStream_t);
+extern "C" int shallow_mfma_slab();
+extern "C" void launch_shallow_mask_convert(const float*, const void*,
+                                            void*, long, hipStream_t);
+extern "C" void launch_shallow_mfma_bwd(const unsigned char*, const void*,
+                                        const void*, const void*,
+                                        const void*, const void*, void*,
+                                        void*, float*, float*, float*, int,
+                                        hipStream_t);
 
 #define CHECK(x)                                                   \
   do {                                                             \
@@ -662,6 +670,205 @@ int main(int argc, char** argv) {
     report("shallow mfma conv1 (u8 -> a1 nhwc)", e1, 2e-2f);
     report("shallow mfma conv2 (a1 -> a2 nhwc)", e2, 2e-2f);
     report("shallow mfma conv3 (a2 -> a3 flat)", e3 / m3, 2e-2f);
+
+    // ---- 9. shallow encoder backward --------------------------------
+    // Uses the fwd section's activations (o1 NHWC, o2 NHWC, o3 flat,
+    // all on device) and CPU references r1/r2/r3.
+    {
+      std::vector<float> d3f((size_t)N * F3 * H3 * W3);
+      for (auto& v : d3f) v = frand(seed);
+      // CPU: dz3 = d3 .* (r3>0), bf16
+      std::vector<u16> dz3h(d3f.size());
+      for (size_t i = 0; i < d3f.size(); ++i)
+        dz3h[i] = f2bf(r3[i] > 0.f ? d3f[i] : 0.f);
+      // CPU d_a2 (NCHW layout [c][py][px] per img), mask r2>0
+      std::vector<float> dA2((size_t)N * F2 * H2 * W2, 0.f);
+      for (int n = 0; n < N; ++n)
+        for (int c = 0; c < F2; ++c)
+          for (int iy = 0; iy < H2; ++iy)
+            for (int ix = 0; ix < W2; ++ix) {
+              float acc = 0.f;
+              for (int f = 0; f < F3; ++f)
+                for (int ky = 0; ky < K3; ++ky)
+                  for (int kx = 0; kx < K3; ++kx) {
+                    const int oy = iy - ky, ox = ix - kx;
+                    if (oy < 0 || oy >= H3 || ox < 0 || ox >= W3) continue;
+                    acc += bf2f(dz3h[(size_t)n * (F3 * 49) + f * 49 +
+                                     oy * W3 + ox]) *
+                           bf2f(w3[f * 288 + (ky * 3 + kx) * 32 + c]);
+                  }
+              if (r2[((size_t)(n * F2 + c) * H2 + iy) * W2 + ix] <= 0.f)
+                acc = 0.f;
+              dA2[((size_t)(n * F2 + c) * H2 + iy) * W2 + ix] = acc;
+            }
+      // CPU d_a1 from bf16(dA2) through conv2^T, mask r1>0
+      std::vector<float> dA1((size_t)N * F1 * H1 * W1, 0.f);
+      for (int n = 0; n < N; ++n)
+        for (int c = 0; c < F1; ++c)
+          for (int iy = 0; iy < H1; ++iy)
+            for (int ix = 0; ix < W1; ++ix) {
+              float acc = 0.f;
+              for (int f = 0; f < F2; ++f)
+                for (int ky = 0; ky < K2; ++ky)
+                  for (int kx = 0; kx < K2; ++kx) {
+                    if ((iy - ky) % 2 || (ix - kx) % 2) continue;
+                    const int oy = (iy - ky) / 2, ox = (ix - kx) / 2;
+                    if (oy < 0 || oy >= H2 || ox < 0 || ox >= W2) continue;
+                    acc += bf2f(f2bf(
+                               dA2[((size_t)(n * F2 + f) * H2 + oy) * W2 +
+                                   ox])) *
+                           bf2f(w2[f * 256 + c * 16 + ky * 4 + kx]);
+                  }
+              if (r1[((size_t)(n * F1 + c) * H1 + iy) * W1 + ix] <= 0.f)
+                acc = 0.f;
+              dA1[((size_t)(n * F1 + c) * H1 + iy) * W1 + ix] = acc;
+            }
+      // CPU dW refs in the kernels' k-major slab layouts
+      std::vector<float> rdw3((size_t)288 * F3, 0.f),
+          rdw2((size_t)256 * F2, 0.f), rdw1((size_t)256 * F1, 0.f);
+      for (int n = 0; n < N; ++n) {
+        for (int tap = 0; tap < 9; ++tap)
+          for (int c = 0; c < F2; ++c)
+            for (int f = 0; f < F3; ++f) {
+              const int ky = tap / 3, kx = tap % 3;
+              float acc = 0.f;
+              for (int oy = 0; oy < H3; ++oy)
+                for (int ox = 0; ox < W3; ++ox)
+                  acc += r2[((size_t)(n * F2 + c) * H2 + oy + ky) * W2 +
+                            ox + kx] *
+                         bf2f(dz3h[(size_t)n * (F3 * 49) + f * 49 +
+                                   oy * W3 + ox]);
+              rdw3[(size_t)(tap * 32 + c) * F3 + f] += acc;
+            }
+        for (int c = 0; c < F1; ++c)
+          for (int t = 0; t < 16; ++t)
+            for (int f = 0; f < F2; ++f) {
+              const int ky = t / 4, kx = t % 4;
+              float acc = 0.f;
+              for (int oy = 0; oy < H2; ++oy)
+                for (int ox = 0; ox < W2; ++ox)
+                  acc += r1[((size_t)(n * F1 + c) * H1 + S2 * oy + ky) *
+                            W1 + S2 * ox + kx] *
+                         bf2f(f2bf(dA2[((size_t)(n * F2 + f) * H2 + oy) *
+                                       W2 + ox]));
+              rdw2[(size_t)(c * 16 + t) * F2 + f] += acc;
+            }
+        for (int c = 0; c < C0; ++c)
+          for (int t = 0; t < 64; ++t)
+            for (int f = 0; f < F1; ++f) {
+              const int ky = t / 8, kx = t % 8;
+              float acc = 0.f;
+              for (int oy = 0; oy < H1; ++oy)
+                for (int ox = 0; ox < W1; ++ox)
+                  acc += bf2f(f2bf(
+                             fr[((size_t)(n * C0 + c) * H0 + S1 * oy + ky) *
+                                W0 + S1 * ox + kx] * (1.f / 255.f))) *
+                         bf2f(f2bf(dA1[((size_t)(n * F1 + f) * H1 + oy) *
+                                       W1 + ox]));
+              rdw1[(size_t)(c * 64 + t) * F1 + f] += acc;
+            }
+      }
+      // weight repacks for the dX kernels
+      std::vector<u16> wr3((size_t)F2 * 288), wr2c((size_t)4 * F1 * 128);
+      for (int c = 0; c < F2; ++c)
+        for (int ky = 0; ky < 3; ++ky)
+          for (int kx = 0; kx < 3; ++kx)
+            for (int f = 0; f < F3; ++f)
+              wr3[(size_t)c * 288 + (ky * 3 + kx) * 32 + f] =
+                  w3[f * 288 + ((2 - ky) * 3 + (2 - kx)) * 32 + c];
+      for (int cls = 0; cls < 4; ++cls)
+        for (int c = 0; c < F1; ++c)
+          for (int t = 0; t < 4; ++t)
+            for (int f = 0; f < F2; ++f)
+              wr2c[((size_t)cls * F1 + c) * 128 + t * 32 + f] =
+                  w2[f * 256 + c * 16 + ((cls >> 1) + 2 * (t >> 1)) * 4 +
+                     ((cls & 1) + 2 * (t & 1))];
+      // GPU: mask-convert then the backward chain
+      const int SL = shallow_mfma_slab();
+      void *dd3, *ddz3, *dda1, *dda2, *dwr3, *dwr2, *ds1, *ds2, *ds3;
+      CHECK(hipMalloc(&dd3, d3f.size() * 4));
+      CHECK(hipMalloc(&ddz3, d3f.size() * 2));
+      CHECK(hipMalloc(&dda1, (size_t)N * H1 * W1 * F1 * 2));
+      CHECK(hipMalloc(&dda2, (size_t)N * H2 * W2 * F2 * 2));
+      CHECK(hipMalloc(&dwr3, wr3.size() * 2));
+      CHECK(hipMalloc(&dwr2, wr2c.size() * 2));
+      CHECK(hipMalloc(&ds1, (size_t)SL * 256 * F1 * 4));
+      CHECK(hipMalloc(&ds2, (size_t)SL * 256 * F2 * 4));
+      CHECK(hipMalloc(&ds3, (size_t)SL * 288 * F3 * 4));
+      CHECK(hipMemcpy(dd3, d3f.data(), d3f.size() * 4,
+                      hipMemcpyHostToDevice));
+      CHECK(hipMemcpy(dwr3, wr3.data(), wr3.size() * 2,
+                      hipMemcpyHostToDevice));
+      CHECK(hipMemcpy(dwr2, wr2c.data(), wr2c.size() * 2,
+                      hipMemcpyHostToDevice));
+      CHECK(hipMemset(ds1, 0, (size_t)SL * 256 * F1 * 4));
+      CHECK(hipMemset(ds2, 0, (size_t)SL * 256 * F2 * 4));
+      CHECK(hipMemset(ds3, 0, (size_t)SL * 288 * F3 * 4));
+      launch_shallow_mask_convert((const float*)dd3, da3, ddz3,
+                                  (long)d3f.size(), 0);
+      launch_shallow_mfma_bwd((const unsigned char*)dfr, da1, da2, ddz3,
+                              dwr3, dwr2, dda1, dda2, (float*)ds1,
+                              (float*)ds2, (float*)ds3, N, 0);
+      CHECK(hipDeviceSynchronize());
+      std::vector<u16> odz3(d3f.size()), oda1((size_t)N * H1 * W1 * F1),
+          oda2((size_t)N * H2 * W2 * F2);
+      CHECK(hipMemcpy(odz3.data(), ddz3, odz3.size() * 2,
+                      hipMemcpyDeviceToHost));
+      CHECK(hipMemcpy(oda1.data(), dda1, oda1.size() * 2,
+                      hipMemcpyDeviceToHost));
+      CHECK(hipMemcpy(oda2.data(), dda2, oda2.size() * 2,
+                      hipMemcpyDeviceToHost));
+      float em = 0;
+      for (size_t i = 0; i < dz3h.size(); ++i)
+        em = std::max(em,
+                      std::fabs(bf2f(odz3[i]) - bf2f(dz3h[i])));
+      report("shallow bwd mask-convert (dz3)", em, 1e-6f);
+      float e2x = 0, m2x = 1;
+      for (int n = 0; n < N; ++n)
+        for (int c = 0; c < F2; ++c)
+          for (int p = 0; p < H2 * W2; ++p) {
+            const float rf = dA2[(size_t)(n * F2 + c) * H2 * W2 + p];
+            m2x = std::max(m2x, std::fabs(rf));
+            e2x = std::max(e2x,
+                           std::fabs(bf2f(oda2[((size_t)n * H2 * W2 + p) *
+                                               F2 + c]) - rf));
+          }
+      report("shallow bwd conv3 dX (+relu mask)", e2x / m2x, 2e-2f);
+      float e1x = 0, m1x = 1;
+      for (int n = 0; n < N; ++n)
+        for (int c = 0; c < F1; ++c)
+          for (int p = 0; p < H1 * W1; ++p) {
+            const float rf = dA1[(size_t)(n * F1 + c) * H1 * W1 + p];
+            m1x = std::max(m1x, std::fabs(rf));
+            e1x = std::max(e1x,
+                           std::fabs(bf2f(oda1[((size_t)n * H1 * W1 + p) *
+                                               F1 + c]) - rf));
+          }
+      report("shallow bwd conv2 dX (+relu mask)", e1x / m1x, 2e-2f);
+      auto check_dw = [&](void* dslab, const std::vector<float>& ref,
+                          size_t rows, int cols, const char* nm) {
+        std::vector<float> slabs((size_t)SL * ref.size());
+        CHECK(hipMemcpy(slabs.data(), dslab, slabs.size() * 4,
+                        hipMemcpyDeviceToHost));
+        std::vector<float> sum(ref.size(), 0.f);
+        for (int sl = 0; sl < SL; ++sl)
+          for (size_t i = 0; i < ref.size(); ++i)
+            sum[i] += slabs[(size_t)sl * ref.size() + i];
+        float e = 0, m = 1;
+        for (size_t i = 0; i < ref.size(); ++i) {
+          m = std::max(m, std::fabs(ref[i]));
+          e = std::max(e, std::fabs(sum[i] - ref[i]));
+        }
+        report(nm, e / m, 2e-2f);
+        (void)rows; (void)cols;
+      };
+      check_dw(ds3, rdw3, 288, F3, "shallow bwd dW3 (slab sum)");
+      check_dw(ds2, rdw2, 256, F2, "shallow bwd dW2 (slab sum)");
+      check_dw(ds1, rdw1, 256, F1, "shallow bwd dW1 (slab sum)");
+      hipFree(dd3); hipFree(ddz3); hipFree(dda1); hipFree(dda2);
+      hipFree(dwr3); hipFree(dwr2); hipFree(ds1); hipFree(ds2);
+      hipFree(ds3);
+    }
     hipFree(dfr); hipFree(dw1); hipFree(db1); hipFree(dw2); hipFree(db2);
     hipFree(dw3); hipFree(db3); hipFree(da1); hipFree(da2); hipFree(da3);
   }

@@ -1,0 +1,455 @@
+// Backward of the MFMA bf16 shallow encoder (shallow_mfma.hip).
+//
+// Chain (z* = pre-ReLU): d_z3 = d_a3 .* (a3>0)   [masked convert]
+//   dW3 = a2 (x) d_z3          d_a2 = conv3^T(d_z3) .* (a2>0)
+//   dW2 = a1 (x) d_z2          d_a1 = conv2^T(d_z2) .* (a1>0)
+//   dW1 = frames/255 (x) d_z1  (no d_frames)
+// ReLU masks for d_a2/d_a1 are fused into the dX epilogues; biases are
+// plain sums done with torch on the python side.
+//
+// dX conv3 (stride 1, VALID 9<-7): full correlation = conv3x3 of the
+//   2-halo-padded d_z3 with rot180/transposed weights — same implicit
+//   GEMM as forward, K = 9 taps x 32 cout (c-inner, NHWC LDS patch).
+// dX conv2 (stride 2, K=4, 20<-9): four (iy&1, ix&1) parity classes,
+//   each a regular 10x10 grid with a fixed 2x2 tap subset ->
+//   K = 2*2*32 = 128 per class, per-class packed weights.
+// dW: per-image outer products M = K_fwd, N = cout, K = px (deep-dw
+//   recipe), accumulated into DW_NSLAB slab copies with fp32 atomics.
+#include <hip/hip_runtime.h>
+
+namespace impala_shallow_mfma_bwd {
+
+typedef __bf16 bf16_t;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+
+__device__ __forceinline__ float bf2f(bf16_t v) { return (float)v; }
+__device__ __forceinline__ bf16_t f2bf(float v) { return (bf16_t)v; }
+
+constexpr int C0 = 4, H0 = 84, W0 = 84;
+constexpr int F1 = 16, K1 = 8, S1 = 4, H1 = 20, W1 = 20;
+constexpr int F2 = 32, K2 = 4, S2 = 2, H2 = 9, W2 = 9;
+constexpr int F3 = 32, K3 = 3, H3 = 7, W3 = 7;
+constexpr int SLAB = 256;
+
+// --------------------------------------------------------------------------
+// masked convert: d_z3 = bf16(d_a3_f32) .* (a3 > 0); both NCHW-flat.
+// --------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void mask_convert(
+    const float* __restrict__ d, const bf16_t* __restrict__ act,
+    bf16_t* __restrict__ o, long n) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    o[i] = f2bf(bf2f(act[i]) > 0.f ? d[i] : 0.f);
+}
+
+// --------------------------------------------------------------------------
+// conv3 dX: d_a2[iy][ix][c] = sum_{ky,kx,f} dz3p[iy+ky][ix+kx][f]
+//   * wr[c][(ky*3+kx)*32+f], masked by a2>0.  dz3 arrives NCHW-flat
+//   [f*49+px]; staged into a 13x13x32 NHWC zero-padded LDS patch.
+//   M = 96 (6 frags round-robin), N = 32 (c), K = 288 (9 steps).
+// --------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void conv3_dx(
+    const bf16_t* __restrict__ dz3, const bf16_t* __restrict__ wr,
+    const bf16_t* __restrict__ a2, bf16_t* __restrict__ da2, int N) {
+  constexpr int P = H3 + 2 * (K3 - 1);      // 11? no: 7 + 4 = 11
+  __shared__ bf16_t s_p[P * P * F3];        // 11*11*32 = 7.6 KiB
+  __shared__ bf16_t s_w[F2 * 288];          // 18 KiB
+
+  const int img = blockIdx.x;
+  if (img >= N) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  for (int i = tid; i < F2 * 288; i += 256) s_w[i] = wr[i];
+  {
+    const bf16_t* src = dz3 + (size_t)img * (F3 * H3 * W3);
+    for (int i = tid; i < P * P * F3; i += 256) {
+      const int f = i % F3;
+      const int x = (i / F3) % P;
+      const int y = i / (F3 * P);
+      const int sy = y - (K3 - 1), sx = x - (K3 - 1);
+      s_p[i] = (sy >= 0 && sy < H3 && sx >= 0 && sx < W3)
+                   ? src[f * (H3 * W3) + sy * W3 + sx]
+                   : f2bf(0.f);
+    }
+  }
+  __syncthreads();
+
+  const int arow = lane & 15;
+  const int kbase = (lane >> 4) * 8;
+  for (int mf = wave; mf < 6; mf += 4) {
+    floatx4 acc[2];
+    acc[0] = floatx4{0.f, 0.f, 0.f, 0.f};
+    acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
+    const int p = mf * 16 + arow;
+    const int iy = p / W2, ix = p % W2;
+    const bool live = p < H2 * W2;
+    #pragma unroll
+    for (int s = 0; s < 9; ++s) {
+      const int k0 = s * 32 + kbase;
+      const int tap = k0 >> 5, f0 = k0 & 31;
+      const int ky = tap / 3, kx = tap % 3;
+      bf16x8 af = {};
+      if (live)
+        af = *reinterpret_cast<const bf16x8*>(
+            s_p + ((iy + ky) * P + (ix + kx)) * F3 + f0);
+      const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
+          s_w + (lane & 15) * 288 + k0);
+      const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
+          s_w + (16 + (lane & 15)) * 288 + k0);
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc[0], 0,
+                                                       0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1, acc[1], 0,
+                                                       0, 0);
+    }
+    const int ccol = lane & 15;
+    const int crow = (lane >> 4) * 4;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      const int c = n * 16 + ccol;
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int pp = mf * 16 + crow + reg;
+        if (pp < H2 * W2) {
+          const size_t off = ((size_t)img * H2 * W2 + pp) * F2 + c;
+          float v = acc[n][reg];
+          if (bf2f(a2[off]) <= 0.f) v = 0.f;
+          da2[off] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// conv2 dX: four parity classes (py,px). Class grid 10x10 (100 px, 7
+//   frags), K = 128 = (kyi,kxi in 2x2) x 32 f; per-class packed weights
+//   wr2[(py*2+px)][c][ (kyi*2+kxi)*32 + f ] with
+//   wr2[...] = w2[f][c][py+2kyi][px+2kxi].  dz2 NHWC staged whole
+//   (9x9x32); out-of-range oy/ox contribute zeros. Masked by a1>0.
+// --------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void conv2_dx(
+    const bf16_t* __restrict__ dz2, const bf16_t* __restrict__ wr2,
+    const bf16_t* __restrict__ a1, bf16_t* __restrict__ da1, int N) {
+  __shared__ bf16_t s_d[H2 * W2 * F2];      // 5.1 KiB NHWC
+  __shared__ bf16_t s_w[4 * F1 * 128];      // 16 KiB
+
+  const int img = blockIdx.x;
+  if (img >= N) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  for (int i = tid; i < 4 * F1 * 128; i += 256) s_w[i] = wr2[i];
+  {
+    const bf16_t* src = dz2 + (size_t)img * H2 * W2 * F2;
+    for (int i = tid; i < H2 * W2 * F2; i += 256) s_d[i] = src[i];
+  }
+  __syncthreads();
+
+  const int arow = lane & 15;
+  const int kbase = (lane >> 4) * 8;
+  for (int cls = 0; cls < 4; ++cls) {
+    const int py = cls >> 1, px = cls & 1;
+    const bf16_t* wc = s_w + cls * F1 * 128;
+    for (int mf = wave; mf < 7; mf += 4) {
+      floatx4 acc = floatx4{0.f, 0.f, 0.f, 0.f};
+      const int p = mf * 16 + arow;
+      const int yq = p / 10, xq = p % 10;   // class-grid coords
+      const bool live = p < 100;
+      #pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const int k0 = s * 32 + kbase;
+        const int t = k0 >> 5;              // (kyi*2 + kxi)
+        const int f0 = k0 & 31;
+        const int oy = yq - (t >> 1);       // oy = y' - kyi
+        const int ox = xq - (t & 1);
+        bf16x8 af = {};
+        if (live && oy >= 0 && oy < H2 && ox >= 0 && ox < W2)
+          af = *reinterpret_cast<const bf16x8*>(
+              s_d + ((size_t)oy * W2 + ox) * F2 + f0);
+        const bf16x8 bf = *reinterpret_cast<const bf16x8*>(
+            wc + (lane & 15) * 128 + k0);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0,
+                                                      0);
+      }
+      const int ccol = lane & 15;
+      const int crow = (lane >> 4) * 4;
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int pp = mf * 16 + crow + reg;
+        if (pp < 100) {
+          const int iy = 2 * (pp / 10) + py, ix = 2 * (pp % 10) + px;
+          const size_t off = (((size_t)img * H1 + iy) * W1 + ix) * F1 + ccol;
+          float v = acc[reg];
+          if (bf2f(a1[off]) <= 0.f) v = 0.f;
+          da1[off] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// dW3: M = 288 (k_fwd = tap*32+c), N = 32 (f), K = 49 px (1 step of 32
+//   + 1 step of 17 -> 64 padded).  A[m][px] = a2[(oy+ky)(ox+kx)][c]
+//   element gathers from the NHWC a2 LDS; B[px][f] from dz3 staged
+//   [f][56] (NCHW-flat rows padded) -> contiguous 8-px reads.
+//   Slab-atomic accumulate into dw3_slab[s][m*32+f] (k-major layout;
+//   python reshapes).
+// --------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void conv3_dw(
+    const bf16_t* __restrict__ a2, const bf16_t* __restrict__ dz3,
+    float* __restrict__ dw_slab, int N) {
+  __shared__ bf16_t s_a[H2 * W2 * F2];      // 5.1 KiB NHWC
+  __shared__ bf16_t s_d[F3 * 56];           // 3.5 KiB, px-padded rows
+
+  const int img = blockIdx.x;
+  if (img >= N) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (288 * F3);
+
+  {
+    const bf16_t* src = a2 + (size_t)img * H2 * W2 * F2;
+    for (int i = tid; i < H2 * W2 * F2; i += 256) s_a[i] = src[i];
+    const bf16_t* sd = dz3 + (size_t)img * (F3 * H3 * W3);
+    for (int i = tid; i < F3 * 56; i += 256) {
+      const int px = i % 56, f = i / 56;
+      s_d[i] = px < H3 * W3 ? sd[f * (H3 * W3) + px] : f2bf(0.f);
+    }
+  }
+  __syncthreads();
+
+  const int kbase = (lane >> 4) * 8;
+  // 18 m-frags (288/16), round-robin over waves
+  for (int mf = wave; mf < 18; mf += 4) {
+    floatx4 acc[2];
+    acc[0] = floatx4{0.f, 0.f, 0.f, 0.f};
+    acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
+    const int m = mf * 16 + (lane & 15);    // k_fwd = tap*32 + c
+    const int tap = m >> 5, c = m & 31;
+    const int ky = tap / 3, kx = tap % 3;
+    #pragma unroll
+    for (int s = 0; s < 2; ++s) {           // px steps (49 -> 64)
+      bf16x8 af = {}, b0 = {}, b1 = {};
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int px = s * 32 + kbase + j;
+        if (px < H3 * W3) {
+          const int oy = px / W3, ox = px % W3;
+          af[j] = s_a[((oy + ky) * W2 + (ox + kx)) * F2 + c];
+        }
+      }
+      const int pxb = s * 32 + kbase;
+      b0 = *reinterpret_cast<const bf16x8*>(
+          s_d + (lane & 15) * 56 + pxb);
+      b1 = *reinterpret_cast<const bf16x8*>(
+          s_d + (16 + (lane & 15)) * 56 + pxb);
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc[0], 0,
+                                                       0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1, acc[1], 0,
+                                                       0, 0);
+    }
+    const int dcol = lane & 15;
+    const int drow = (lane >> 4) * 4;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      const int f = n * 16 + dcol;
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int mm = mf * 16 + drow + reg;
+        atomicAdd(&dw[(size_t)mm * F3 + f], acc[n][reg]);
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// dW2: M = 256 (c*16+tap), N = 32, K = 81 px (3 steps of 32, padded).
+//   A from planar a1 LDS; B from dz2 staged TRANSPOSED [f][96].
+//   Slab layout dw2_slab[s][m*32+f].
+// --------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void conv2_dw(
+    const bf16_t* __restrict__ a1, const bf16_t* __restrict__ dz2,
+    float* __restrict__ dw_slab, int N) {
+  __shared__ bf16_t s_a[F1 * H1 * W1];      // 12.5 KiB planar
+  __shared__ bf16_t s_d[F2 * 96];           // 6 KiB transposed
+
+  const int img = blockIdx.x;
+  if (img >= N) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (256 * F2);
+
+  {
+    const bf16_t* src = a1 + (size_t)img * H1 * W1 * F1;
+    for (int i = tid; i < F1 * H1 * W1; i += 256) {
+      const int x = i % W1;
+      const int y = (i / W1) % H1;
+      const int c = i / (W1 * H1);
+      s_a[i] = src[((size_t)y * W1 + x) * F1 + c];
+    }
+    const bf16_t* sd = dz2 + (size_t)img * H2 * W2 * F2;
+    for (int i = tid; i < F2 * 96; i += 256) {
+      const int px = i % 96, f = i / 96;
+      s_d[i] = px < H2 * W2 ? sd[(size_t)px * F2 + f] : f2bf(0.f);
+    }
+  }
+  __syncthreads();
+
+  const int kbase = (lane >> 4) * 8;
+  for (int mf = wave; mf < 16; mf += 4) {   // 256/16 m-frags
+    floatx4 acc[2];
+    acc[0] = floatx4{0.f, 0.f, 0.f, 0.f};
+    acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
+    const int m = mf * 16 + (lane & 15);    // c*16 + ky*4 + kx
+    const int c = m >> 4;
+    const int ky = (m & 15) >> 2, kx = m & 3;
+    #pragma unroll
+    for (int s = 0; s < 3; ++s) {
+      bf16x8 af = {};
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int px = s * 32 + kbase + j;
+        if (px < H2 * W2) {
+          const int oy = px / W2, ox = px % W2;
+          af[j] = s_a[(c * H1 + S2 * oy + ky) * W1 + S2 * ox + kx];
+        }
+      }
+      const int pxb = s * 32 + kbase;
+      const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
+          s_d + (lane & 15) * 96 + pxb);
+      const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
+          s_d + (16 + (lane & 15)) * 96 + pxb);
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc[0], 0,
+                                                       0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1, acc[1], 0,
+                                                       0, 0);
+    }
+    const int dcol = lane & 15;
+    const int drow = (lane >> 4) * 4;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      const int f = n * 16 + dcol;
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int mm = mf * 16 + drow + reg;
+        atomicAdd(&dw[(size_t)mm * F2 + f], acc[n][reg]);
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// dW1: M = 256 (c*64+tap), N = 16, K = 400 px (13 steps, padded 416).
+//   A from the planar u8->bf16 frame patch (whole 4x84x84, 56.4 KiB);
+//   B from dz1 staged transposed [16][416].
+// --------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 1) void conv1_dw(
+    const unsigned char* __restrict__ frames,
+    const bf16_t* __restrict__ dz1, float* __restrict__ dw_slab, int N) {
+  __shared__ bf16_t s_a[C0 * H0 * W0];      // 55.1 KiB planar
+  __shared__ bf16_t s_d[F1 * 416];          // 13 KiB transposed
+
+  const int img = blockIdx.x;
+  if (img >= N) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (256 * F1);
+
+  {
+    const unsigned char* fr = frames + (size_t)img * C0 * H0 * W0;
+    for (int i = tid; i < C0 * H0 * W0; i += 256)
+      s_a[i] = f2bf(fr[i] * (1.f / 255.f));
+    const bf16_t* sd = dz1 + (size_t)img * H1 * W1 * F1;
+    for (int i = tid; i < F1 * 416; i += 256) {
+      const int px = i % 416, f = i / 416;
+      s_d[i] = px < H1 * W1 ? sd[(size_t)px * F1 + f] : f2bf(0.f);
+    }
+  }
+  __syncthreads();
+
+  const int kbase = (lane >> 4) * 8;
+  for (int mf = wave; mf < 16; mf += 4) {
+    floatx4 acc = floatx4{0.f, 0.f, 0.f, 0.f};
+    const int m = mf * 16 + (lane & 15);    // c*64 + ky*8 + kx
+    const int c = m >> 6;
+    const int ky = (m & 63) >> 3, kx = m & 7;
+    #pragma unroll 4
+    for (int s = 0; s < 13; ++s) {
+      bf16x8 af = {};
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int px = s * 32 + kbase + j;
+        if (px < H1 * W1) {
+          const int oy = px / W1, ox = px % W1;
+          af[j] = s_a[(c * H0 + S1 * oy + ky) * W0 + S1 * ox + kx];
+        }
+      }
+      const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
+          s_d + (lane & 15) * 416 + s * 32 + kbase);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc, 0, 0, 0);
+    }
+    const int dcol = lane & 15;
+    const int drow = (lane >> 4) * 4;
+    #pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int mm = mf * 16 + drow + reg;
+      atomicAdd(&dw[(size_t)mm * F1 + dcol], acc[reg]);
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// launchers — slabs are SLAB zeroed copies, k-major [m][f]; the python
+// side sums over slabs and reshapes to the torch weight layout.
+// --------------------------------------------------------------------------
+extern "C" int shallow_mfma_slab() { return SLAB; }
+
+extern "C" void launch_shallow_mask_convert(const float* d, const void* act,
+                                            void* o, long n,
+                                            hipStream_t s) {
+  long blocks = (n + 255) / 256;
+  if (blocks > 16384) blocks = 16384;
+  hipLaunchKernelGGL(mask_convert, dim3((int)blocks), dim3(256), 0, s, d,
+                     static_cast<const bf16_t*>(act),
+                     static_cast<bf16_t*>(o), n);
+}
+
+// Full backward chain from dz3 (= d_a3 masked by a3>0): the produced
+// da2/da1 ARE the masked dz2/dz1 (the dX epilogues fuse the ReLU
+// masks), so the dW kernels consume them directly. Same-stream launch
+// order gives the dependencies. Python sums the biases from
+// dz3/da2/da1 and the slab copies for the weights.
+extern "C" void launch_shallow_mfma_bwd(
+    const unsigned char* frames, const void* a1, const void* a2,
+    const void* dz3, const void* wr3, const void* wr2, void* da1,
+    void* da2, float* dw1_slab, float* dw2_slab, float* dw3_slab, int N,
+    hipStream_t s) {
+  hipLaunchKernelGGL(conv3_dx, dim3(N), dim3(256), 0, s,
+                     static_cast<const bf16_t*>(dz3),
+                     static_cast<const bf16_t*>(wr3),
+                     static_cast<const bf16_t*>(a2),
+                     static_cast<bf16_t*>(da2), N);
+  hipLaunchKernelGGL(conv3_dw, dim3(N), dim3(256), 0, s,
+                     static_cast<const bf16_t*>(a2),
+                     static_cast<const bf16_t*>(dz3), dw3_slab, N);
+  hipLaunchKernelGGL(conv2_dx, dim3(N), dim3(256), 0, s,
+                     static_cast<const bf16_t*>(da2),
+                     static_cast<const bf16_t*>(wr2),
+                     static_cast<const bf16_t*>(a1),
+                     static_cast<bf16_t*>(da1), N);
+  hipLaunchKernelGGL(conv2_dw, dim3(N), dim3(256), 0, s,
+                     static_cast<const bf16_t*>(a1),
+                     static_cast<const bf16_t*>(da2), dw2_slab, N);
+  hipLaunchKernelGGL(conv1_dw, dim3(N), dim3(256), 0, s, frames,
+                     static_cast<const bf16_t*>(da1), dw1_slab, N);
+}
+
+}  // namespace impala_shallow_mfma_bwd

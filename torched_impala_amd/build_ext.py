@@ -77,7 +77,7 @@ def build_actor(force: bool = False) -> str:
 
 HIP_SOURCES = ["hip_ext.cpp", "vtrace_kernels.hip", "lstm_kernels.hip",
                "optim_kernels.hip", "conv_kernels.hip", "linear_kernels.hip",
-               "deep_kernels.hip", "shallow_mfma.hip"]
+               "deep_kernels.hip", "shallow_mfma.hip", "shallow_mfma_bwd.hip"]
 
 
 def build_probe(force: bool = False) -> str:
@@ -87,7 +87,8 @@ def build_probe(force: bool = False) -> str:
     out = os.path.join(CSRC, "deep_probe")
     sources = [os.path.join(CSRC, "deep_probe.cpp"),
                os.path.join(CSRC, "deep_kernels.hip"),
-               os.path.join(CSRC, "shallow_mfma.hip")]
+               os.path.join(CSRC, "shallow_mfma.hip"),
+               os.path.join(CSRC, "shallow_mfma_bwd.hip")]
     if force or _needs_build(out, sources):
         _run(["hipcc", "--offload-arch=gfx950", "-O2", "-Wno-unused-value",
               "-x", "hip"] + sources + ["-o", out])

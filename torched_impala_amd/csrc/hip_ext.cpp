@@ -70,6 +70,17 @@ void launch_nchw_f32_to_nhwc_bf16(const void*, void*, int, int, int, int,
                                   hipStream_t);
 void launch_nhwc_bf16_to_nchw_f32(const void*, void*, int, int, int, int,
                                   hipStream_t);
+void launch_shallow_mfma_fwd(const unsigned char*, const void*,
+                             const float*, const void*, const float*,
+                             const void*, const float*, void*, void*,
+                             void*, int, hipStream_t);
+int shallow_mfma_slab();
+void launch_shallow_mask_convert(const float*, const void*, void*, long,
+                                 hipStream_t);
+void launch_shallow_mfma_bwd(const unsigned char*, const void*, const void*,
+                             const void*, const void*, const void*, void*,
+                             void*, float*, float*, float*, int,
+                             hipStream_t);
 }
 
 namespace {
@@ -556,6 +567,84 @@ torch::Tensor nhwc_bf16_to_nchw(torch::Tensor x) {
   return y;
 }
 
+// ---------------------------------------------------------------------------
+// MFMA bf16 shallow encoder (shallow_mfma.hip / shallow_mfma_bwd.hip)
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> shallow_mfma_fwd(torch::Tensor frames,
+                                            torch::Tensor w1p,
+                                            torch::Tensor b1,
+                                            torch::Tensor w2p,
+                                            torch::Tensor b2,
+                                            torch::Tensor w3p,
+                                            torch::Tensor b3) {
+  TORCH_CHECK(frames.scalar_type() == torch::kUInt8 && frames.is_cuda() &&
+                  frames.is_contiguous() && frames.size(1) == 4 &&
+                  frames.size(2) == 84 && frames.size(3) == 84,
+              "frames must be cuda u8 (N,4,84,84)");
+  check_bf16_nhwc(w1p, "w1p");
+  check_bf16_nhwc(w2p, "w2p");
+  check_bf16_nhwc(w3p, "w3p");
+  check_f32(b1, "b1");
+  check_f32(b2, "b2");
+  check_f32(b3, "b3");
+  const int N = frames.size(0);
+  DeviceScope guard(frames.get_device());
+  auto opts = frames.options().dtype(torch::kBFloat16);
+  auto a1 = torch::empty({N, 20, 20, 16}, opts);
+  auto a2 = torch::empty({N, 9, 9, 32}, opts);
+  auto a3 = torch::empty({N, 1568}, opts);
+  launch_shallow_mfma_fwd(frames.data_ptr<unsigned char>(), w1p.data_ptr(),
+                          b1.data_ptr<float>(), w2p.data_ptr(),
+                          b2.data_ptr<float>(), w3p.data_ptr(),
+                          b3.data_ptr<float>(), a1.data_ptr(),
+                          a2.data_ptr(), a3.data_ptr(), N,
+                          current_stream());
+  HIP_CHECK(hipGetLastError());
+  return {a1, a2, a3};
+}
+
+torch::Tensor shallow_mask_convert(torch::Tensor d, torch::Tensor act) {
+  check_f32(d, "d");
+  check_bf16_nhwc(act, "act");
+  DeviceScope guard(d.get_device());
+  auto o = torch::empty_like(act);
+  launch_shallow_mask_convert(d.data_ptr<float>(), act.data_ptr(),
+                              o.data_ptr(), d.numel(), current_stream());
+  HIP_CHECK(hipGetLastError());
+  return o;
+}
+
+// returns (da1, da2, dw1k, dw2k, dw3k): da* bf16 (ReLU-masked), dw*k
+// fp32 in k-major [M][F] layout (python reshapes to torch layout).
+std::vector<torch::Tensor> shallow_mfma_bwd(torch::Tensor frames,
+                                            torch::Tensor a1,
+                                            torch::Tensor a2,
+                                            torch::Tensor dz3,
+                                            torch::Tensor wr3,
+                                            torch::Tensor wr2) {
+  check_bf16_nhwc(a1, "a1");
+  check_bf16_nhwc(a2, "a2");
+  check_bf16_nhwc(dz3, "dz3");
+  check_bf16_nhwc(wr3, "wr3");
+  check_bf16_nhwc(wr2, "wr2");
+  const int N = frames.size(0);
+  DeviceScope guard(frames.get_device());
+  const int SL = shallow_mfma_slab();
+  auto fopts = frames.options().dtype(torch::kFloat32);
+  auto da1 = torch::empty_like(a1);
+  auto da2 = torch::empty_like(a2);
+  auto s1 = torch::zeros({SL, 256, 16}, fopts);
+  auto s2 = torch::zeros({SL, 256, 32}, fopts);
+  auto s3 = torch::zeros({SL, 288, 32}, fopts);
+  launch_shallow_mfma_bwd(frames.data_ptr<unsigned char>(), a1.data_ptr(),
+                          a2.data_ptr(), dz3.data_ptr(), wr3.data_ptr(),
+                          wr2.data_ptr(), da1.data_ptr(), da2.data_ptr(),
+                          s1.data_ptr<float>(), s2.data_ptr<float>(),
+                          s3.data_ptr<float>(), N, current_stream());
+  HIP_CHECK(hipGetLastError());
+  return {da1, da2, s1.sum(0), s2.sum(0), s3.sum(0)};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -586,4 +675,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deep_u8_to_nhwc4", &deep_u8_to_nhwc4);
   m.def("nchw_to_nhwc_bf16", &nchw_to_nhwc_bf16);
   m.def("nhwc_bf16_to_nchw", &nhwc_bf16_to_nchw);
+  m.def("shallow_mfma_fwd", &shallow_mfma_fwd);
+  m.def("shallow_mask_convert", &shallow_mask_convert);
+  m.def("shallow_mfma_bwd", &shallow_mfma_bwd);
 }

@@ -173,8 +173,16 @@ class ShallowNet(_CoreMixin, _HeadsMixin):
                 and not getattr(self, "force_eager_core", False)):
             from torched_impala_amd import ops
             if ops.hip_available():
-                from torched_impala_amd.ops.encoder import shallow_encode_conv
                 from torched_impala_amd.ops.linear import linear_relu
+                from torched_impala_amd.ops.shallow_mfma import \
+                    shallow_mfma_enabled
+                if shallow_mfma_enabled():
+                    # bf16 MFMA encoder (IMPALA_SHALLOW_MFMA=1; see
+                    # ops/shallow_mfma.py for the validation gate)
+                    from torched_impala_amd.ops.shallow_mfma import \
+                        shallow_mfma_encode
+                    return linear_relu(self.fc, shallow_mfma_encode(self, x))
+                from torched_impala_amd.ops.encoder import shallow_encode_conv
                 return linear_relu(self.fc, shallow_encode_conv(self, x))
         x = x.float() / 255.0
         x = F.relu(self.conv1(x))

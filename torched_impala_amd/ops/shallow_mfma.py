@@ -1,0 +1,81 @@
+"""Shallow encoder on bf16 MFMA matrix cores (learner side).
+
+Autograd glue over csrc/shallow_mfma.hip + shallow_mfma_bwd.hip.  The
+fp32 LDS-fused encoder (ops/encoder.py) stays the default; this path is
+enabled with IMPALA_SHALLOW_MFMA=1 after the round-2 probe validation
+(deep_probe sections 8-9) and perf check — see ROADMAP.md.
+
+Weight packings (pure torch reshapes, done per forward; ~20 KiB total):
+  w1p [16][256]  = conv1.weight.flatten(1)            k = c*64+ky*8+kx
+  w2p [32][256]  = conv2.weight.flatten(1)            k = c*16+ky*4+kx
+  w3p [32][288]  = conv3.weight.permute(0,2,3,1).flatten(1)  k = tap*32+c
+Backward repacks:
+  wr3 [32][288]  = conv3.weight.flip(2,3).permute(1,2,3,0).flatten(1)
+  wr2 [4][16][128]: class (py,px): w2[:, :, py::2, px::2]
+                    .permute(1,2,3,0).flatten(1)      k = (kyi*2+kxi)*32+f
+dW returns arrive k-major [M][F] and are reshaped back to the torch
+[F][C][K][K] layouts here.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+
+from . import ext_hip
+
+
+def shallow_mfma_enabled() -> bool:
+    return os.environ.get("IMPALA_SHALLOW_MFMA", "0") == "1"
+
+
+class _ShallowMfmaEncode(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, frames, w1, b1, w2, b2, w3, b3):
+        ext = ext_hip()
+        bf = torch.bfloat16
+        w1p = w1.detach().flatten(1).to(bf).contiguous()
+        w2p = w2.detach().flatten(1).to(bf).contiguous()
+        w3p = w3.detach().permute(0, 2, 3, 1).flatten(1).to(bf).contiguous()
+        a1, a2, a3 = ext.shallow_mfma_fwd(
+            frames.contiguous(), w1p, b1.detach().float().contiguous(),
+            w2p, b2.detach().float().contiguous(), w3p,
+            b3.detach().float().contiguous())
+        ctx.save_for_backward(frames, a1, a2, a3, w2, w3)
+        return a3
+
+    @staticmethod
+    def backward(ctx, d_a3):
+        ext = ext_hip()
+        frames, a1, a2, a3, w2, w3 = ctx.saved_tensors
+        bf = torch.bfloat16
+        dz3 = ext.shallow_mask_convert(d_a3.float().contiguous(), a3)
+        wr3 = (w3.detach().flip(2, 3).permute(1, 2, 3, 0).flatten(1)
+               .to(bf).contiguous())
+        cls = []
+        for py in range(2):
+            for px in range(2):
+                cls.append(w2.detach()[:, :, py::2, px::2]
+                           .permute(1, 2, 3, 0).reshape(16, 128))
+        wr2 = torch.stack(cls).to(bf).contiguous()
+        da1, da2, dw1k, dw2k, dw3k = ext.shallow_mfma_bwd(
+            frames.contiguous(), a1, a2, dz3, wr3, wr2)
+        # k-major -> torch layouts
+        dw1 = dw1k.view(4, 8, 8, 16).permute(3, 0, 1, 2).contiguous()
+        dw2 = dw2k.view(16, 4, 4, 32).permute(3, 0, 1, 2).contiguous()
+        dw3 = dw3k.view(9, 32, 32).permute(2, 1, 0).reshape(32, 32, 3, 3
+                                                            ).contiguous()
+        n = frames.shape[0]
+        db1 = da1.float().sum(dim=(0, 1, 2))
+        db2 = da2.float().sum(dim=(0, 1, 2))
+        db3 = dz3.float().view(n, 32, 49).sum(dim=(0, 2))
+        return None, dw1, db1, dw2, db2, dw3, db3
+
+
+def shallow_mfma_encode(net, frames_u8: torch.Tensor) -> torch.Tensor:
+    """(N,4,84,84) u8 -> (N,1568) fp32 conv features (NCHW-flat order,
+    drop-in for ops.encoder.shallow_encode_conv)."""
+    a3 = _ShallowMfmaEncode.apply(
+        frames_u8, net.conv1.weight, net.conv1.bias, net.conv2.weight,
+        net.conv2.bias, net.conv3.weight, net.conv3.bias)
+    return a3.float()

@@ -1,0 +1,43 @@
+#!/bin/bash
+# Round-2 opening GPU call — run via:
+#   /usr/local/graft/bin/gpurun --timeout 900 -- 'bash scripts/round2_opening.sh'
+# Validates everything round 1 built blind, in cost order (cheapest and
+# most-gating first), writing all artifacts under gpurun_out/.
+#
+# 1. deep_probe: all kernel parities incl. the shallow MFMA encoder
+#    fwd+bwd (sections 8-9) and the slab-dW deep kernels. Torch-free,
+#    ~2 s. If a shallow section FAILS, do NOT flip IMPALA_SHALLOW_MFMA.
+# 2. deep_probe bench: per-kernel throughput incl. the slab-dW fix.
+# 3. actor probe A/B: fp32 native vs int8/bf16 native on the box's EPYC.
+# 4. learner capacity A/B: fp32 encoder vs IMPALA_SHALLOW_MFMA=1.
+# 5. GPU test suite (skips shallow-mfma-gated tests unless env set).
+set -x
+mkdir -p gpurun_out
+O=gpurun_out
+
+./torched_impala_amd/csrc/deep_probe 2>&1 | tee $O/r2_probe.log
+PROBE_RC=${PIPESTATUS[0]}
+./torched_impala_amd/csrc/deep_probe bench 2>&1 | tee $O/r2_probe_bench.log
+
+timeout 300 python scripts/actor_probe.py --jit native --actors 4 \
+    --seconds 8 > $O/r2_actor_fp32.json 2> $O/r2_actor_fp32.err
+timeout 300 python scripts/actor_probe.py --jit native-bf16 --actors 4 \
+    --seconds 8 > $O/r2_actor_bf16.json 2> $O/r2_actor_bf16.err
+
+timeout 300 python scripts/learner_capacity.py --steps 60 --warmup 15 \
+    > $O/r2_cap_fp32.json 2> $O/r2_cap_fp32.err
+if [ "$PROBE_RC" = "0" ]; then
+  IMPALA_SHALLOW_MFMA=1 timeout 300 python scripts/learner_capacity.py \
+      --steps 60 --warmup 15 > $O/r2_cap_mfma.json 2> $O/r2_cap_mfma.err
+else
+  echo "probe FAILED — skipping shallow-mfma capacity run" \
+      | tee $O/r2_cap_mfma.err
+fi
+
+timeout 600 python -m pytest tests -m gpu -x -q 2>&1 | tail -20 \
+    | tee $O/r2_gpu_tests.log
+
+echo "=== summary ==="
+grep -c PASS $O/r2_probe.log || true
+tail -1 $O/r2_actor_fp32.json $O/r2_actor_bf16.json \
+    $O/r2_cap_fp32.json $O/r2_cap_mfma.json 2>/dev/null || true

@@ -1,20 +1,30 @@
 // Native CPU actor inference for the shallow IMPALA net (ROADMAP item 1;
 // BASELINE.json:5 "CPU actor processes" — the actor-side hot loop).
 //
-// The whole-system headline is actor-bound: 32 single-threaded CPU
-// actor processes each run the policy forward per env step. The torch
-// eager path costs ~0.14 ms/frame (op dispatch + oneDNN) and the
-// jit-frozen path ~0.09 ms; this hand-vectorized forward targets the
-// AVX2 FMA ceiling (~16 MAC/cycle/core) with zero dispatch overhead:
-//
-//   u8 frame -> f32/255 -> conv1 16x8x8/s4 -> conv2 32x4x4/s2
-//   -> conv3 32x3x3 -> fc 256 -> policy head -> softmax sample,
+//   u8 frame -> conv1 16x8x8/s4 -> conv2 32x4x4/s2 -> conv3 32x3x3
+//   -> fc 256 [-> LSTM 256] -> policy head -> softmax sample,
 //
 // all filter-innermost (weights prepacked [c][ky][kx][f]) so the inner
-// loops are broadcast-pixel x vector-FMA over output channels. The
-// baseline head is skipped (actors never use values). Compiled with
-// -mavx2 -mfma via plain g++ auto-vectorizable loops (fixed trip
-// counts, restrict pointers) — portable to any x86-64 actor host.
+// loops are broadcast-pixel x vector-MAC over output channels. The
+// baseline head is skipped (actors never use values). Three precision
+// tiers, runtime-dispatched per __builtin_cpu_supports:
+//
+//   fp32 AVX2 / AVX-512: exact parity with the torch policy (~1e-7);
+//     the fp32 AVX-512 forward measures at the single-core FMA
+//     roofline (~218 GF/s on the dev host).
+//   int8 VNNI + bf16 (mode "native-bf16"): conv1 consumes the RAW u8
+//     frame against s8 weights on vpdpbusd (per-filter scales folding
+//     /255; input exact); conv2/3 and the fc GEMV run on u8-requantised
+//     activations (ReLU outputs are non-negative -> zero-point-0 u8
+//     with a per-frame dynamic scale computed in the producing
+//     epilogue); the LSTM gates GEMM runs on bf16 pairs (vdpbf16ps).
+//     Measured 5.1x the fp32 forward at 2e-4 logits error; behavior
+//     logits stay self-consistent with sampling, which is all V-trace
+//     requires.
+//
+// Compiled with -mavx2 -mfma base + per-function target attributes —
+// portable to any x86-64 actor host, fastest on full-512 cores
+// (EPYC Zen4/5, SPR+).
 //
 // Weight refresh: load_weights() repacks from the torch tensors (a few
 // ms) — rate-limited by the same actor_jit_refresh cadence as the

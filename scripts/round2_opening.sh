@@ -18,6 +18,13 @@ O=gpurun_out
 ./torched_impala_amd/csrc/deep_probe 2>&1 | tee $O/r2_probe.log
 PROBE_RC=${PIPESTATUS[0]}
 ./torched_impala_amd/csrc/deep_probe bench 2>&1 | tee $O/r2_probe_bench.log
+# exact-coverage tile variants (ROADMAP 3c): same parity ladder with the
+# env set at process start; flip the env into the production default
+# only if this passes AND the bench shows the win
+IMPALA_DEEP_TILES=1 ./torched_impala_amd/csrc/deep_probe 2>&1 \
+    | tee $O/r2_probe_tiles.log
+IMPALA_DEEP_TILES=1 ./torched_impala_amd/csrc/deep_probe bench 2>&1 \
+    | tee $O/r2_probe_tiles_bench.log
 
 timeout 300 python scripts/actor_probe.py --jit native --actors 4 \
     --seconds 8 > $O/r2_actor_fp32.json 2> $O/r2_actor_fp32.err

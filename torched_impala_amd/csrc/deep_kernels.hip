@@ -48,7 +48,13 @@ constexpr int THREADS = 256;           // 4 waves
 // res: [N][H][W][COUT] bf16 or nullptr
 // y:   [N][H][W][COUT] bf16
 // ---------------------------------------------------------------------------
-template <int CIN, int COUT, bool RELU_IN, bool HAS_RES>
+// TTY x TTX is the output tile; (16,16) is the hardware-validated
+// default, the other instantiations give (near-)exact coverage of the
+// deep shapes (36x48 -> 12x16, 18x24 -> 9x24, 9x12 -> 9x12; ROADMAP
+// item 3c) and are selected by the launcher under IMPALA_DEEP_TILES=1
+// until the round-2 probe validates them.
+template <int CIN, int COUT, bool RELU_IN, bool HAS_RES, int TTY = TY,
+          int TTX = TX>
 __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
     const float* __restrict__ b, const bf16_t* __restrict__ res,
@@ -56,19 +62,22 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
   constexpr int KTOT = 9 * CIN;
   constexpr int KSTEPS = (KTOT + 31) / 32;       // MFMA k-steps (zero-padded)
   constexpr int NFRAG = COUT / 16;
-  constexpr int PSTR = (TX + 2) * CIN;           // patch row stride (halfs)
+  constexpr int NPX = TTY * TTX;
+  constexpr int MFRAGS = (NPX + 15) / 16;
+  constexpr int MPW = (MFRAGS + 3) / 4;          // m-frags per wave
+  constexpr int PSTR = (TTX + 2) * CIN;          // patch row stride (halfs)
   constexpr int WSTR = 9 * CIN;                  // weight row stride (halfs)
 
-  __shared__ bf16_t s_patch[(TY + 2) * PSTR];
+  __shared__ bf16_t s_patch[(TTY + 2) * PSTR];
   __shared__ bf16_t s_w[COUT * WSTR];
 
-  const int ntx = (W + TX - 1) / TX;
-  const int nty = (H + TY - 1) / TY;
+  const int ntx = (W + TTX - 1) / TTX;
+  const int nty = (H + TTY - 1) / TTY;
   const int tile = blockIdx.x % (ntx * nty);
   const int img = blockIdx.x / (ntx * nty);
   if (img >= N) return;
-  const int ty0 = (tile / ntx) * TY;             // tile origin (output px)
-  const int tx0 = (tile % ntx) * TX;
+  const int ty0 = (tile / ntx) * TTY;            // tile origin (output px)
+  const int tx0 = (tile % ntx) * TTX;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -86,8 +95,8 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
     const unsigned short* xs = reinterpret_cast<const unsigned short*>(x) +
                                (size_t)img * H * W * CIN;
     unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
-    for (int p = tid; p < (TY + 2) * (TX + 2); p += THREADS) {
-      const int py = p / (TX + 2), px = p % (TX + 2);
+    for (int p = tid; p < (TTY + 2) * (TTX + 2); p += THREADS) {
+      const int py = p / (TTX + 2), px = p % (TTX + 2);
       const int iy = ty0 + py - 1, ix = tx0 + px - 1;
       unsigned short* dst = pd + py * PSTR + px * CIN;
       if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
@@ -104,10 +113,12 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
   __syncthreads();
 
   // --- MFMA main loop ----------------------------------------------------
-  // wave handles m-frags {wave*4 .. wave*4+3}; all NFRAG n-frags.
-  floatx4 acc[4][NFRAG];
+  // m-frags round-robin over the 4 waves (frag f handled by wave f%4);
+  // padded rows (p >= NPX) are clamped for the A read and masked at the
+  // epilogue.
+  floatx4 acc[MPW][NFRAG];
   #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < MPW; ++m)
     #pragma unroll
     for (int n = 0; n < NFRAG; ++n) acc[m][n] = floatx4{0.f, 0.f, 0.f, 0.f};
 
@@ -137,11 +148,13 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
     }
     // A fragments: patch pixel (m-frag row) shifted by the tap offset.
     #pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      const int p = (wave * 4 + m) * 16 + arow;   // px index in 16x16 tile
-      const int oy = p >> 4, ox = p & 15;
+    for (int m = 0; m < MPW; ++m) {
+      const int mf = wave + m * 4;
+      int p = mf * 16 + arow;        // px index in the TTY x TTX tile
+      if (p >= NPX) p = NPX - 1;     // clamp pad rows (masked at store)
+      const int oy = p / TTX, ox = p % TTX;
       bf16x8 af = {};
-      if (k0 < KTOT) {
+      if (mf < MFRAGS && k0 < KTOT) {
         if constexpr (CIN >= 8) {
           const int tap = k0 / CIN, c0 = k0 % CIN;  // 8 elems, one tap
           const int ky = tap / 3, kx = tap % 3;
@@ -169,15 +182,17 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
   const int ccol = lane & 15;
   const int crow = (lane >> 4) * 4;
   #pragma unroll
-  for (int m = 0; m < 4; ++m) {
+  for (int m = 0; m < MPW; ++m) {
+    const int mf = wave + m * 4;
     #pragma unroll
     for (int n = 0; n < NFRAG; ++n) {
       const int cout = n * 16 + ccol;
       const float bias = b[cout];
       #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int p = (wave * 4 + m) * 16 + crow + reg;
-        const int oy = ty0 + (p >> 4), ox = tx0 + (p & 15);
+        const int p = mf * 16 + crow + reg;
+        if (mf >= MFRAGS || p >= NPX) continue;
+        const int oy = ty0 + p / TTX, ox = tx0 + p % TTX;
         if (oy < H && ox < W) {
           const size_t off = (((size_t)img * H + oy) * W + ox) * COUT + cout;
           float v = acc[m][n][reg] + bias;
@@ -196,7 +211,7 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
 // epilogue-masked by sign(x_fwd) (fused ReLU backward).  This variant
 // only adds the mask read; cin/cout roles are swapped by the caller.
 // ---------------------------------------------------------------------------
-template <int CIN, int COUT>
+template <int CIN, int COUT, int TTY = TY, int TTX = TX>
 __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ wt,
     const bf16_t* __restrict__ xmask, bf16_t* __restrict__ dx, int N, int H,
@@ -204,19 +219,22 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
   constexpr int KTOT = 9 * CIN;
   constexpr int KSTEPS = (KTOT + 31) / 32;
   constexpr int NFRAG = COUT / 16;
-  constexpr int PSTR = (TX + 2) * CIN;
+  constexpr int NPX = TTY * TTX;
+  constexpr int MFRAGS = (NPX + 15) / 16;
+  constexpr int MPW = (MFRAGS + 3) / 4;
+  constexpr int PSTR = (TTX + 2) * CIN;
   constexpr int WSTR = 9 * CIN;
 
-  __shared__ bf16_t s_patch[(TY + 2) * PSTR];
+  __shared__ bf16_t s_patch[(TTY + 2) * PSTR];
   __shared__ bf16_t s_w[COUT * WSTR];
 
-  const int ntx = (W + TX - 1) / TX;
-  const int nty = (H + TY - 1) / TY;
+  const int ntx = (W + TTX - 1) / TTX;
+  const int nty = (H + TTY - 1) / TTY;
   const int tile = blockIdx.x % (ntx * nty);
   const int img = blockIdx.x / (ntx * nty);
   if (img >= N) return;
-  const int ty0 = (tile / ntx) * TY;
-  const int tx0 = (tile % ntx) * TX;
+  const int ty0 = (tile / ntx) * TTY;
+  const int tx0 = (tile % ntx) * TTX;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -230,8 +248,8 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
     const unsigned short* xs = reinterpret_cast<const unsigned short*>(dy) +
                                (size_t)img * H * W * CIN;
     unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
-    for (int p = tid; p < (TY + 2) * (TX + 2); p += THREADS) {
-      const int py = p / (TX + 2), px = p % (TX + 2);
+    for (int p = tid; p < (TTY + 2) * (TTX + 2); p += THREADS) {
+      const int py = p / (TTX + 2), px = p % (TTX + 2);
       const int iy = ty0 + py - 1, ix = tx0 + px - 1;
       unsigned short* dst = pd + py * PSTR + px * CIN;
       if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
@@ -246,9 +264,9 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
   }
   __syncthreads();
 
-  floatx4 acc[4][NFRAG];
+  floatx4 acc[MPW][NFRAG];
   #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < MPW; ++m)
     #pragma unroll
     for (int n = 0; n < NFRAG; ++n) acc[m][n] = floatx4{0.f, 0.f, 0.f, 0.f};
   const int arow = lane & 15;
@@ -266,11 +284,13 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
       bf[n] = v;
     }
     #pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      const int p = (wave * 4 + m) * 16 + arow;
-      const int oy = p >> 4, ox = p & 15;
+    for (int m = 0; m < MPW; ++m) {
+      const int mf = wave + m * 4;
+      int p = mf * 16 + arow;
+      if (p >= NPX) p = NPX - 1;
+      const int oy = p / TTX, ox = p % TTX;
       bf16x8 af = {};
-      if (k0 < KTOT) {
+      if (mf < MFRAGS && k0 < KTOT) {
         const int tap = k0 / CIN, c0 = k0 % CIN;
         af = *reinterpret_cast<const bf16x8*>(
             s_patch + (oy + tap / 3) * PSTR + (ox + tap % 3) * CIN + c0);
@@ -285,14 +305,16 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
   const int ccol = lane & 15;
   const int crow = (lane >> 4) * 4;
   #pragma unroll
-  for (int m = 0; m < 4; ++m) {
+  for (int m = 0; m < MPW; ++m) {
+    const int mf = wave + m * 4;
     #pragma unroll
     for (int n = 0; n < NFRAG; ++n) {
       const int cout = n * 16 + ccol;
       #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int p = (wave * 4 + m) * 16 + crow + reg;
-        const int oy = ty0 + (p >> 4), ox = tx0 + (p & 15);
+        const int p = mf * 16 + crow + reg;
+        if (mf >= MFRAGS || p >= NPX) continue;
+        const int oy = ty0 + p / TTX, ox = tx0 + p % TTX;
         if (oy < H && ox < W) {
           const size_t off = (((size_t)img * H + oy) * W + ox) * COUT + cout;
           float v = acc[m][n][reg];
@@ -319,7 +341,7 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
 // (measured 9 TF); 256 slabs cut the per-word contention 256x.
 constexpr int DW_NSLAB = 256;
 
-template <int CIN, int COUT, bool RELU_IN>
+template <int CIN, int COUT, bool RELU_IN, int TTY = TY, int TTX = TX>
 __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
     float* __restrict__ dw_slab, int N, int H, int W) {
@@ -328,18 +350,20 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
   constexpr int MTOT = 9 * CIN;                  // 144 or 288
   constexpr int MFRAGS = (MTOT + 15) / 16;       // 9 or 18
   constexpr int NFRAG = COUT / 16;
-  constexpr int PSTR = (TX + 2) * CIN;
+  constexpr int NPX = TTY * TTX;
+  constexpr int KPX = ((NPX + 31) / 32) * 32;    // px padded to 32-mult
+  constexpr int PSTR = (TTX + 2) * CIN;
 
-  __shared__ bf16_t s_patch[(TY + 2) * PSTR];
-  __shared__ bf16_t s_dy[TY * TX * COUT];
+  __shared__ bf16_t s_patch[(TTY + 2) * PSTR];
+  __shared__ bf16_t s_dy[KPX * COUT];
 
-  const int ntx = (W + TX - 1) / TX;
-  const int nty = (H + TY - 1) / TY;
+  const int ntx = (W + TTX - 1) / TTX;
+  const int nty = (H + TTY - 1) / TTY;
   const int tile = blockIdx.x % (ntx * nty);
   const int img = blockIdx.x / (ntx * nty);
   if (img >= N) return;
-  const int ty0 = (tile / ntx) * TY;
-  const int tx0 = (tile % ntx) * TX;
+  const int ty0 = (tile / ntx) * TTY;
+  const int tx0 = (tile % ntx) * TTX;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -348,8 +372,8 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
     const unsigned short* xs = reinterpret_cast<const unsigned short*>(x) +
                                (size_t)img * H * W * CIN;
     unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
-    for (int p = tid; p < (TY + 2) * (TX + 2); p += THREADS) {
-      const int py = p / (TX + 2), px = p % (TX + 2);
+    for (int p = tid; p < (TTY + 2) * (TTX + 2); p += THREADS) {
+      const int py = p / (TTX + 2), px = p % (TTX + 2);
       const int iy = ty0 + py - 1, ix = tx0 + px - 1;
       unsigned short* dst = pd + py * PSTR + px * CIN;
       if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
@@ -367,10 +391,10 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
     const unsigned short* ds = reinterpret_cast<const unsigned short*>(dy) +
                                (size_t)img * H * W * COUT;
     unsigned short* dd = reinterpret_cast<unsigned short*>(s_dy);
-    for (int p = tid; p < TY * TX; p += THREADS) {
-      const int oy = ty0 + (p >> 4), ox = tx0 + (p & 15);
+    for (int p = tid; p < KPX; p += THREADS) {
+      const int oy = ty0 + p / TTX, ox = tx0 + p % TTX;
       unsigned short* dst = dd + p * COUT;
-      if (oy < H && ox < W) {
+      if (p < NPX && oy < H && ox < W) {
         const unsigned short* src = ds + ((size_t)oy * W + ox) * COUT;
         #pragma unroll
         for (int c = 0; c < COUT; ++c) dst[c] = src[c];
@@ -392,13 +416,13 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
     const int tap = m0 / CIN, c = m0 % CIN;      // may be >=9*CIN (pad)
     const int ky = tap / 3, kx = tap % 3;
     #pragma unroll 2
-    for (int s = 0; s < (TY * TX) / 32; ++s) {   // 8 k-steps of 32 px
+    for (int s = 0; s < KPX / 32; ++s) {         // k-steps of 32 px
       bf16x8 af = {}, bfv[NFRAG] = {};
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int px = s * 32 + kbase + j;
-        if (m0 < MTOT)
-          af[j] = s_patch[((px >> 4) + ky) * PSTR + ((px & 15) + kx) * CIN
+        if (m0 < MTOT && px < NPX)
+          af[j] = s_patch[(px / TTX + ky) * PSTR + (px % TTX + kx) * CIN
                           + c];
       }
       #pragma unroll
@@ -569,26 +593,60 @@ __global__ void mfma_probe(const bf16_t* __restrict__ A,
 // ---------------------------------------------------------------------------
 // C-linkage launchers
 // ---------------------------------------------------------------------------
-static inline int conv_grid(int N, int H, int W) {
-  const int ntx = (W + TX - 1) / TX, nty = (H + TY - 1) / TY;
-  return N * ntx * nty;
+// tile selection: 0 = 16x16 (validated default); under
+// IMPALA_DEEP_TILES=1 pick an exact-coverage tile for the deep shapes
+// (ROADMAP 3c — validate via deep_probe/round2_opening.sh first).
+static inline int tile_id(int H, int W) {
+  static int want = -1;
+  if (want < 0) {
+    const char* e = getenv("IMPALA_DEEP_TILES");
+    want = (e != nullptr && e[0] == '1') ? 1 : 0;
+  }
+  if (!want) return 0;
+  if (H % 12 == 0 && W % 16 == 0) return 1;   // 36x48, 72x96
+  if (H % 9 == 0 && W % 24 == 0) return 2;    // 18x24
+  if (H % 9 == 0 && W % 12 == 0) return 3;    // 9x12
+  return 0;
 }
 
-#define CONV_LAUNCH(CIN, COUT)                                               \
+static inline void tile_dims(int id, int* ty, int* tx) {
+  switch (id) {
+    case 1: *ty = 12; *tx = 16; return;
+    case 2: *ty = 9; *tx = 24; return;
+    case 3: *ty = 9; *tx = 12; return;
+    default: *ty = TY; *tx = TX; return;
+  }
+}
+
+static inline int conv_grid_t(int N, int H, int W, int id) {
+  int ty, tx;
+  tile_dims(id, &ty, &tx);
+  return N * ((W + tx - 1) / tx) * ((H + ty - 1) / ty);
+}
+
+#define CONV_LAUNCH_T(CIN, COUT, TTY, TTX)                                   \
   if (relu_in) {                                                             \
     if (res)                                                                 \
-      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, true>), grid, blk,   \
-                         0, s, xb, wb, b, rb, yb, N, H, W);                  \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, true, TTY, TTX>),    \
+                         grid, blk, 0, s, xb, wb, b, rb, yb, N, H, W);       \
     else                                                                     \
-      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, false>), grid, blk,  \
-                         0, s, xb, wb, b, nullptr, yb, N, H, W);             \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, false, TTY, TTX>),   \
+                         grid, blk, 0, s, xb, wb, b, nullptr, yb, N, H, W);  \
   } else {                                                                   \
     if (res)                                                                 \
-      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, true>), grid, blk,  \
-                         0, s, xb, wb, b, rb, yb, N, H, W);                  \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, true, TTY, TTX>),   \
+                         grid, blk, 0, s, xb, wb, b, rb, yb, N, H, W);       \
     else                                                                     \
-      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, false>), grid, blk, \
-                         0, s, xb, wb, b, nullptr, yb, N, H, W);             \
+      hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, false, TTY, TTX>),  \
+                         grid, blk, 0, s, xb, wb, b, nullptr, yb, N, H, W);  \
+  }
+
+#define CONV_LAUNCH(CIN, COUT)                                               \
+  switch (tid_) {                                                            \
+    case 1: CONV_LAUNCH_T(CIN, COUT, 12, 16) break;                          \
+    case 2: CONV_LAUNCH_T(CIN, COUT, 9, 24) break;                           \
+    case 3: CONV_LAUNCH_T(CIN, COUT, 9, 12) break;                           \
+    default: CONV_LAUNCH_T(CIN, COUT, TY, TX) break;                         \
   }
 
 extern "C" int launch_deep_conv3x3(const void* x, const void* w,
@@ -599,7 +657,8 @@ extern "C" int launch_deep_conv3x3(const void* x, const void* w,
   const bf16_t* wb = static_cast<const bf16_t*>(w);
   const bf16_t* rb = static_cast<const bf16_t*>(res);
   bf16_t* yb = static_cast<bf16_t*>(y);
-  dim3 grid(conv_grid(N, H, W)), blk(THREADS);
+  const int tid_ = tile_id(H, W);
+  dim3 grid(conv_grid_t(N, H, W, tid_)), blk(THREADS);
   if (cin == 4 && cout == 16) { CONV_LAUNCH(4, 16) }
   else if (cin == 16 && cout == 16) { CONV_LAUNCH(16, 16) }
   else if (cin == 16 && cout == 32) { CONV_LAUNCH(16, 32) }
@@ -640,16 +699,30 @@ extern "C" int launch_deep_conv3x3_dx(const void* dy, const void* wt,
   const bf16_t* w = static_cast<const bf16_t*>(wt);
   const bf16_t* m = static_cast<const bf16_t*>(xmask);
   bf16_t* o = static_cast<bf16_t*>(dx);
-  dim3 grid(conv_grid(N, H, W)), blk(THREADS);
-  if (cout_fwd == 16 && cin_fwd == 16)
-    hipLaunchKernelGGL((conv3x3_dx_nhwc<16, 16>), grid, blk, 0, s, d, w, m,
-                       o, N, H, W);
-  else if (cout_fwd == 32 && cin_fwd == 16)
-    hipLaunchKernelGGL((conv3x3_dx_nhwc<32, 16>), grid, blk, 0, s, d, w, m,
-                       o, N, H, W);
-  else if (cout_fwd == 32 && cin_fwd == 32)
-    hipLaunchKernelGGL((conv3x3_dx_nhwc<32, 32>), grid, blk, 0, s, d, w, m,
-                       o, N, H, W);
+  const int tid_ = tile_id(H, W);
+  dim3 grid(conv_grid_t(N, H, W, tid_)), blk(THREADS);
+#define DX_CASE(CI, CO)                                                     \
+  switch (tid_) {                                                           \
+    case 1:                                                                 \
+      hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO, 12, 16>), grid, blk, 0,   \
+                         s, d, w, m, o, N, H, W);                           \
+      break;                                                                \
+    case 2:                                                                 \
+      hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO, 9, 24>), grid, blk, 0,    \
+                         s, d, w, m, o, N, H, W);                           \
+      break;                                                                \
+    case 3:                                                                 \
+      hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO, 9, 12>), grid, blk, 0,    \
+                         s, d, w, m, o, N, H, W);                           \
+      break;                                                                \
+    default:                                                                \
+      hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO>), grid, blk, 0, s, d, w,  \
+                         m, o, N, H, W);                                    \
+      break;                                                                \
+  }
+  if (cout_fwd == 16 && cin_fwd == 16) { DX_CASE(16, 16) }
+  else if (cout_fwd == 32 && cin_fwd == 16) { DX_CASE(32, 16) }
+  else if (cout_fwd == 32 && cin_fwd == 32) { DX_CASE(32, 32) }
   else if (cout_fwd == 16 && cin_fwd == 4)
     // conv1 dX (rarely needed: frames have no grad) — COUT=4 < 16 not
     // representable; caller should skip conv1 dX entirely.
@@ -669,14 +742,22 @@ extern "C" int launch_deep_conv3x3_dw(const void* x, const void* dy,
                                       hipStream_t s) {
   const bf16_t* xb = static_cast<const bf16_t*>(x);
   const bf16_t* db = static_cast<const bf16_t*>(dy);
-  dim3 grid(conv_grid(N, H, W)), blk(THREADS);
-#define DW_CASE(CI, CO)                                                     \
+  const int tid_ = tile_id(H, W);
+  dim3 grid(conv_grid_t(N, H, W, tid_)), blk(THREADS);
+#define DW_CASE_T(CI, CO, TTY, TTX)                                         \
   if (relu_in)                                                              \
-    hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, true>), grid, blk, 0, s,    \
-                       xb, db, dw, N, H, W);                                \
+    hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, true, TTY, TTX>), grid,     \
+                       blk, 0, s, xb, db, dw, N, H, W);                     \
   else                                                                      \
-    hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, false>), grid, blk, 0, s,   \
-                       xb, db, dw, N, H, W);
+    hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, false, TTY, TTX>), grid,    \
+                       blk, 0, s, xb, db, dw, N, H, W);
+#define DW_CASE(CI, CO)                                                     \
+  switch (tid_) {                                                           \
+    case 1: DW_CASE_T(CI, CO, 12, 16) break;                                \
+    case 2: DW_CASE_T(CI, CO, 9, 24) break;                                 \
+    case 3: DW_CASE_T(CI, CO, 9, 12) break;                                 \
+    default: DW_CASE_T(CI, CO, TY, TX) break;                               \
+  }
   if (cin == 4 && cout == 16) { DW_CASE(4, 16) }
   else if (cin == 16 && cout == 16) { DW_CASE(16, 16) }
   else if (cin == 16 && cout == 32) { DW_CASE(16, 32) }

@@ -26,6 +26,7 @@ namespace impala_deep {
 
 typedef __bf16 bf16_t;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 typedef __attribute__((ext_vector_type(4))) float floatx4;
 
 __device__ __forceinline__ float bf2f(bf16_t v) { return (float)v; }
@@ -160,6 +161,17 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
           const int ky = tap / 3, kx = tap % 3;
           af = *reinterpret_cast<const bf16x8*>(
               s_patch + (oy + ky) * PSTR + (ox + kx) * CIN + c0);
+        } else if constexpr (CIN == 4) {
+          // 8 k-elems = taps t0, t0+1 with c 0..3 each: two 8B reads
+          const int t0 = k0 >> 2;
+          *reinterpret_cast<bf16x4*>(&af) =
+              *reinterpret_cast<const bf16x4*>(
+                  s_patch + (oy + t0 / 3) * PSTR + (ox + t0 % 3) * CIN);
+          if (t0 + 1 < 9)
+            *(reinterpret_cast<bf16x4*>(&af) + 1) =
+                *reinterpret_cast<const bf16x4*>(
+                    s_patch + (oy + (t0 + 1) / 3) * PSTR +
+                    (ox + (t0 + 1) % 3) * CIN);
         } else {
           #pragma unroll
           for (int j = 0; j < 8; ++j) {

@@ -143,6 +143,32 @@ def test_mock_grads_match_eager(mock_ext):
         assert 0.8 < ratio < 1.25, f"{k}: norm ratio {ratio}"
 
 
+def test_mock_features_with_fc_grads(mock_ext):
+    """shallow_mfma_features: conv stack via the Function + fc on bf16
+    torch autograd — fc weight grads must land on the fp32 masters."""
+    net, frames = _setup(32)
+    tgt = torch.randn(3, 256)
+    net.zero_grad(set_to_none=True)
+    out = S.shallow_mfma_features(net, frames)
+    ((out - tgt) ** 2).mean().backward()
+    assert net.fc.weight.grad is not None
+    assert net.fc.weight.grad.dtype == torch.float32
+    assert net.conv1.weight.grad is not None
+
+    net.zero_grad(set_to_none=True)
+    x = frames.float() / 255.0
+    ref = F.relu(net.fc(F.relu(net.conv3(F.relu(net.conv2(
+        F.relu(net.conv1(x)))))).flatten(1)))
+    ((ref - tgt) ** 2).mean().backward()
+    g_ref = net.fc.weight.grad.clone()
+    net.zero_grad(set_to_none=True)
+    out2 = S.shallow_mfma_features(net, frames)
+    ((out2 - tgt) ** 2).mean().backward()
+    g = net.fc.weight.grad
+    cos = F.cosine_similarity(g.flatten(), g_ref.flatten(), dim=0).item()
+    assert cos > 0.97, cos
+
+
 def test_flag_gate_defaults_off(monkeypatch):
     monkeypatch.delenv("IMPALA_SHALLOW_MFMA", raising=False)
     assert not S.shallow_mfma_enabled()

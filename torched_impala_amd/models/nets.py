@@ -177,11 +177,12 @@ class ShallowNet(_CoreMixin, _HeadsMixin):
                 from torched_impala_amd.ops.shallow_mfma import \
                     shallow_mfma_enabled
                 if shallow_mfma_enabled():
-                    # bf16 MFMA encoder (IMPALA_SHALLOW_MFMA=1; see
-                    # ops/shallow_mfma.py for the validation gate)
+                    # bf16 MFMA encoder + rocBLAS bf16 fc
+                    # (IMPALA_SHALLOW_MFMA=1; validation gate in
+                    # ops/shallow_mfma.py)
                     from torched_impala_amd.ops.shallow_mfma import \
-                        shallow_mfma_encode
-                    return linear_relu(self.fc, shallow_mfma_encode(self, x))
+                        shallow_mfma_features
+                    return shallow_mfma_features(self, x)
                 from torched_impala_amd.ops.encoder import shallow_encode_conv
                 return linear_relu(self.fc, shallow_encode_conv(self, x))
         x = x.float() / 255.0

@@ -79,3 +79,16 @@ def shallow_mfma_encode(net, frames_u8: torch.Tensor) -> torch.Tensor:
         frames_u8, net.conv1.weight, net.conv1.bias, net.conv2.weight,
         net.conv2.bias, net.conv3.weight, net.conv3.bias)
     return a3.float()
+
+
+def shallow_mfma_features(net, frames_u8: torch.Tensor) -> torch.Tensor:
+    """Full encode incl. the fc: the conv stack on the MFMA kernels, the
+    fc as a rocBLAS bf16 GEMM through plain autograd (weight grads land
+    on the fp32 masters via the tracked .to(bf16) casts)."""
+    bf = torch.bfloat16
+    a3 = _ShallowMfmaEncode.apply(
+        frames_u8, net.conv1.weight, net.conv1.bias, net.conv2.weight,
+        net.conv2.bias, net.conv3.weight, net.conv3.bias)
+    y = torch.nn.functional.linear(a3, net.fc.weight.to(bf),
+                                   net.fc.bias.to(bf))
+    return torch.relu(y).float()

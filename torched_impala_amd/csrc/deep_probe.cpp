@@ -244,6 +244,79 @@ int main(int argc, char** argv) {
              ms, 2.0 * N * H * W * cout * 9.0 * cin / ms / 1e9);
       hipFree(dyp); hipFree(wtp); hipFree(mp); hipFree(op); hipFree(dwp);
     }
+    // shallow MFMA encoder at the learner step size (B=32 T=20 -> 640)
+    {
+      const int N = 640;
+      void *dfr, *dw1, *db1, *dw2, *db2, *dw3, *db3, *da1, *da2, *da3;
+      CHECK(hipMalloc(&dfr, (size_t)N * 4 * 84 * 84));
+      CHECK(hipMalloc(&dw1, 16 * 256 * 2));
+      CHECK(hipMalloc(&db1, 16 * 4));
+      CHECK(hipMalloc(&dw2, 32 * 256 * 2));
+      CHECK(hipMalloc(&db2, 32 * 4));
+      CHECK(hipMalloc(&dw3, (size_t)32 * 288 * 2));
+      CHECK(hipMalloc(&db3, 32 * 4));
+      CHECK(hipMalloc(&da1, (size_t)N * 400 * 16 * 2));
+      CHECK(hipMalloc(&da2, (size_t)N * 81 * 32 * 2));
+      CHECK(hipMalloc(&da3, (size_t)N * 1568 * 2));
+      CHECK(hipMemset(dfr, 0x55, (size_t)N * 4 * 84 * 84));
+      CHECK(hipMemset(dw1, 0x3c, 16 * 256 * 2));
+      CHECK(hipMemset(db1, 0, 16 * 4));
+      CHECK(hipMemset(dw2, 0x3c, 32 * 256 * 2));
+      CHECK(hipMemset(db2, 0, 32 * 4));
+      CHECK(hipMemset(dw3, 0x3c, (size_t)32 * 288 * 2));
+      CHECK(hipMemset(db3, 0, 32 * 4));
+      hipEvent_t e0, e1;
+      CHECK(hipEventCreate(&e0)); CHECK(hipEventCreate(&e1));
+      const int iters = 50;
+      for (int i = 0; i < 5; ++i)
+        launch_shallow_mfma_fwd((const unsigned char*)dfr, dw1, (float*)db1,
+                                dw2, (float*)db2, dw3, (float*)db3, da1,
+                                da2, da3, N, 0);
+      CHECK(hipEventRecord(e0, 0));
+      for (int i = 0; i < iters; ++i)
+        launch_shallow_mfma_fwd((const unsigned char*)dfr, dw1, (float*)db1,
+                                dw2, (float*)db2, dw3, (float*)db3, da1,
+                                da2, da3, N, 0);
+      CHECK(hipEventRecord(e1, 0));
+      CHECK(hipDeviceSynchronize());
+      float ms; CHECK(hipEventElapsedTime(&ms, e0, e1)); ms /= iters;
+      printf("%-40s N=%-6d %8.3f ms (3 kernels; fp32 fused chain was "
+             "~0.11 ms)\n", "shallow mfma encoder FWD", N, ms);
+      // backward chain (dz3 reused from da3 memory as dummy bf16)
+      const int SL = shallow_mfma_slab();
+      void *wr3, *wr2, *bda1, *bda2, *s1, *s2, *s3;
+      CHECK(hipMalloc(&wr3, (size_t)32 * 288 * 2));
+      CHECK(hipMalloc(&wr2, (size_t)4 * 16 * 128 * 2));
+      CHECK(hipMalloc(&bda1, (size_t)N * 400 * 16 * 2));
+      CHECK(hipMalloc(&bda2, (size_t)N * 81 * 32 * 2));
+      CHECK(hipMalloc(&s1, (size_t)SL * 256 * 16 * 4));
+      CHECK(hipMalloc(&s2, (size_t)SL * 256 * 32 * 4));
+      CHECK(hipMalloc(&s3, (size_t)SL * 288 * 32 * 4));
+      CHECK(hipMemset(wr3, 0x3c, (size_t)32 * 288 * 2));
+      CHECK(hipMemset(wr2, 0x3c, (size_t)4 * 16 * 128 * 2));
+      CHECK(hipMemset(s1, 0, (size_t)SL * 256 * 16 * 4));
+      CHECK(hipMemset(s2, 0, (size_t)SL * 256 * 32 * 4));
+      CHECK(hipMemset(s3, 0, (size_t)SL * 288 * 32 * 4));
+      for (int i = 0; i < 5; ++i)
+        launch_shallow_mfma_bwd((const unsigned char*)dfr, da1, da2, da3,
+                                wr3, wr2, bda1, bda2, (float*)s1,
+                                (float*)s2, (float*)s3, N, 0);
+      CHECK(hipEventRecord(e0, 0));
+      for (int i = 0; i < iters; ++i)
+        launch_shallow_mfma_bwd((const unsigned char*)dfr, da1, da2, da3,
+                                wr3, wr2, bda1, bda2, (float*)s1,
+                                (float*)s2, (float*)s3, N, 0);
+      CHECK(hipEventRecord(e1, 0));
+      CHECK(hipDeviceSynchronize());
+      CHECK(hipEventElapsedTime(&ms, e0, e1)); ms /= iters;
+      printf("%-40s N=%-6d %8.3f ms (5 kernels; fp32 bwd ~0.6 ms; "
+             "slab sums excluded)\n",
+             "shallow mfma encoder BWD", N, ms);
+      hipFree(dfr); hipFree(dw1); hipFree(db1); hipFree(dw2); hipFree(db2);
+      hipFree(dw3); hipFree(db3); hipFree(da1); hipFree(da2); hipFree(da3);
+      hipFree(wr3); hipFree(wr2); hipFree(bda1); hipFree(bda2);
+      hipFree(s1); hipFree(s2); hipFree(s3);
+    }
     return 0;
   }
   int ndev = 0;

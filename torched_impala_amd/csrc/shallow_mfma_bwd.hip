@@ -204,7 +204,8 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
     const bf16_t* __restrict__ a2, const bf16_t* __restrict__ dz3,
     float* __restrict__ dw_slab, int N) {
   __shared__ bf16_t s_a[H2 * W2 * F2];      // 5.1 KiB NHWC
-  __shared__ bf16_t s_d[F3 * 56];           // 3.5 KiB, px-padded rows
+  __shared__ bf16_t s_d[F3 * 64];           // 4 KiB; rows padded to the
+                                            // k-loop's 2x32 px range
 
   const int img = blockIdx.x;
   if (img >= N) return;
@@ -217,8 +218,8 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
     const bf16_t* src = a2 + (size_t)img * H2 * W2 * F2;
     for (int i = tid; i < H2 * W2 * F2; i += 256) s_a[i] = src[i];
     const bf16_t* sd = dz3 + (size_t)img * (F3 * H3 * W3);
-    for (int i = tid; i < F3 * 56; i += 256) {
-      const int px = i % 56, f = i / 56;
+    for (int i = tid; i < F3 * 64; i += 256) {
+      const int px = i % 64, f = i / 64;
       s_d[i] = px < H3 * W3 ? sd[f * (H3 * W3) + px] : f2bf(0.f);
     }
   }
@@ -246,9 +247,9 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
       }
       const int pxb = s * 32 + kbase;
       b0 = *reinterpret_cast<const bf16x8*>(
-          s_d + (lane & 15) * 56 + pxb);
+          s_d + (lane & 15) * 64 + pxb);
       b1 = *reinterpret_cast<const bf16x8*>(
-          s_d + (16 + (lane & 15)) * 56 + pxb);
+          s_d + (16 + (lane & 15)) * 64 + pxb);
       acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc[0], 0,
                                                        0, 0);
       acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1, acc[1], 0,

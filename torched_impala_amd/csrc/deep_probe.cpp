@@ -896,10 +896,15 @@ int main(int argc, char** argv) {
         em = std::max(em,
                       std::fabs(bf2f(odz3[i]) - bf2f(dz3h[i])));
       report("shallow bwd mask-convert (dz3)", em, 1e-6f);
+      // skip ReLU-mask-ambiguous elements (activation within rounding
+      // noise of 0: CPU/GPU fp32 accumulation order can flip the mask
+      // bit there, a legitimate 1-ulp divergence, not a kernel bug)
       float e2x = 0, m2x = 1;
       for (int n = 0; n < N; ++n)
         for (int c = 0; c < F2; ++c)
           for (int p = 0; p < H2 * W2; ++p) {
+            if (std::fabs(r2[(size_t)(n * F2 + c) * H2 * W2 + p]) < 1e-5f)
+              continue;
             const float rf = dA2[(size_t)(n * F2 + c) * H2 * W2 + p];
             m2x = std::max(m2x, std::fabs(rf));
             e2x = std::max(e2x,
@@ -911,6 +916,8 @@ int main(int argc, char** argv) {
       for (int n = 0; n < N; ++n)
         for (int c = 0; c < F1; ++c)
           for (int p = 0; p < H1 * W1; ++p) {
+            if (std::fabs(r1[(size_t)(n * F1 + c) * H1 * W1 + p]) < 1e-5f)
+              continue;
             const float rf = dA1[(size_t)(n * F1 + c) * H1 * W1 + p];
             m1x = std::max(m1x, std::fabs(rf));
             e1x = std::max(e1x,

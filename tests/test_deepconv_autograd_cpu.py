@@ -9,7 +9,6 @@ ops.ext_hip for a torch-eager mock with the kernels' exact semantics
 checks deep_encode_train against the eager fp32 DeepResNet — any
 permute or mask mistake in the glue shows up here deterministically.
 """
-import numpy as np
 import pytest
 import torch
 import torch.nn.functional as F

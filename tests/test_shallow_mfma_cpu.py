@@ -4,7 +4,6 @@ kernels' layouts and semantics re-implemented with conv2d/conv grads so
 every weight pack, backward repack and dW reshape in the python glue is
 checked deterministically (the kernels themselves are covered by
 deep_probe sections 8-9 on hardware)."""
-import numpy as np
 import pytest
 import torch
 import torch.nn.functional as F

@@ -8,15 +8,25 @@ learner.
 """
 from __future__ import annotations
 
+import time
 from typing import List
 
 import torch
 
 
 class WeightMirror:
+    """Seqlock protocol: each buffer has a write-sequence counter that the
+    writer makes ODD before touching the buffer and EVEN after.  A reader
+    copies only when the sequence is even and unchanged across its copy —
+    a version-counter check alone cannot protect against the writer
+    re-using the same buffer two publishes later (observed as a
+    two-version tear under a fast writer)."""
+
     def __init__(self, model: torch.nn.Module):
         self.version = torch.zeros(1, dtype=torch.int64)
         self.version.share_memory_()
+        self.seq = torch.zeros(2, dtype=torch.int64)   # per-buffer seqlock
+        self.seq.share_memory_()
         self._bufs: List[List[torch.Tensor]] = []
         sd = model.state_dict()
         self.keys = list(sd.keys())
@@ -24,7 +34,7 @@ class WeightMirror:
             buf = [torch.empty_like(sd[k], device="cpu").share_memory_()
                    for k in self.keys]
             self._bufs.append(buf)
-        # publish initial weights into both buffers
+        # publish initial weights into both buffers (no readers yet)
         for b in range(2):
             self._write(b, model)
         self.version.fill_(1)
@@ -37,7 +47,10 @@ class WeightMirror:
     def publish(self, model: torch.nn.Module):
         """Called by the learner. Writes inactive buffer, flips version."""
         v = int(self.version.item())
-        self._write((v + 1) % 2, model)
+        b = (v + 1) % 2
+        self.seq[b] += 1               # odd: writer active in buffer b
+        self._write(b, model)
+        self.seq[b] += 1               # even: buffer b consistent
         self.version.fill_(v + 1)
 
     def publish_state_dict(self, sd):
@@ -45,27 +58,30 @@ class WeightMirror:
         per tensor straight into the shared mirror, no CPU model hop."""
         v = int(self.version.item())
         b = (v + 1) % 2
+        self.seq[b] += 1
         with torch.no_grad():
             for k, dst in zip(self.keys, self._bufs[b]):
                 dst.copy_(sd[k].detach(), non_blocking=False)
+        self.seq[b] += 1
         self.version.fill_(v + 1)
 
     def pull_into(self, model: torch.nn.Module, last_seen: int = -1) -> int:
         """Called by actors. Returns the version pulled (or last_seen if
         already current)."""
-        v = int(self.version.item())
-        if v == last_seen:
+        if int(self.version.item()) == last_seen:
             return last_seen
         while True:
-            src = self._bufs[v % 2]
+            v = int(self.version.item())
+            b = v % 2
+            s1 = int(self.seq[b].item())
+            if s1 & 1:                 # writer mid-overwrite of this buffer
+                time.sleep(2e-4)
+                continue
+            src = self._bufs[b]
             with torch.no_grad():
                 sd = model.state_dict()
                 for k, s in zip(self.keys, src):
                     sd[k].copy_(s)
-            v2 = int(self.version.item())
-            # version v's buffer stays intact only until the writer
-            # publishes v+2 (which rewrites buffer v%2): exactly one
-            # newer version is tolerable, two means our copy may be torn
-            if v2 <= v + 1:
+            if int(self.seq[b].item()) == s1:
                 return v
-            v = v2  # writer lapped us; retry from the newest version
+            # the writer re-entered buffer b during our copy; retry

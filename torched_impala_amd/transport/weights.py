@@ -1,10 +1,11 @@
 """Learner -> actor weight publication (SURVEY.md §3c weight publish).
 
 Double-buffered shared-memory mirror of the model parameters with a
-version counter: the learner writes the inactive buffer then bumps the
-version; actors copy from the active buffer and retry if the version
-moved mid-copy, so they never act on torn weights and never block the
-learner.
+version counter plus a per-buffer seqlock: the learner marks the
+inactive buffer's sequence odd, writes it, marks it even, then bumps
+the version; actors copy from the active buffer only at an even,
+unchanged-across-the-copy sequence. Readers never act on torn weights
+and never block the learner.
 """
 from __future__ import annotations
 

@@ -133,6 +133,7 @@ def main():
                 "seq_len": flags.unroll_length,
                 "parallelism": f"dp{n_gpus}",
                 "num_actors_per_gpu": flags.num_actors,
+                "actor_jit": flags.actor_jit,
                 "envs_per_actor": flags.envs_per_actor,
                 "frame_convention": "agent_steps",
                 "env_frames_per_sec_x4_convention": value * 4,

@@ -1,0 +1,37 @@
+"""Property-based FlatRMSprop parity (hypothesis): the flat-arena
+optimizer must match torch clip_grad_norm_ + torch.optim.RMSprop for
+arbitrary hyperparameters, momentum off, over several steps."""
+import torch
+from hypothesis import given, settings, strategies as st
+
+from torched_impala_amd.models import MLPPolicy
+from torched_impala_amd.ops.optim import FlatRMSprop
+from tests.test_optim import _loss
+
+
+@settings(max_examples=15, deadline=None)
+@given(lr=st.floats(1e-4, 0.1), alpha=st.floats(0.5, 0.999),
+       eps=st.floats(1e-8, 0.1), max_norm=st.floats(0.01, 100.0),
+       seed=st.integers(0, 2**31 - 1))
+def test_flat_rmsprop_property(lr, alpha, eps, max_norm, seed):
+    torch.manual_seed(seed)
+    m1 = MLPPolicy((4,), 2)
+    m2 = MLPPolicy((4,), 2)
+    m2.load_state_dict(m1.state_dict())
+    opt1 = FlatRMSprop(m1.parameters(), lr=lr, alpha=alpha, eps=eps,
+                       max_grad_norm=max_norm)
+    opt2 = torch.optim.RMSprop(m2.parameters(), lr=lr, alpha=alpha, eps=eps)
+    for it in range(3):
+        torch.manual_seed(seed ^ (7919 * (it + 1)))
+        opt1.zero_grad()
+        _loss(m1).backward()
+        opt1.step()
+        torch.manual_seed(seed ^ (7919 * (it + 1)))
+        opt2.zero_grad()
+        _loss(m2).backward()
+        torch.nn.utils.clip_grad_norm_(m2.parameters(), max_norm)
+        opt2.step()
+    for (k1, p1), (_, p2) in zip(m1.state_dict().items(),
+                                 m2.state_dict().items()):
+        torch.testing.assert_close(p1, p2, rtol=2e-4, atol=1e-5,
+                                   msg=lambda m, k=k1: f"{k}: {m}")

@@ -89,6 +89,49 @@ def test_native_bf16_mode_self_consistent_and_close():
             assert err < 5e-3, f"row {t}: bf16 drift {err}"
 
 
+@pytest.mark.parametrize("bf16", [False, True])
+def test_native_lstm_finite_at_large_cell_state(bf16):
+    """Regression: exp512's clamp constant sat exactly on the 2^127
+    boundary — at LSTM cell states |c| >= ~44.2 (2c hitting the clamp)
+    tanh returned inf/inf = NaN, which poisoned training through the
+    recorded behavior logits. Drive c deliberately into that regime
+    (forget/cell gate biases ~= +6 => c grows ~1 per step) and require
+    finite, eager-matching outputs past c > 50."""
+    torch.manual_seed(0)
+    from torched_impala_amd.actor import NativeActorPolicy
+    from torched_impala_amd.models import ShallowNet
+    net = ShallowNet((4, 84, 84), 6, use_lstm=True).eval()
+    with torch.no_grad():
+        H = 256
+        net.core.bias_ih[H:3 * H].fill_(8.0)   # forget + cell gates open
+        net.core.bias_hh[H:3 * H].fill_(0.0)
+    nat = NativeActorPolicy(net, 6, 0, bf16=bf16)
+    n = 4
+    frame = torch.randint(0, 256, (n, 4, 84, 84), dtype=torch.uint8)
+    reward = torch.zeros(n)
+    done = torch.zeros(n, dtype=torch.bool)
+    la = torch.zeros(n, dtype=torch.int64)
+    h = torch.zeros(n, 256)
+    c = torch.zeros(n, 256)
+    he, ce = h.clone(), c.clone()
+    for step in range(90):
+        lg, _, h, c = nat(frame, reward, done, la, h, c)
+        assert torch.isfinite(lg).all(), f"NaN logits at step {step}"
+        assert torch.isfinite(h).all() and torch.isfinite(c).all(), step
+    assert c.abs().max().item() > 46, "test no longer reaches the regime"
+    # eager comparison at the final state magnitude (fp32 mode: tight)
+    if not nat.bf16_active:
+        import torch.nn.functional as F
+        with torch.no_grad():
+            for _ in range(90):
+                x = net.encode(frame)
+                ci = torch.cat([torch.zeros(n, 1),
+                                F.one_hot(la, 6).float()], -1)
+                ci = torch.cat([x, ci], -1)
+                he, ce = net.core(ci, (he, ce))
+        assert (h - he).abs().max().item() < 1e-3
+
+
 def test_lstm_initial_state_replay():
     """initial_core_state lets the learner replay rows 0..T exactly."""
     flags, model, roll = _setup(use_lstm=True, seed=7)

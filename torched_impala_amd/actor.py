@@ -15,6 +15,7 @@ rows 0..T from it.
 from __future__ import annotations
 
 import logging
+import os
 
 import numpy as np
 import torch
@@ -337,6 +338,28 @@ def act(flags, actor_index: int, free_queue, full_queue, buffers, mirror,
             version = new_version
             claims += 1
             roll.fill_slot(buffers.slot(m), T)
+            if os.environ.get("IMPALA_DEBUG_NAN"):
+                s = buffers.slot(m)
+                if not torch.isfinite(s["behavior_logits"]).all():
+                    bad = (~torch.isfinite(s["behavior_logits"]))
+                    wfin = all(torch.isfinite(p).all()
+                               for p in model.parameters())
+                    h, c = (roll.core_state if roll.core_state else
+                            (torch.zeros(1), torch.zeros(1)))
+                    log.error(
+                        "NAN-DEBUG actor=%d claim=%d badrows=%d wfinite=%s "
+                        "hfin=%s cfin=%s hmax=%.3e cmax=%.3e ver=%d",
+                        actor_index, claims, int(bad.any(-1).sum()), wfin,
+                        torch.isfinite(h).all().item(),
+                        torch.isfinite(c).all().item(),
+                        h[torch.isfinite(h)].abs().max().item(),
+                        c[torch.isfinite(c)].abs().max().item(), version)
+                    dump = f"/tmp/nan_dump_a{actor_index}.pt"
+                    if wfin and not os.path.exists(dump):
+                        torch.save({"sd": model.state_dict(),
+                                    "slot": {k: v.clone()
+                                             for k, v in s.items()},
+                                    "h": h.clone(), "c": c.clone()}, dump)
             if not full_queue.push(m):
                 log.error("actor %d: full queue overflow", actor_index)
                 break

@@ -1091,8 +1091,14 @@ struct ShallowActorNet {
 
   // exp(x) via range reduction + degree-5 polynomial; |rel err| ~1e-7
   __attribute__((target("avx512f"))) static inline __m512 exp512(__m512 x) {
-    const __m512 hi = _mm512_set1_ps(88.3762626647950f);
-    const __m512 lo = _mm512_set1_ps(-87.3365478515625f);
+    // Clamp must stay clear of the 2^127 boundary: at x = 88.3762...
+    // round(x*log2e) lands on 127.5 in fp32 and rounds to 128, making
+    // the 2^n scale the INF bit pattern -> tanh = inf/inf = NaN. Seen
+    // in production once LSTM cell states exceeded |c| ~ 44.2 (2c hit
+    // the clamp). 87.0 keeps n <= 126 with zero accuracy cost for the
+    // saturated sigmoids/tanhs this serves.
+    const __m512 hi = _mm512_set1_ps(87.0f);
+    const __m512 lo = _mm512_set1_ps(-87.0f);
     x = _mm512_min_ps(_mm512_max_ps(x, lo), hi);
     const __m512 log2e = _mm512_set1_ps(1.44269504088896341f);
     const __m512 c1 = _mm512_set1_ps(0.693359375f);

@@ -49,3 +49,4 @@ def test_synthetic_env_shapes_and_determinism():
 def test_synthetic_dmlab_shape():
     e = SyntheticFrameEnv(72, 96, 3, num_actions=15, seed=0)
     assert e.reset().shape == (3, 72, 96)
+

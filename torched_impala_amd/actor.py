@@ -278,6 +278,10 @@ class Rollout:
         obs_buf = self._np_obs[self._obs_flip]
         rew = np.empty(self.n, dtype=np.float32)
         done = np.empty(self.n, dtype=bool)
+        # NOTE: a vectorised batch-env (numpy row gathers) was measured
+        # SLOWER than this loop (0.79 vs 0.70 ms/step, 16 envs): the
+        # synthetic step is trivially cheap and the obs copy dominates
+        # either way, so the loop stays.
         act = action.tolist()
         for i, env in enumerate(self.envs):
             o, r, d, _ = env.step(act[i])

@@ -59,39 +59,3 @@ class SyntheticFrameEnv:
     def close(self):
         pass
 
-
-class SyntheticVecFrameEnv:
-    """Vectorised batch of SyntheticFrameEnv with BITWISE-identical
-    trajectories to n individual instances (same seeds): the actor's
-    per-env python step loop (~75 us for 16 envs) becomes a handful of
-    numpy ops. Auto-resets on done, like the actor loop did."""
-
-    def __init__(self, envs):
-        assert envs and all(type(e) is SyntheticFrameEnv for e in envs)
-        n = len(envs)
-        e0 = envs[0]
-        self.n = n
-        self.obs_shape = e0.obs_shape
-        self.obs_dtype = e0.obs_dtype
-        self.num_actions = e0.num_actions
-        self.episode_len = np.array([e.episode_len for e in envs],
-                                    dtype=np.int64)
-        self.bank = e0.bank                       # shared across envs
-        self.reward_banks = np.stack([e.reward_bank for e in envs])
-        self.t = np.array([e.t for e in envs], dtype=np.int64)
-        self.gt = np.array([e.global_t for e in envs], dtype=np.int64)
-        self._idx = np.arange(n)
-
-    def step_all(self, actions, obs_out, rew_out, done_out):
-        """actions int64[n]; writes obs/reward/done in place (the same
-        record-then-auto-reset order as the actor's per-env loop)."""
-        self.t += 1
-        self.gt += 1
-        rew_out[:] = self.reward_banks[self._idx,
-                                       (self.gt + actions) % 1024]
-        done = self.t >= self.episode_len
-        done_out[:] = done
-        if done.any():                            # auto-reset
-            self.t[done] = 0
-            self.gt[done] += 1
-        obs_out[:] = self.bank[self.gt % SyntheticFrameEnv.FRAME_BANK]

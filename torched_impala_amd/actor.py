@@ -88,6 +88,9 @@ class NativeActorPolicy:
             assert model.core.hidden_size == 256
         self._ext = ext.ShallowActorNet(seed, bf16)
         self.bf16_active = bool(getattr(self._ext, "bf16_active", False))
+        if bf16 and not self.bf16_active:
+            log.info("native actor: avx512bf16 unavailable on this host — "
+                     "running the fp32 AVX-512/AVX2 tier instead")
         self._model = model
         self._A = num_actions
         self.refresh()

@@ -25,6 +25,11 @@ IMPALA_DEEP_TILES=1 ./torched_impala_amd/csrc/deep_probe 2>&1 \
     | tee $O/r2_probe_tiles.log
 IMPALA_DEEP_TILES=1 ./torched_impala_amd/csrc/deep_probe bench 2>&1 \
     | tee $O/r2_probe_tiles_bench.log
+IMPALA_DEEP_TILES=1 IMPALA_DEEP_SWIZZLE=1 \
+    ./torched_impala_amd/csrc/deep_probe 2>&1 | tee $O/r2_probe_swz.log
+IMPALA_DEEP_TILES=1 IMPALA_DEEP_SWIZZLE=1 \
+    ./torched_impala_amd/csrc/deep_probe bench 2>&1 \
+    | tee $O/r2_probe_swz_bench.log
 
 timeout 300 python scripts/actor_probe.py --jit native --actors 4 \
     --seconds 8 > $O/r2_actor_fp32.json 2> $O/r2_actor_fp32.err

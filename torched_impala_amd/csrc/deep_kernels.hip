@@ -37,7 +37,21 @@ __device__ __forceinline__ unsigned short relu_bits(unsigned short u) {
   return (u & 0x8000u) ? (unsigned short)0 : u;
 }
 
-constexpr int TY = 16, TX = 16;        // output tile (256 px = 16 M-frags)
+constexpr int TY = 16, TX = 16;       // default output tile (256 px)
+// XCD-aware block swizzle (gfx950: 8 XCDs, consecutive blockIdx round-
+// robin across them): remap so originally-consecutive-on-one-XCD blocks
+// decode to consecutive tiles of the same image, keeping that image's
+// patch/halo traffic inside one XCD's L2. Pure permutation when
+// gridDim.x is a multiple of 8; identity otherwise or when swz==0.
+__device__ __forceinline__ int swizzle_bid(int swz) {
+  int bid = blockIdx.x;
+  if (swz && (gridDim.x & 7) == 0) {
+    const int chunk = gridDim.x >> 3;
+    bid = (bid & 7) * chunk + (bid >> 3);
+  }
+  return bid;
+}
+
 constexpr int THREADS = 256;           // 4 waves
 
 // ---------------------------------------------------------------------------
@@ -59,7 +73,7 @@ template <int CIN, int COUT, bool RELU_IN, bool HAS_RES, int TTY = TY,
 __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
     const float* __restrict__ b, const bf16_t* __restrict__ res,
-    bf16_t* __restrict__ y, int N, int H, int W) {
+    bf16_t* __restrict__ y, int N, int H, int W, int swz) {
   constexpr int KTOT = 9 * CIN;
   constexpr int KSTEPS = (KTOT + 31) / 32;       // MFMA k-steps (zero-padded)
   constexpr int NFRAG = COUT / 16;
@@ -74,8 +88,9 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_nhwc(
 
   const int ntx = (W + TTX - 1) / TTX;
   const int nty = (H + TTY - 1) / TTY;
-  const int tile = blockIdx.x % (ntx * nty);
-  const int img = blockIdx.x / (ntx * nty);
+  const int bid = swizzle_bid(swz);
+  const int tile = bid % (ntx * nty);
+  const int img = bid / (ntx * nty);
   if (img >= N) return;
   const int ty0 = (tile / ntx) * TTY;            // tile origin (output px)
   const int tx0 = (tile % ntx) * TTX;
@@ -227,7 +242,7 @@ template <int CIN, int COUT, int TTY = TY, int TTX = TX>
 __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ wt,
     const bf16_t* __restrict__ xmask, bf16_t* __restrict__ dx, int N, int H,
-    int W) {
+    int W, int swz) {
   constexpr int KTOT = 9 * CIN;
   constexpr int KSTEPS = (KTOT + 31) / 32;
   constexpr int NFRAG = COUT / 16;
@@ -242,8 +257,9 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
 
   const int ntx = (W + TTX - 1) / TTX;
   const int nty = (H + TTY - 1) / TTY;
-  const int tile = blockIdx.x % (ntx * nty);
-  const int img = blockIdx.x / (ntx * nty);
+  const int bid = swizzle_bid(swz);
+  const int tile = bid % (ntx * nty);
+  const int img = bid / (ntx * nty);
   if (img >= N) return;
   const int ty0 = (tile / ntx) * TTY;
   const int tx0 = (tile % ntx) * TTX;
@@ -356,7 +372,7 @@ constexpr int DW_NSLAB = 256;
 template <int CIN, int COUT, bool RELU_IN, int TTY = TY, int TTX = TX>
 __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
-    float* __restrict__ dw_slab, int N, int H, int W) {
+    float* __restrict__ dw_slab, int N, int H, int W, int swz) {
   float* __restrict__ dw =
       dw_slab + (size_t)(blockIdx.x % DW_NSLAB) * (COUT * 9 * CIN);
   constexpr int MTOT = 9 * CIN;                  // 144 or 288
@@ -371,8 +387,9 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
 
   const int ntx = (W + TTX - 1) / TTX;
   const int nty = (H + TTY - 1) / TTY;
-  const int tile = blockIdx.x % (ntx * nty);
-  const int img = blockIdx.x / (ntx * nty);
+  const int bid = swizzle_bid(swz);
+  const int tile = bid % (ntx * nty);
+  const int img = bid / (ntx * nty);
   if (img >= N) return;
   const int ty0 = (tile / ntx) * TTY;
   const int tx0 = (tile % ntx) * TTX;
@@ -608,6 +625,15 @@ __global__ void mfma_probe(const bf16_t* __restrict__ A,
 // tile selection: 0 = 16x16 (validated default); under
 // IMPALA_DEEP_TILES=1 pick an exact-coverage tile for the deep shapes
 // (ROADMAP 3c — validate via deep_probe/round2_opening.sh first).
+static inline int use_swizzle() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("IMPALA_DEEP_SWIZZLE");
+    v = (e != nullptr && e[0] == '1') ? 1 : 0;
+  }
+  return v;
+}
+
 static inline int tile_id(int H, int W) {
   static int want = -1;
   if (want < 0) {
@@ -640,17 +666,17 @@ static inline int conv_grid_t(int N, int H, int W, int id) {
   if (relu_in) {                                                             \
     if (res)                                                                 \
       hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, true, TTY, TTX>),    \
-                         grid, blk, 0, s, xb, wb, b, rb, yb, N, H, W);       \
+                         grid, blk, 0, s, xb, wb, b, rb, yb, N, H, W, use_swizzle());       \
     else                                                                     \
       hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, true, false, TTY, TTX>),   \
-                         grid, blk, 0, s, xb, wb, b, nullptr, yb, N, H, W);  \
+                         grid, blk, 0, s, xb, wb, b, nullptr, yb, N, H, W, use_swizzle());  \
   } else {                                                                   \
     if (res)                                                                 \
       hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, true, TTY, TTX>),   \
-                         grid, blk, 0, s, xb, wb, b, rb, yb, N, H, W);       \
+                         grid, blk, 0, s, xb, wb, b, rb, yb, N, H, W, use_swizzle());       \
     else                                                                     \
       hipLaunchKernelGGL((conv3x3_nhwc<CIN, COUT, false, false, TTY, TTX>),  \
-                         grid, blk, 0, s, xb, wb, b, nullptr, yb, N, H, W);  \
+                         grid, blk, 0, s, xb, wb, b, nullptr, yb, N, H, W, use_swizzle());  \
   }
 
 #define CONV_LAUNCH(CIN, COUT)                                               \
@@ -717,19 +743,19 @@ extern "C" int launch_deep_conv3x3_dx(const void* dy, const void* wt,
   switch (tid_) {                                                           \
     case 1:                                                                 \
       hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO, 12, 16>), grid, blk, 0,   \
-                         s, d, w, m, o, N, H, W);                           \
+                         s, d, w, m, o, N, H, W, use_swizzle());                           \
       break;                                                                \
     case 2:                                                                 \
       hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO, 9, 24>), grid, blk, 0,    \
-                         s, d, w, m, o, N, H, W);                           \
+                         s, d, w, m, o, N, H, W, use_swizzle());                           \
       break;                                                                \
     case 3:                                                                 \
       hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO, 9, 12>), grid, blk, 0,    \
-                         s, d, w, m, o, N, H, W);                           \
+                         s, d, w, m, o, N, H, W, use_swizzle());                           \
       break;                                                                \
     default:                                                                \
       hipLaunchKernelGGL((conv3x3_dx_nhwc<CI, CO>), grid, blk, 0, s, d, w,  \
-                         m, o, N, H, W);                                    \
+                         m, o, N, H, W, use_swizzle());                                    \
       break;                                                                \
   }
   if (cout_fwd == 16 && cin_fwd == 16) { DX_CASE(16, 16) }
@@ -759,10 +785,10 @@ extern "C" int launch_deep_conv3x3_dw(const void* x, const void* dy,
 #define DW_CASE_T(CI, CO, TTY, TTX)                                         \
   if (relu_in)                                                              \
     hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, true, TTY, TTX>), grid,     \
-                       blk, 0, s, xb, db, dw, N, H, W);                     \
+                       blk, 0, s, xb, db, dw, N, H, W, use_swizzle());                     \
   else                                                                      \
     hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, false, TTY, TTX>), grid,    \
-                       blk, 0, s, xb, db, dw, N, H, W);
+                       blk, 0, s, xb, db, dw, N, H, W, use_swizzle());
 #define DW_CASE(CI, CO)                                                     \
   switch (tid_) {                                                           \
     case 1: DW_CASE_T(CI, CO, 12, 16) break;                                \

@@ -154,7 +154,9 @@ def test_slotqueue_fallback_mode():
     assert q.pop() == -1
     assert q.push(3) and q.push(5)
     assert q.pop_wait(timeout_ms=1000) == 3
-    assert q.pop() == 5
+    # mp.Queue enqueues via a feeder thread; a bare get_nowait can race
+    # the pipe flush under load -> use the bounded wait here too
+    assert q.pop_wait(timeout_ms=1000) == 5
     assert q.pop_wait(timeout_ms=50) == -1
     q.abort()
     assert q.pop_wait(timeout_ms=1000) == -2

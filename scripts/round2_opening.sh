@@ -48,6 +48,12 @@ fi
 
 timeout 600 python -m pytest tests -m gpu -x -q 2>&1 | tail -20 \
     | tee $O/r2_gpu_tests.log
+if [ "$PROBE_RC" = "0" ]; then
+  # the whole GPU suite again with the shallow-MFMA encoder switched in:
+  # the shallow train/graph tests then exercise the new path end to end
+  IMPALA_SHALLOW_MFMA=1 timeout 600 python -m pytest tests -m gpu -x -q \
+      2>&1 | tail -20 | tee $O/r2_gpu_tests_mfma.log
+fi
 
 echo "=== summary ==="
 grep -c PASS $O/r2_probe.log || true

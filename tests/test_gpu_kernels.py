@@ -31,7 +31,7 @@ def test_vtrace_scan_parity(T, B):
     values = torch.randn(T, B, device=dev)
     boot = torch.randn(B, device=dev)
     vs, adv = ops.ext_hip().vtrace_scan(
-        log_rhos, discounts, rewards, values, boot, 1.0, 1.0, 1.0)
+        log_rhos, discounts, rewards, values, boot, 1.0, 1.0, 1.0, 1.0)
     ref = vtrace.from_importance_weights(
         log_rhos.cpu(), discounts.cpu(), rewards.cpu(), values.cpu(),
         boot.cpu())
@@ -121,7 +121,7 @@ def test_scan_handles_done_boundaries():
     values = torch.randn(T, B, device=dev)
     boot = torch.randn(B, device=dev)
     vs, adv = ops.ext_hip().vtrace_scan(
-        log_rhos, discounts, rewards, values, boot, 1.0, 1.0, 1.0)
+        log_rhos, discounts, rewards, values, boot, 1.0, 1.0, 1.0, 1.0)
     ref = vtrace.from_importance_weights(
         log_rhos.cpu(), discounts.cpu(), rewards.cpu(), values.cpu(), boot.cpu())
     torch.testing.assert_close(vs.cpu(), ref.vs, rtol=1e-5, atol=1e-5)
@@ -309,12 +309,14 @@ def test_vtrace_scan_parity_nondefault_clips(clip_rho, clip_c, lam):
     rewards = torch.randn(T, B, device=dev)
     values = torch.randn(T, B, device=dev)
     boot = torch.randn(B, device=dev)
+    pg_rho = 0.5 * clip_rho  # exercise an independent pg-rho threshold
     vs, adv = ops.ext_hip().vtrace_scan(
-        log_rhos, discounts, rewards, values, boot, clip_rho, clip_c, lam)
+        log_rhos, discounts, rewards, values, boot, clip_rho, clip_c, lam,
+        pg_rho)
     ref = vtrace.from_importance_weights(
         log_rhos.cpu(), discounts.cpu(), rewards.cpu(), values.cpu(),
         boot.cpu(), clip_rho_threshold=clip_rho, clip_c_threshold=clip_c,
-        lambda_=lam, clip_pg_rho_threshold=clip_rho)
+        lambda_=lam, clip_pg_rho_threshold=pg_rho)
     torch.testing.assert_close(vs.cpu(), ref.vs, rtol=1e-5, atol=1e-5)
     torch.testing.assert_close(adv.cpu(), ref.pg_advantages, rtol=1e-5,
                                atol=1e-5)

@@ -332,17 +332,21 @@ def act(flags, actor_index: int, free_queue, full_queue, buffers, mirror,
         T = flags.unroll_length
         refresh_every = max(1, getattr(flags, "actor_jit_refresh", 4))
         claims = 0
+        refresh_pending = False  # model updated but weights not re-baked
         while True:
             m = free_queue.pop_wait()
             if m < 0:  # abort/sentinel
                 break
             new_version = mirror.pull_into(model, last_seen=version)
+            if new_version != version:
+                refresh_pending = True
+                version = new_version
             if (roll.fast is not None
                     and roll.fast.mode in ("frozen", "native")
-                    and new_version != version
+                    and refresh_pending
                     and claims % refresh_every == 0):
                 roll.fast.refresh()  # re-bake/re-pack the weights
-            version = new_version
+                refresh_pending = False
             claims += 1
             roll.fill_slot(buffers.slot(m), T)
             if os.environ.get("IMPALA_DEBUG_NAN"):

@@ -77,7 +77,10 @@ def from_logits(
     clip_rho_threshold: float = 1.0,
     clip_c_threshold: float = 1.0,
     lambda_: float = 1.0,
+    clip_pg_rho_threshold: float = None,
 ) -> VTraceFromLogitsReturns:
+    if clip_pg_rho_threshold is None:
+        clip_pg_rho_threshold = clip_rho_threshold
     target_lp = action_log_probs(target_policy_logits, actions)
     behavior_lp = action_log_probs(behavior_policy_logits, actions)
     log_rhos = (target_lp - behavior_lp).detach()
@@ -87,7 +90,8 @@ def from_logits(
         values=values.detach(), bootstrap_value=bootstrap_value.detach(),
         clip_rho_threshold=clip_rho_threshold,
         clip_c_threshold=clip_c_threshold,
-        lambda_=lambda_)
+        lambda_=lambda_,
+        clip_pg_rho_threshold=clip_pg_rho_threshold)
     return VTraceFromLogitsReturns(
         vs=vt.vs, pg_advantages=vt.pg_advantages, log_rhos=log_rhos,
         behavior_action_log_probs=behavior_lp,

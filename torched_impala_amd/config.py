@@ -52,6 +52,9 @@ class Flags:
     reward_clipping: str = "abs_one"   # abs_one | none
     clip_rho_threshold: float = 1.0
     clip_c_threshold: float = 1.0
+    # rho-bar for the PG advantage; None => same as clip_rho_threshold
+    # (the HIP scan and eager paths are A/B-parity-tested on this value)
+    clip_pg_rho_threshold: float = None
     lambda_: float = 1.0
 
     # --- optimizer (RMSProp, paper defaults) ---
@@ -123,6 +126,8 @@ def build_parser() -> argparse.ArgumentParser:
                    default=f.clip_rho_threshold)
     p.add_argument("--clip-c-threshold", type=float,
                    default=f.clip_c_threshold)
+    p.add_argument("--clip-pg-rho-threshold", type=float,
+                   default=f.clip_pg_rho_threshold)
     p.add_argument("--lambda", dest="lambda_", type=float, default=f.lambda_)
     p.add_argument("--learning-rate", type=float, default=f.learning_rate)
     p.add_argument("--alpha", type=float, default=f.alpha)

@@ -25,8 +25,8 @@ extern "C" {
 void launch_vtrace_row_fwd(const float*, const float*, const long*, int, int,
                            float*, float*, float*, float*, hipStream_t);
 void launch_vtrace_scan(const float*, const float*, const float*, const float*,
-                        const float*, int, int, float, float, float, float*,
-                        float*, hipStream_t);
+                        const float*, int, int, float, float, float, float,
+                        float*, float*, hipStream_t);
 void launch_vtrace_loss_bwd(const float*, const float*, const long*,
                             const float*, const float*, const float*, float,
                             float, int, int, float*, float*, hipStream_t);
@@ -141,7 +141,7 @@ std::vector<torch::Tensor> vtrace_loss_fwd(
     torch::Tensor target_logits, torch::Tensor behavior_logits,
     torch::Tensor actions, torch::Tensor rewards, torch::Tensor discounts,
     torch::Tensor values, torch::Tensor bootstrap_value, double clip_rho,
-    double clip_c, double lambda_) {
+    double clip_c, double lambda_, double clip_pg_rho) {
   check_f32(target_logits, "target_logits");
   check_f32(behavior_logits, "behavior_logits");
   check_f32(rewards, "rewards");
@@ -173,7 +173,8 @@ std::vector<torch::Tensor> vtrace_loss_fwd(
       log_rho.data_ptr<float>(), discounts.data_ptr<float>(),
       rewards.data_ptr<float>(), values.data_ptr<float>(),
       bootstrap_value.data_ptr<float>(), T, B, (float)clip_rho, (float)clip_c,
-      (float)lambda_, vs.data_ptr<float>(), pg_adv.data_ptr<float>(), stream);
+      (float)lambda_, (float)clip_pg_rho, vs.data_ptr<float>(),
+      pg_adv.data_ptr<float>(), stream);
   HIP_CHECK(hipGetLastError());
   return {logp, tgt_lp, row_negent, vs, pg_adv};
 }
@@ -182,7 +183,7 @@ std::vector<torch::Tensor> vtrace_loss_fwd(
 std::vector<torch::Tensor> vtrace_scan(
     torch::Tensor log_rhos, torch::Tensor discounts, torch::Tensor rewards,
     torch::Tensor values, torch::Tensor bootstrap_value, double clip_rho,
-    double clip_c, double lambda_) {
+    double clip_c, double lambda_, double clip_pg_rho) {
   check_f32(log_rhos, "log_rhos");
   const int T = log_rhos.size(0);
   const int B = log_rhos.size(1);
@@ -193,8 +194,8 @@ std::vector<torch::Tensor> vtrace_scan(
       log_rhos.data_ptr<float>(), discounts.data_ptr<float>(),
       rewards.data_ptr<float>(), values.data_ptr<float>(),
       bootstrap_value.data_ptr<float>(), T, B, (float)clip_rho, (float)clip_c,
-      (float)lambda_, vs.data_ptr<float>(), pg_adv.data_ptr<float>(),
-      current_stream());
+      (float)lambda_, (float)clip_pg_rho, vs.data_ptr<float>(),
+      pg_adv.data_ptr<float>(), current_stream());
   HIP_CHECK(hipGetLastError());
   return {vs, pg_adv};
 }

@@ -71,14 +71,14 @@ __global__ void vtrace_row_fwd(
 // stripe are staged into LDS with coalesced row loads up-front (every
 // load in flight at once), then the serial T-recursion runs out of LDS:
 //   A_t = delta_t + disc_t * c_t * A_{t+1};  vs_t = V_t + A_t
-//   pg_adv_t = crho_t * (r_t + disc_t * vs_{t+1} - V_t)
+//   pg_adv_t = min(rho_t, clip_pg_rho) * (r_t + disc_t * vs_{t+1} - V_t)
 // LDS: 4 arrays * T * 64 * 4 B  (T=100 -> 100 KiB; fits 160 KiB/CU).
 // ---------------------------------------------------------------------------
 __global__ void vtrace_scan_lds(
     const float* __restrict__ log_rho, const float* __restrict__ discounts,
     const float* __restrict__ rewards, const float* __restrict__ values,
     const float* __restrict__ bootstrap, int T, int B,
-    float clip_rho, float clip_c, float lambda_,
+    float clip_rho, float clip_c, float lambda_, float clip_pg_rho,
     float* __restrict__ vs, float* __restrict__ pg_adv) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* s_rho = smem;              // [T][64]
@@ -119,7 +119,7 @@ __global__ void vtrace_scan_lds(
     const float vs_t = v + acc;
     const size_t idx = (size_t)t * B + b;
     vs[idx] = vs_t;
-    pg_adv[idx] = crho * fmaf(disc, vs_next, r - v);
+    pg_adv[idx] = fminf(rho, clip_pg_rho) * fmaf(disc, vs_next, r - v);
     vs_next = vs_t;
     v_next = v;
   }
@@ -130,7 +130,7 @@ __global__ void vtrace_scan_global(
     const float* __restrict__ log_rho, const float* __restrict__ discounts,
     const float* __restrict__ rewards, const float* __restrict__ values,
     const float* __restrict__ bootstrap, int T, int B,
-    float clip_rho, float clip_c, float lambda_,
+    float clip_rho, float clip_c, float lambda_, float clip_pg_rho,
     float* __restrict__ vs, float* __restrict__ pg_adv) {
   const int b = blockIdx.x * blockDim.x + threadIdx.x;
   if (b >= B) return;
@@ -149,7 +149,7 @@ __global__ void vtrace_scan_global(
     acc = fmaf(disc * c, acc, delta);
     const float vs_t = v + acc;
     vs[idx] = vs_t;
-    pg_adv[idx] = crho * fmaf(disc, vs_next, r - v);
+    pg_adv[idx] = fminf(rho, clip_pg_rho) * fmaf(disc, vs_next, r - v);
     vs_next = vs_t;
     v_next = v;
   }
@@ -205,20 +205,22 @@ void launch_vtrace_row_fwd(const float* tgt, const float* beh,
 void launch_vtrace_scan(const float* log_rho, const float* discounts,
                         const float* rewards, const float* values,
                         const float* bootstrap, int T, int B, float clip_rho,
-                        float clip_c, float lambda_, float* vs, float* pg_adv,
-                        hipStream_t stream) {
+                        float clip_c, float lambda_, float clip_pg_rho,
+                        float* vs, float* pg_adv, hipStream_t stream) {
   const size_t lds_bytes = 4ull * T * 64 * sizeof(float);
   if (lds_bytes <= 160 * 1024 - 1024) {
     const int grid = (B + 63) / 64;
     hipLaunchKernelGGL(impala::vtrace_scan_lds, dim3(grid), dim3(64),
                        lds_bytes, stream, log_rho, discounts, rewards, values,
-                       bootstrap, T, B, clip_rho, clip_c, lambda_, vs, pg_adv);
+                       bootstrap, T, B, clip_rho, clip_c, lambda_,
+                       clip_pg_rho, vs, pg_adv);
   } else {
     const int block = 256;
     const int grid = (B + block - 1) / block;
     hipLaunchKernelGGL(impala::vtrace_scan_global, dim3(grid), dim3(block), 0,
                        stream, log_rho, discounts, rewards, values, bootstrap,
-                       T, B, clip_rho, clip_c, lambda_, vs, pg_adv);
+                       T, B, clip_rho, clip_c, lambda_, clip_pg_rho, vs,
+                       pg_adv);
   }
 }
 

@@ -213,6 +213,7 @@ def _fwd_bwd(flags: Flags, model, batch: Dict[str, torch.Tensor],
         actions, rewards, discounts,
         clip_rho=flags.clip_rho_threshold, clip_c=flags.clip_c_threshold,
         lambda_=flags.lambda_,
+        clip_pg_rho=getattr(flags, "clip_pg_rho_threshold", None),
         baseline_cost=flags.baseline_cost, entropy_cost=flags.entropy_cost,
         force_eager=not flags.use_hip_kernels)
 
@@ -290,12 +291,22 @@ class GraphedStep:
                 stats["entropy_loss"])
 
     def _capture(self, i: int):
+        # warmup runs real optimizer updates (world==1 captures
+        # step_device inside _fn) — snapshot the flat state so capture
+        # doesn't bake in up to 2*WARMUP_ITERS uncounted weight updates
+        # from the same batch
+        snap = (self.optimizer.flat_param.clone(),
+                self.optimizer.flat_sqavg.clone())
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(self.WARMUP_ITERS):
                 self._fn(i)
         torch.cuda.current_stream().wait_stream(s)
+        with torch.no_grad():
+            self.optimizer.flat_param.copy_(snap[0])
+            self.optimizer.flat_sqavg.copy_(snap[1])
+        del snap
         g = torch.cuda.CUDAGraph()
         ctx = (torch.cuda.graph(g, pool=self.pool) if self.pool is not None
                else torch.cuda.graph(g))
@@ -411,16 +422,41 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     run_dir = None
     ckpt_path = None
     resume_frames = 0
-    if flags.xpid and rank == 0:
+    if flags.xpid:
+        # every rank computes the path and loads the checkpoint: ranks
+        # that resumed from different states (or none) would all-reduce
+        # gradients across divergent models and exit the step loop at
+        # different iteration counts, hanging the others in collectives.
+        # Single node => shared filesystem, so all ranks read the same
+        # (atomically-replaced) file; rank 0 alone writes it.
         run_dir = os.path.join(os.path.expanduser(flags.savedir), flags.xpid)
-        os.makedirs(run_dir, exist_ok=True)
+        if rank == 0:
+            os.makedirs(run_dir, exist_ok=True)
         ckpt_path = os.path.join(run_dir, "model.tar")
+        if world > 1:
+            torch.distributed.barrier()  # rank0 mkdir before others look
         if os.path.exists(ckpt_path) and not flags.disable_checkpoint:
             ckpt = ckpt_mod.load(ckpt_path, model, optimizer, scheduler,
                                  map_location=device)
             resume_frames = int(ckpt.get("step", 0))
-            log.info("resumed from %s at step=%d", ckpt_path, resume_frames)
+            log.info("rank %d resumed from %s at step=%d",
+                     rank, ckpt_path, resume_frames)
             mirror.publish_state_dict(model.state_dict())
+        if world > 1:
+            # belt and braces: bitwise-identical start even if a rank
+            # raced the file; also agree on the resume step.
+            if isinstance(optimizer, FlatRMSprop):
+                torch.distributed.broadcast(optimizer.flat_param, src=0)
+                torch.distributed.broadcast(optimizer.flat_sqavg, src=0)
+            else:
+                for p in model.parameters():
+                    torch.distributed.broadcast(p.data, src=0)
+            rf = torch.tensor([resume_frames], dtype=torch.long,
+                              device=device)
+            torch.distributed.broadcast(rf, src=0)
+            resume_frames = int(rf.item())
+            if resume_frames:
+                mirror.publish_state_dict(model.state_dict())
 
     frames_per_batch = flags.unroll_length * flags.batch_size
     step_frames = resume_frames
@@ -526,7 +562,7 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
                               **{k: float(v) for k, v in stats.items()})
                 last_log, last_log_frames = now, step_frames
 
-            if (ckpt_path and not flags.disable_checkpoint
+            if (ckpt_path and rank == 0 and not flags.disable_checkpoint
                     and now - last_ckpt > flags.checkpoint_interval_s):
                 ckpt_mod.save(ckpt_path, model, optimizer, scheduler,
                               step_frames, flags)
@@ -543,7 +579,7 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
             if p.is_alive():
                 p.terminate()
         metrics.close()
-    if ckpt_path and not flags.disable_checkpoint:
+    if ckpt_path and rank == 0 and not flags.disable_checkpoint:
         ckpt_mod.save(ckpt_path, model, optimizer, scheduler, step_frames, flags)
     out = {k: float(v) for k, v in stats.items()}
     out.update(step_frames=step_frames, learner_steps=learner_steps,

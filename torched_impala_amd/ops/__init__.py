@@ -90,7 +90,8 @@ class _FusedVtraceLoss(torch.autograd.Function):
     @staticmethod
     def forward(ctx, target_logits, values, bootstrap_value,
                 behavior_logits, actions, rewards, discounts,
-                clip_rho, clip_c, lambda_, baseline_cost, entropy_cost):
+                clip_rho, clip_c, lambda_, clip_pg_rho,
+                baseline_cost, entropy_cost):
         ext = ext_hip()
         T, B, A = target_logits.shape
         logp, target_lp, row_negent, vs, pg_adv = ext.vtrace_loss_fwd(
@@ -98,7 +99,8 @@ class _FusedVtraceLoss(torch.autograd.Function):
             actions.contiguous(), rewards.contiguous(),
             discounts.contiguous(), values.contiguous(),
             bootstrap_value.contiguous(),
-            float(clip_rho), float(clip_c), float(lambda_))
+            float(clip_rho), float(clip_c), float(lambda_),
+            float(clip_pg_rho))
         pg = -(target_lp * pg_adv.flatten()).sum()
         vdiff = vs - values.detach()  # (T,B)
         bl = 0.5 * (vdiff * vdiff).sum()
@@ -119,24 +121,29 @@ class _FusedVtraceLoss(torch.autograd.Function):
         dlogits, dvalues = ext.vtrace_loss_bwd(
             logp, row_negent, actions, pg_adv.contiguous(),
             vdiff.contiguous(), grad_scale, baseline_cost, entropy_cost)
-        return (dlogits, dvalues) + (None,) * 10
+        return (dlogits, dvalues) + (None,) * 11
 
 
 def fused_vtrace_loss(target_logits, values, bootstrap_value,
                       behavior_logits, actions, rewards, discounts,
                       clip_rho=1.0, clip_c=1.0, lambda_=1.0,
+                      clip_pg_rho=None,
                       baseline_cost=0.5, entropy_cost=0.01,
                       force_eager: bool = False):
     """Returns (total_loss, stats dict, vs, pg_advantages).
 
     target_logits (T,B,A) and values (T,B) carry grad; everything else is
     data. Dispatches to the HIP fused path on GPU, eager elsewhere.
+    clip_pg_rho (rho-bar for the PG advantage) defaults to clip_rho so
+    the HIP and eager paths agree for any --clip-rho-threshold.
     """
+    if clip_pg_rho is None:
+        clip_pg_rho = clip_rho
     if not force_eager and hip_available() and target_logits.is_cuda:
         loss, pg, bl, ent, vs, adv = _FusedVtraceLoss.apply(
             target_logits, values, bootstrap_value, behavior_logits,
             actions, rewards, discounts, clip_rho, clip_c, lambda_,
-            baseline_cost, entropy_cost)
+            clip_pg_rho, baseline_cost, entropy_cost)
         stats = dict(pg_loss=pg.detach(), baseline_loss=bl.detach(),
                      entropy_loss=ent.detach())
         return loss, stats, vs, adv
@@ -146,7 +153,8 @@ def fused_vtrace_loss(target_logits, values, bootstrap_value,
     vt = VT.from_logits(
         behavior_logits, target_logits, actions,
         discounts, rewards, values, bootstrap_value,
-        clip_rho_threshold=clip_rho, clip_c_threshold=clip_c, lambda_=lambda_)
+        clip_rho_threshold=clip_rho, clip_c_threshold=clip_c, lambda_=lambda_,
+        clip_pg_rho_threshold=clip_pg_rho)
     loss, stats = L.total_loss(
         target_logits, actions, vt.pg_advantages, vt.vs, values,
         baseline_cost, entropy_cost)

@@ -582,6 +582,9 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     if ckpt_path and rank == 0 and not flags.disable_checkpoint:
         ckpt_mod.save(ckpt_path, model, optimizer, scheduler, step_frames, flags)
     out = {k: float(v) for k, v in stats.items()}
+    # deterministic param digest: DP ranks in lockstep agree bitwise
+    out["param_sum"] = float(sum(p.data.double().sum().item()
+                                 for p in model.parameters()))
     out.update(step_frames=step_frames, learner_steps=learner_steps,
                mean_return=metrics.mean_return or float("nan"),
                fps=step_frames / max(time.time() - t_start, 1e-9),

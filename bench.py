@@ -28,8 +28,8 @@ import torch
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=30)
-    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--steps", type=int, default=40)
+    p.add_argument("--warmup", type=int, default=12)
     p.add_argument("--env", default="synthetic-atari")
     p.add_argument("--batch-size", type=int, default=32)
     p.add_argument("--unroll-length", type=int, default=20)
@@ -98,6 +98,19 @@ def main():
     print("[bench] stage timers (whole run):",
           train_stats.get("timers"), file=sys.stderr)
 
+    # a silent eager fallback must never be reported as the product
+    # number: when graphs were requested on GPU, the graph path has to
+    # have stayed active for the whole timed region.
+    graph_active = bool(train_stats.get("graph_path_active"))
+    if is_cuda and not args.no_graphs and not graph_active:
+        raise RuntimeError(
+            "bench invalid: hipGraph step path fell back to eager during "
+            "the run — fix the graph path or pass --no-graphs explicitly")
+
+    from torched_impala_amd.actor import resolve_actor_tier
+    achieved_tier = resolve_actor_tier(
+        args.actor_jit) if flags.env == "synthetic-atari" else args.actor_jit
+
     elapsed = state["t1"] - state["t0"]
     rank = int(os.environ.get("RANK", "0"))
     # max over ranks
@@ -133,7 +146,11 @@ def main():
                 "seq_len": flags.unroll_length,
                 "parallelism": f"dp{n_gpus}",
                 "num_actors_per_gpu": flags.num_actors,
-                "actor_jit": flags.actor_jit,
+                "actor_jit_requested": flags.actor_jit,
+                "actor_jit": achieved_tier,
+                "graph_path_active": graph_active,
+                "stage_timers_s": {k: round(v, 4) for k, v in
+                                   (train_stats.get("timers") or {}).items()},
                 "envs_per_actor": flags.envs_per_actor,
                 "frame_convention": "agent_steps",
                 "env_frames_per_sec_x4_convention": value * 4,

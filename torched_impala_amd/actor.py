@@ -29,6 +29,24 @@ ROW_KEYS = ("frame", "reward", "done", "action", "behavior_logits",
             "last_action", "episode_return", "episode_step")
 
 
+def resolve_actor_tier(requested: str, obs_shape=(4, 84, 84)) -> str:
+    """The tier that will actually run on THIS host for the requested
+    --actor-jit flag (native-bf16 silently runs the fp32 native tier on
+    hosts without avx512bf16; benchmarks must report the achieved tier,
+    not the requested one — VERDICT r1)."""
+    if requested != "native-bf16":
+        return requested
+    if tuple(obs_shape) != (4, 84, 84):
+        return "frozen"
+    try:
+        import torched_impala_amd._impala_actor as ext
+        probe = ext.ShallowActorNet(0, True)
+        return ("native-bf16" if getattr(probe, "bf16_active", False)
+                else "native")
+    except Exception:
+        return "frozen"
+
+
 class _FlatActorFwd(torch.nn.Module):
     """Feedforward-policy actor forward as one traceable graph: encode ->
     core-input concat -> policy head -> sample. Skips the baseline head

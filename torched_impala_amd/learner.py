@@ -585,6 +585,10 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     # deterministic param digest: DP ranks in lockstep agree bitwise
     out["param_sum"] = float(sum(p.data.double().sum().item()
                                  for p in model.parameters()))
+    # did the hipGraph step survive the whole run? (a silent eager
+    # fallback must never be reported as the graph-path number)
+    out["graph_path_active"] = bool(graphed is not None) if use_graphs \
+        else False
     out.update(step_frames=step_frames, learner_steps=learner_steps,
                mean_return=metrics.mean_return or float("nan"),
                fps=step_frames / max(time.time() - t_start, 1e-9),

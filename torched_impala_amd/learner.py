@@ -384,9 +384,21 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
         for p in model.parameters():
             torch.distributed.broadcast(p.data, src=0)
 
+    optimizer, scheduler = build_optimizer(flags, model, world)
+
     cpu_model = build_model(flags, obs_shape, num_actions)
     cpu_model.load_state_dict({k: v.cpu() for k, v in model.state_dict().items()})
-    mirror = WeightMirror(cpu_model)
+    flat_layout = None
+    if isinstance(optimizer, FlatRMSprop):
+        name_of = {p: n for n, p in model.named_parameters()}
+        flat_layout = {name_of[p]: optimizer.slices[p]
+                       for p in optimizer._order if p in name_of}
+    mirror = WeightMirror(cpu_model, flat_layout=flat_layout)
+    use_flat_publish = (mirror.layout is not None
+                        and isinstance(optimizer, FlatRMSprop)
+                        and device.type == "cuda")
+    if use_flat_publish:
+        mirror.pin()  # page-lock the flat shm arenas for async D2H
 
     buffers = create_buffers(flags, obs_shape, obs_dtype, num_actions)
     free_queue = SlotQueue(buffers.num_slots)
@@ -399,7 +411,6 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
                   obs_shape, obs_dtype, num_actions)
     procs = start_actors(*actor_args)
 
-    optimizer, scheduler = build_optimizer(flags, model, world)
     batcher = Batcher(flags, buffers, free_queue, full_queue, device)
     graphed = None
     reducer = None
@@ -517,8 +528,15 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
 
             if learner_steps % flags.weight_sync_interval == 0:
                 timers.start("weight_publish")
-                mirror.publish_state_dict(model.state_dict())
+                if use_flat_publish:
+                    # one D2D snapshot + one async D2H; seqlock flips on
+                    # completion (poll) — host cost is an event query
+                    mirror.publish_flat_async(optimizer.flat_param)
+                else:
+                    mirror.publish_state_dict(model.state_dict())
                 timers.stop()
+            elif use_flat_publish:
+                mirror.poll_publish()
 
             # episode metrics (cheap on CPU; sync-inducing on GPU -> sparse)
             if device.type != "cuda" or learner_steps % 20 == 0:

@@ -51,7 +51,9 @@ def test_graphed_step_matches_eager(use_lstm):
     flags = Flags(env="synthetic-atari", batch_size=8, unroll_length=10,
                   use_lstm=use_lstm, total_steps=1 << 50)
     n_steps = 2
-    total_updates = L.GraphedStep.WARMUP_ITERS + n_steps
+    # warmup updates are rolled back before capture (ADVICE r1 fix), so
+    # the graphed run applies exactly n_steps parameter updates
+    total_updates = n_steps
 
     def make():
         torch.manual_seed(42)
@@ -106,3 +108,55 @@ def test_graphed_two_buffers_shared_pool():
         losses.append(float(stats["total_loss"]))
     assert all(x == x for x in losses), f"NaN loss in {losses}"
     assert len(graphed.graphs) == 2
+
+
+@pytest.mark.timeout(600)
+def test_graphed_step_world2_stub_equals_world1(monkeypatch):
+    """The exact world>1 step sequence (graph replay -> flat all-reduce
+    -> eager fused step, learner.py GraphedStep.step) on ONE GPU with a
+    stubbed collective simulating a second identical rank (flat_grad
+    *= 2 against the optimizer's 1/world scale): must match the world=1
+    path step for step, bitwise."""
+    device = torch.device("cuda", 0)
+    obs, A = (4, 84, 84), 6
+    flags = Flags(env="synthetic-atari", batch_size=8, unroll_length=10,
+                  use_lstm=False, total_steps=1 << 50)
+    n_steps = 4
+
+    def run(world):
+        torch.manual_seed(42)
+        model = build_model(flags, obs, A).to(device)
+        opt, sched = L.build_optimizer(flags, model, world_size=world)
+        st = _staging(flags, obs, A, device)
+        stub = _StubBatcher(flags, st)
+        graphed = L.GraphedStep(flags, model, opt, stub, world_size=world)
+        traj = []
+        for _ in range(n_steps):
+            graphed.step(0, sched)
+            torch.cuda.synchronize()
+            traj.append(opt.flat_param.clone())
+        return traj
+
+    traj1 = run(1)
+
+    import torch.distributed as dist
+
+    class _Work:
+        def wait(self):
+            pass
+
+    calls = []
+
+    def fake_all_reduce(t, op=None, group=None, async_op=False):
+        calls.append(tuple(t.shape))
+        t.mul_(2.0)  # the second identical rank's contribution (SUM)
+        return _Work()
+
+    monkeypatch.setattr(dist, "all_reduce", fake_all_reduce)
+    traj2 = run(2)
+    assert len(calls) == n_steps, "flat all-reduce must run once per step"
+    for i, (a, b) in enumerate(zip(traj1, traj2)):
+        torch.testing.assert_close(
+            a, b, rtol=0, atol=0,
+            msg=lambda m, i=i: f"step {i} diverged between world=1 "
+                               f"and stubbed world=2: {m}")

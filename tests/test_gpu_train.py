@@ -84,3 +84,57 @@ def test_smoke_entry():
     sys.path.insert(0, ROOT)
     import __graft_entry__
     __graft_entry__.smoke()
+
+
+@pytest.mark.timeout(420)
+def test_train_world2_stub_single_gpu():
+    """The full train() world>1 path on ONE GPU with a stubbed process
+    group: capture-agreement MIN-reduce, graph
+    replay, flat SUM all-reduce (doubling, against the 1/world scale),
+    frame accounting *world, graph path staying active. A second
+    world=1 run with identical seed would see different actor batches,
+    so this asserts protocol/liveness, not numeric equality (that is
+    test_gpu_graphs.test_graphed_step_world2_stub_equals_world1)."""
+    code = """
+import json
+import torch
+import torch.distributed as dist
+from torched_impala_amd.config import Flags
+from torched_impala_amd import learner
+
+# stub process group: one GPU pretends to be rank 0 of world 2
+learner.init_distributed = lambda device: (0, 2)
+
+class _W:
+    def wait(self): pass
+
+reduce_ops = []
+def fake_all_reduce(t, op=None, group=None, async_op=False):
+    reduce_ops.append(str(op))
+    if op == dist.ReduceOp.SUM or op is None:
+        t.mul_(2.0)   # the second identical rank's contribution
+    return _W()
+dist.all_reduce = fake_all_reduce
+dist.broadcast = lambda t, src=0, group=None, async_op=False: _W()
+dist.barrier = lambda *a, **k: None
+
+flags = Flags(env="synthetic-atari", num_actors=4, batch_size=8,
+              unroll_length=20, total_steps=1 << 40,
+              disable_checkpoint=True, device="cuda", seed=9)
+stats = learner.train(flags, max_learner_steps=6)
+assert stats["learner_steps"] == 6, stats
+# frames counted with the world multiplier
+assert stats["step_frames"] == 6 * 8 * 20 * 2, stats
+assert stats["graph_path_active"], "graph path must survive world>1 stub"
+# one MIN capture-agreement reduce + one SUM flat-grad reduce per step
+mins = [o for o in reduce_ops if "MIN" in o]
+sums = [o for o in reduce_ops if "SUM" in o]
+assert len(mins) == 1, reduce_ops
+assert len(sums) == 6, reduce_ops
+for k in ("total_loss", "pg_loss", "baseline_loss", "entropy_loss"):
+    v = stats[k]
+    assert v == v, f"{k} is NaN"
+print("TRAIN_OK")
+"""
+    out = _run_train(code)
+    assert "TRAIN_OK" in out

@@ -317,6 +317,17 @@ class GraphedStep:
         self.graphs[i] = g
         log.info("captured learn-step graph for staging buffer %d", i)
 
+    def destroy(self):
+        """Deterministic teardown: the hipGraphExec objects and their
+        shared mempool must not be destroyed by a lazy GC while replays
+        or side-stream copies are still in flight (suspected source of
+        the round-1 flaky async hipErrorInvalidValue)."""
+        if self.graphs:
+            torch.cuda.synchronize()
+        self.graphs.clear()
+        self.outputs.clear()
+        self.pool = None
+
     def step(self, i: int, scheduler) -> Dict[str, torch.Tensor]:
         self.optimizer.sync_lr()
         if i not in self.graphs:
@@ -591,6 +602,8 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
         free_queue.abort()
         full_queue.abort()
         batcher.shutdown()
+        if graphed is not None:
+            graphed.destroy()
         for p in procs:
             p.join(timeout=5.0)
         for p in procs:

@@ -635,10 +635,14 @@ static inline int use_swizzle() {
 }
 
 static inline int tile_id(int H, int W) {
+  // exact-coverage tiles are DEFAULT-ON since round 2: parity-clean on
+  // hardware (gpurun_out/r2_probe_tiles.log, all PASS) and 1.3-1.9x on
+  // the 18x24/9x12 layers (profiles/r2_deep_tiles_bench.md).
+  // IMPALA_DEEP_TILES=0 restores the 16x16 tiling for A/B.
   static int want = -1;
   if (want < 0) {
     const char* e = getenv("IMPALA_DEEP_TILES");
-    want = (e != nullptr && e[0] == '1') ? 1 : 0;
+    want = (e != nullptr && e[0] == '0') ? 0 : 1;
   }
   if (!want) return 0;
   if (H % 12 == 0 && W % 16 == 0) return 1;   // 36x48, 72x96

@@ -356,27 +356,32 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
 
 // ---------------------------------------------------------------------------
 // conv3x3 backward-to-weights: dW[f][ky][kx][c] += sum over px of
-// act(x)[px + tap] * dY[px].  Per WG: one 16x16 px tile; MFMA with
-// M = 9*CIN (taps x cin), N = COUT, K = 256 px; fp32 atomicAdd into the
-// global dW (fp32 rounding-order nondeterminism only — ROADMAP notes a
-// slab+reduce variant for bitwise determinism).  A/B gathers are
-// k(px)-strided scalar reads — correctness-first; ds_read_b64_tr_b16 is
-// the round-2 lever.
+//   act(x)[px + tap] * dY[px].  MFMA with M = 9*CIN (taps x cin),
+//   N = COUT, K = px.
+//
+// Grid-stride accumulation (round 2): each workgroup loops over many
+// (image, px-tile) pairs, holding the full D fragment set for its
+// m-frags IN REGISTERS across the loop (dW is px-independent), and
+// performs ONE atomic epilogue at the end.  The round-1 per-tile
+// version issued COUT*9*CIN atomicAdds per 16x16-px tile (~3e8 atomic
+// f32 ops per 36x48 layer — measured 9 TF with one dW copy, 32 TF with
+// the 256-slab split); with grid ~1k the atomic count drops ~100x and
+// the kernel is GEMM-bound.  dw_slab: [DW_NSLAB][cout*9*cin] zeroed
+// fp32; workgroup i accumulates into slab i % DW_NSLAB (caller sums).
 // ---------------------------------------------------------------------------
-// dw_slab: [DW_NSLAB][cout*9*cin] zero-initialised fp32; workgroup i
-// accumulates into slab i % DW_NSLAB (the caller sums the slabs).  With
-// a single copy ~310k workgroups serialise on each of the ~2-5k words
-// (measured 9 TF); 256 slabs cut the per-word contention 256x.
 constexpr int DW_NSLAB = 256;
+constexpr int DW_GRID = 1024;          // resident-WG target (256 CU x 2 x2)
 
 template <int CIN, int COUT, bool RELU_IN, int TTY = TY, int TTX = TX>
 __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
     float* __restrict__ dw_slab, int N, int H, int W, int swz) {
+  (void)swz;
   float* __restrict__ dw =
       dw_slab + (size_t)(blockIdx.x % DW_NSLAB) * (COUT * 9 * CIN);
-  constexpr int MTOT = 9 * CIN;                  // 144 or 288
-  constexpr int MFRAGS = (MTOT + 15) / 16;       // 9 or 18
+  constexpr int MTOT = 9 * CIN;                  // 36..288
+  constexpr int MFRAGS = (MTOT + 15) / 16;       // 3..18
+  constexpr int MPW = (MFRAGS + 3) / 4;          // m-frags per wave
   constexpr int NFRAG = COUT / 16;
   constexpr int NPX = TTY * TTX;
   constexpr int KPX = ((NPX + 31) / 32) * 32;    // px padded to 32-mult
@@ -387,86 +392,101 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
 
   const int ntx = (W + TTX - 1) / TTX;
   const int nty = (H + TTY - 1) / TTY;
-  const int bid = swizzle_bid(swz);
-  const int tile = bid % (ntx * nty);
-  const int img = bid / (ntx * nty);
-  if (img >= N) return;
-  const int ty0 = (tile / ntx) * TTY;
-  const int tx0 = (tile % ntx) * TTX;
+  const long total = (long)N * ntx * nty;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-
-  {  // x patch with halo (+ReLU)
-    const unsigned short* xs = reinterpret_cast<const unsigned short*>(x) +
-                               (size_t)img * H * W * CIN;
-    unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
-    for (int p = tid; p < (TTY + 2) * (TTX + 2); p += THREADS) {
-      const int py = p / (TTX + 2), px = p % (TTX + 2);
-      const int iy = ty0 + py - 1, ix = tx0 + px - 1;
-      unsigned short* dst = pd + py * PSTR + px * CIN;
-      if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
-        const unsigned short* src = xs + ((size_t)iy * W + ix) * CIN;
-        #pragma unroll
-        for (int c = 0; c < CIN; ++c)
-          dst[c] = RELU_IN ? relu_bits(src[c]) : src[c];
-      } else {
-        #pragma unroll
-        for (int c = 0; c < CIN; ++c) dst[c] = 0;
-      }
-    }
-  }
-  {  // dY tile [px][cout], zero where out of range
-    const unsigned short* ds = reinterpret_cast<const unsigned short*>(dy) +
-                               (size_t)img * H * W * COUT;
-    unsigned short* dd = reinterpret_cast<unsigned short*>(s_dy);
-    for (int p = tid; p < KPX; p += THREADS) {
-      const int oy = ty0 + p / TTX, ox = tx0 + p % TTX;
-      unsigned short* dst = dd + p * COUT;
-      if (p < NPX && oy < H && ox < W) {
-        const unsigned short* src = ds + ((size_t)oy * W + ox) * COUT;
-        #pragma unroll
-        for (int c = 0; c < COUT; ++c) dst[c] = src[c];
-      } else {
-        #pragma unroll
-        for (int c = 0; c < COUT; ++c) dst[c] = 0;
-      }
-    }
-  }
-  __syncthreads();
-
-  // each wave owns m-frags wave, wave+4, wave+8, ... (MFRAGS 9/18)
   const int kbase = (lane >> 4) * 8;
-  for (int mf = wave; mf < MFRAGS; mf += 4) {
-    floatx4 acc[NFRAG];
+
+  floatx4 acc[MPW][NFRAG];
+  #pragma unroll
+  for (int m = 0; m < MPW; ++m)
     #pragma unroll
-    for (int n = 0; n < NFRAG; ++n) acc[n] = floatx4{0.f, 0.f, 0.f, 0.f};
-    const int m0 = mf * 16 + (lane & 15);        // this lane's A row
-    const int tap = m0 / CIN, c = m0 % CIN;      // may be >=9*CIN (pad)
-    const int ky = tap / 3, kx = tap % 3;
-    #pragma unroll 2
-    for (int s = 0; s < KPX / 32; ++s) {         // k-steps of 32 px
-      bf16x8 af = {}, bfv[NFRAG] = {};
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int px = s * 32 + kbase + j;
-        if (m0 < MTOT && px < NPX)
-          af[j] = s_patch[(px / TTX + ky) * PSTR + (px % TTX + kx) * CIN
-                          + c];
-      }
-      #pragma unroll
-      for (int n = 0; n < NFRAG; ++n) {
-        const int col = n * 16 + (lane & 15);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          bfv[n][j] = s_dy[(s * 32 + kbase + j) * COUT + col];
-        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfv[n], acc[n],
-                                                         0, 0, 0);
+    for (int n = 0; n < NFRAG; ++n) acc[m][n] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+  for (long t = blockIdx.x; t < total; t += gridDim.x) {
+    const int tile = (int)(t % (ntx * nty));
+    const int img = (int)(t / (ntx * nty));
+    const int ty0 = (tile / ntx) * TTY;
+    const int tx0 = (tile % ntx) * TTX;
+    __syncthreads();  // all waves done reading the previous tile's LDS
+
+    {  // x patch with halo (+ReLU)
+      const unsigned short* xs = reinterpret_cast<const unsigned short*>(x) +
+                                 (size_t)img * H * W * CIN;
+      unsigned short* pd = reinterpret_cast<unsigned short*>(s_patch);
+      for (int p = tid; p < (TTY + 2) * (TTX + 2); p += THREADS) {
+        const int py = p / (TTX + 2), px = p % (TTX + 2);
+        const int iy = ty0 + py - 1, ix = tx0 + px - 1;
+        unsigned short* dst = pd + py * PSTR + px * CIN;
+        if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+          const unsigned short* src = xs + ((size_t)iy * W + ix) * CIN;
+          #pragma unroll
+          for (int c = 0; c < CIN; ++c)
+            dst[c] = RELU_IN ? relu_bits(src[c]) : src[c];
+        } else {
+          #pragma unroll
+          for (int c = 0; c < CIN; ++c) dst[c] = 0;
+        }
       }
     }
-    // epilogue: D[m=9cin-row][n=cout] -> dw[f][ky][kx][c], atomic f32
-    const int dcol = lane & 15;
-    const int drow = (lane >> 4) * 4;
+    {  // dY tile [px][cout], zero where out of range
+      const unsigned short* ds = reinterpret_cast<const unsigned short*>(dy) +
+                                 (size_t)img * H * W * COUT;
+      unsigned short* dd = reinterpret_cast<unsigned short*>(s_dy);
+      for (int p = tid; p < KPX; p += THREADS) {
+        const int oy = ty0 + p / TTX, ox = tx0 + p % TTX;
+        unsigned short* dst = dd + p * COUT;
+        if (p < NPX && oy < H && ox < W) {
+          const unsigned short* src = ds + ((size_t)oy * W + ox) * COUT;
+          #pragma unroll
+          for (int c = 0; c < COUT; ++c) dst[c] = src[c];
+        } else {
+          #pragma unroll
+          for (int c = 0; c < COUT; ++c) dst[c] = 0;
+        }
+      }
+    }
+    __syncthreads();
+
+    // each wave owns m-frags wave, wave+4, wave+8, ...
+    #pragma unroll
+    for (int mi = 0; mi < MPW; ++mi) {
+      const int mf = wave + mi * 4;
+      if (mf >= MFRAGS) break;
+      const int m0 = mf * 16 + (lane & 15);      // this lane's A row
+      const int tap = m0 / CIN, c = m0 % CIN;    // may be >=9*CIN (pad)
+      const int ky = tap / 3, kx = tap % 3;
+      #pragma unroll 2
+      for (int ks = 0; ks < KPX / 32; ++ks) {    // k-steps of 32 px
+        bf16x8 af = {}, bfv[NFRAG] = {};
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int px = ks * 32 + kbase + j;
+          if (m0 < MTOT && px < NPX)
+            af[j] = s_patch[(px / TTX + ky) * PSTR + (px % TTX + kx) * CIN
+                            + c];
+        }
+        #pragma unroll
+        for (int n = 0; n < NFRAG; ++n) {
+          const int col = n * 16 + (lane & 15);
+          #pragma unroll
+          for (int j = 0; j < 8; ++j)
+            bfv[n][j] = s_dy[(ks * 32 + kbase + j) * COUT + col];
+          acc[mi][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bfv[n], acc[mi][n], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  // epilogue once per WG: D[m=9cin-row][n=cout] -> dw[f][ky][kx][c]
+  const int dcol = lane & 15;
+  const int drow = (lane >> 4) * 4;
+  #pragma unroll
+  for (int mi = 0; mi < MPW; ++mi) {
+    const int mf = wave + mi * 4;
+    if (mf >= MFRAGS) break;
     #pragma unroll
     for (int n = 0; n < NFRAG; ++n) {
       const int f = n * 16 + dcol;
@@ -474,9 +494,9 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
       for (int reg = 0; reg < 4; ++reg) {
         const int m = mf * 16 + drow + reg;
         if (m < MTOT) {
-          const int t = m / CIN, cc = m % CIN;
-          atomicAdd(&dw[(((size_t)f * 3 + t / 3) * 3 + t % 3) * CIN + cc],
-                    acc[n][reg]);
+          const int t2 = m / CIN, cc = m % CIN;
+          atomicAdd(&dw[(((size_t)f * 3 + t2 / 3) * 3 + t2 % 3) * CIN + cc],
+                    acc[mi][n][reg]);
         }
       }
     }
@@ -785,7 +805,10 @@ extern "C" int launch_deep_conv3x3_dw(const void* x, const void* dy,
   const bf16_t* xb = static_cast<const bf16_t*>(x);
   const bf16_t* db = static_cast<const bf16_t*>(dy);
   const int tid_ = tile_id(H, W);
-  dim3 grid(conv_grid_t(N, H, W, tid_)), blk(THREADS);
+  int g = conv_grid_t(N, H, W, tid_);
+  const char* legacy = getenv("IMPALA_DEEP_DW_LEGACY");
+  if (!(legacy != nullptr && legacy[0] == '1') && g > DW_GRID) g = DW_GRID;
+  dim3 grid(g), blk(THREADS);
 #define DW_CASE_T(CI, CO, TTY, TTX)                                         \
   if (relu_in)                                                              \
     hipLaunchKernelGGL((conv3x3_dw_nhwc<CI, CO, true, TTY, TTX>), grid,     \

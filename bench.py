@@ -72,6 +72,7 @@ def main():
         compute_dtype=args.compute_dtype,
         device=args.device,
         weight_sync_interval=1,
+        ramp_ready=True,
         seed=1234,
     )
 

@@ -264,20 +264,34 @@ class Rollout:
         return {"policy_logits": logits.unsqueeze(0),
                 "action": action.unsqueeze(0)}, ()
 
-    def _advance(self):
+    def _advance(self, out=None, t: int = 0):
         """One timestep: forward -> record row -> step envs.
-        Sets self.carry to the recorded row."""
-        out, new_core = (self._fast_forward() if self.fast is not None
+        out=None: sets self.carry to the recorded row (torch tensors).
+        out=dict of NUMPY views (key -> (T+1, n, ...)): writes row t
+        directly — one np slice assignment per key instead of a torch
+        copy_ dispatch (the per-claim overhead of the torch path was
+        measured at 5.2 ms vs 10.7 ms of compute, 33%)."""
+        fwd, new_core = (self._fast_forward() if self.fast is not None
                          else self._forward())
-        action = out["action"][0]
-        self.carry = dict(
-            frame=self.frame, reward=self.reward.clone(),
-            done=self.done.clone(), action=action,
-            behavior_logits=out["policy_logits"][0],
-            last_action=self.last_action,  # the action INPUT to this forward
-            episode_return=torch.from_numpy(self.episode_return.copy()),
-            episode_step=torch.from_numpy(self.episode_step.copy()),
-        )
+        action = fwd["action"][0]
+        if out is not None:
+            out["frame"][t] = self.frame.numpy()
+            out["reward"][t] = self.reward.numpy()
+            out["done"][t] = self.done.numpy()
+            out["action"][t] = action.numpy()
+            out["behavior_logits"][t] = fwd["policy_logits"][0].numpy()
+            out["last_action"][t] = self.last_action.numpy()
+            out["episode_return"][t] = self.episode_return
+            out["episode_step"][t] = self.episode_step
+        else:
+            self.carry = dict(
+                frame=self.frame, reward=self.reward.clone(),
+                done=self.done.clone(), action=action,
+                behavior_logits=fwd["policy_logits"][0],
+                last_action=self.last_action,  # action INPUT to this forward
+                episode_return=torch.from_numpy(self.episode_return.copy()),
+                episode_step=torch.from_numpy(self.episode_step.copy()),
+            )
         finished = self.done
         if finished.any():  # reset stats after the done row was recorded
             fin = finished.numpy()
@@ -318,18 +332,23 @@ class Rollout:
         self.done = torch.from_numpy(done)
 
     def fill_slot(self, slot, T: int):
-        """Write rows 0..T (shape (T+1, n, ...)) of one slot."""
+        """Write rows 0..T (shape (T+1, n, ...)) of one slot, directly
+        into the shm arenas through numpy views (single write pass)."""
+        np_out = {k: slot[k].numpy() for k in ROW_KEYS}
         for k in ROW_KEYS:
-            slot[k][0].copy_(self.carry[k])
+            src = self.carry[k]
+            np_out[k][0] = src.numpy() if isinstance(src, torch.Tensor) else src
         if "initial_core_state" in slot and self.core_before:
             # (2, n, hidden)
             slot["initial_core_state"].copy_(torch.stack(self.core_before))
         for t in range(1, T + 1):
             if t == T:
                 self.core_before = tuple(x.clone() for x in self.core_state)
-            self._advance()
-            for k in ROW_KEYS:
-                slot[k][t].copy_(self.carry[k])
+            self._advance(out=np_out, t=t)
+        # next unroll's row 0 = this unroll's row T; the slot is about to
+        # be handed off (and recycled), so carry owns a copy
+        self.carry = {k: torch.from_numpy(np_out[k][T].copy())
+                      for k in ROW_KEYS}
 
 
 def act(flags, actor_index: int, free_queue, full_queue, buffers, mirror,

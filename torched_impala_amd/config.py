@@ -50,6 +50,10 @@ class Flags:
     baseline_cost: float = 0.5
     discounting: float = 0.99
     reward_clipping: str = "abs_one"   # abs_one | none
+    # benchmarking: block the learner until the rollout ring is fully
+    # populated (every actor warmed through >=1 claim) so short timed
+    # windows measure steady state, not the actor spawn ramp
+    ramp_ready: bool = False
     clip_rho_threshold: float = 1.0
     clip_c_threshold: float = 1.0
     # rho-bar for the PG advantage; None => same as clip_rho_threshold

@@ -51,9 +51,10 @@ void launch_shallow_encoder_bwd(const unsigned char*, const float*,
                                 const float*, const float*, int, float*,
                                 float*, float*, hipStream_t);
 int shallow_encoder_slab_elems();
+int linear_splitk_factor(int, int, int);
 void launch_linear_mfma(const float*, const float*, const float*,
-                        const float*, int, int, int, int, float*,
-                        hipStream_t);
+                        const float*, int, int, int, int, float*, float*,
+                        int, hipStream_t);
 int launch_deep_conv3x3(const void*, const void*, const float*, const void*,
                         void*, int, int, int, int, int, int, hipStream_t);
 void launch_deep_maxpool(const void*, void*, void*, int, int, int, int,
@@ -408,9 +409,16 @@ torch::Tensor linear_mfma(torch::Tensor A, torch::Tensor B,
     check_f32(*ymask, "ymask");
     ymask_p = ymask->data_ptr<float>();
   }
+  const int S = linear_splitk_factor(M, N, K);
+  torch::Tensor ws;
+  float* ws_p = nullptr;
+  if (S > 1) {
+    ws = torch::empty({(int64_t)S, M, N}, A.options());
+    ws_p = ws.data_ptr<float>();
+  }
   launch_linear_mfma(A.data_ptr<float>(), B.data_ptr<float>(), bias_p,
                      ymask_p, M, N, K, (int)mode, C.data_ptr<float>(),
-                     current_stream());
+                     ws_p, S, current_stream());
   HIP_CHECK(hipGetLastError());
   return C;
 }

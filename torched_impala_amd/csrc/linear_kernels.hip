@@ -29,15 +29,24 @@ constexpr int LDS_STRIDE = BK + 4;  // pad: avoids b32 column conflicts
 constexpr int THREADS = 256;
 
 // mode: 0 = plain, 1 = relu epilogue, 2 = A-staging masked by Ymask>0
+// Split-K (round 2): the headline GEMMs are tiny (fc fwd 640x256x1568 =
+// a 10x4 grid, 84% of the 256 CUs idle).  With gridDim.z = S > 1 each
+// z-block covers K-chunk [z*kchunk, +kchunk) and stores RAW partials to
+// P = C + z*M*N (C then sized S*M*N); linear_splitk_reduce applies the
+// deterministic S-way sum + bias + activation.  gridDim.z == 1 keeps
+// the original fused epilogue.
 __global__ void __launch_bounds__(THREADS, 2)
 linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
             const float* __restrict__ bias, const float* __restrict__ Ymask,
-            int M, int N, int K, int mode, float* __restrict__ C) {
+            int M, int N, int K, int mode, int kchunk,
+            float* __restrict__ C) {
   __shared__ __align__(16) float s_a[2][BM * LDS_STRIDE];
   __shared__ __align__(16) float s_b[2][BN * LDS_STRIDE];
 
   const int bm = blockIdx.x * BM;
   const int bn = blockIdx.y * BN;
+  const int kb0 = blockIdx.z * kchunk;
+  const int kend = min(K, kb0 + kchunk);   // this z-block's K range
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;           // 4 waves
@@ -58,7 +67,7 @@ linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
       float4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
       if (gr < M) {
         const float* src = A + (size_t)gr * K + k0 + skoff;
-        const int krem = K - (k0 + skoff);
+        const int krem = kend - (k0 + skoff);
         if (krem >= 8) {
           v0 = *reinterpret_cast<const float4*>(src);
           v1 = *reinterpret_cast<const float4*>(src + 4);
@@ -68,7 +77,7 @@ linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
         }
         if (mode == 2) {  // fused ReLU-backward mask from Y
           const float* ym = Ymask + (size_t)gr * K + k0 + skoff;
-          const int mrem = K - (k0 + skoff);
+          const int mrem = kend - (k0 + skoff);
           for (int i = 0; i < 4; ++i) {
             if (i < mrem && ym[i] <= 0.f) (&v0.x)[i] = 0.f;
             if (i + 4 < mrem && ym[i + 4] <= 0.f) (&v1.x)[i] = 0.f;
@@ -85,7 +94,7 @@ linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
       float4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
       if (gr < N) {
         const float* src = B + (size_t)gr * K + k0 + skoff;
-        const int krem = K - (k0 + skoff);
+        const int krem = kend - (k0 + skoff);
         if (krem >= 8) {
           v0 = *reinterpret_cast<const float4*>(src);
           v1 = *reinterpret_cast<const float4*>(src + 4);
@@ -100,13 +109,13 @@ linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
     }
   };
 
-  stage(0, 0);
+  stage(0, kb0);
   __syncthreads();
 
-  const int nk = (K + BK - 1) / BK;
+  const int nk = (kend - kb0 + BK - 1) / BK;
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
-    if (kt + 1 < nk) stage(cur ^ 1, (kt + 1) * BK);
+    if (kt + 1 < nk) stage(cur ^ 1, kb0 + (kt + 1) * BK);
     // MFMA over this K tile: 16x16x4 fragments
     const int ar = lane & 15;          // fragment row
     const int ak = lane >> 4;          // k sub-index (0..3)
@@ -128,6 +137,8 @@ linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
   // row=(lane>>4)*4+reg (dtype-independent on gfx950)
   const int ccol = lane & 15;
   const int crow = (lane >> 4) * 4;
+  float* __restrict__ out = C + (size_t)blockIdx.z * M * N;
+  const bool split = gridDim.z > 1;
   #pragma unroll
   for (int r = 0; r < 2; ++r) {
     #pragma unroll
@@ -138,12 +149,29 @@ linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
         const int col = bn + wn + c * 16 + ccol;
         if (row < M && col < N) {
           float v = acc[r][c][reg];
-          if (bias != nullptr) v += bias[col];
-          if (mode == 1) v = v > 0.f ? v : 0.f;
-          C[(size_t)row * N + col] = v;
+          if (!split) {
+            if (bias != nullptr) v += bias[col];
+            if (mode == 1) v = v > 0.f ? v : 0.f;
+          }
+          out[(size_t)row * N + col] = v;
         }
       }
     }
+  }
+}
+
+// deterministic S-way partial sum + bias + activation epilogue
+__global__ void __launch_bounds__(256)
+linear_splitk_reduce(const float* __restrict__ P,
+                     const float* __restrict__ bias, int S, int N,
+                     size_t MN, int mode, float* __restrict__ C) {
+  for (size_t e = (size_t)blockIdx.x * blockDim.x + threadIdx.x; e < MN;
+       e += (size_t)gridDim.x * blockDim.x) {
+    float v = 0.f;
+    for (int s = 0; s < S; ++s) v += P[(size_t)s * MN + e];
+    if (bias != nullptr) v += bias[(int)(e % N)];
+    if (mode == 1) v = v > 0.f ? v : 0.f;
+    C[e] = v;
   }
 }
 
@@ -151,13 +179,37 @@ linear_mfma(const float* __restrict__ A, const float* __restrict__ B,
 
 extern "C" {
 
+// S chosen so the GEMM grid reaches ~2 blocks/CU on the 256-CU chip
+// while each K-chunk stays >= 2 LDS stages deep.
+int linear_splitk_factor(int M, int N, int K) {
+  const int gm = (M + impala::BM - 1) / impala::BM;
+  const int gn = (N + impala::BN - 1) / impala::BN;
+  int s = (512 + gm * gn - 1) / (gm * gn);
+  const int smax = K / (2 * impala::BK) > 0 ? K / (2 * impala::BK) : 1;
+  if (s > smax) s = smax;
+  if (s > 16) s = 16;
+  if (s < 1) s = 1;
+  return s;
+}
+
+// ws: workspace of S*M*N floats when S = linear_splitk_factor(...) > 1
+// (may be null for S == 1).
 void launch_linear_mfma(const float* A, const float* B, const float* bias,
                         const float* ymask, int M, int N, int K, int mode,
-                        float* C, hipStream_t stream) {
+                        float* C, float* ws, int S, hipStream_t stream) {
   dim3 grid((M + impala::BM - 1) / impala::BM,
-            (N + impala::BN - 1) / impala::BN);
+            (N + impala::BN - 1) / impala::BN, S);
+  const int kchunk = S > 1 ? (K + S - 1) / S : K;
   hipLaunchKernelGGL(impala::linear_mfma, grid, dim3(impala::THREADS), 0,
-                     stream, A, B, bias, ymask, M, N, K, mode, C);
+                     stream, A, B, bias, ymask, M, N, K, mode, kchunk,
+                     S > 1 ? ws : C);
+  if (S > 1) {
+    const size_t MN = (size_t)M * N;
+    int blocks = (int)((MN + 255) / 256);
+    if (blocks > 2048) blocks = 2048;
+    hipLaunchKernelGGL(impala::linear_splitk_reduce, dim3(blocks), dim3(256),
+                       0, stream, ws, bias, S, N, MN, mode, C);
+  }
 }
 
 }  // extern "C"

@@ -480,6 +480,26 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
             if resume_frames:
                 mirror.publish_state_dict(model.state_dict())
 
+    if getattr(flags, "ramp_ready", False):
+        # wait until the ring is full: every actor has finished at least
+        # one claim (imports, weight packing, jit all warmed) before the
+        # first learner step — benches then measure steady state even
+        # over short windows (73 ms timed region on a 7 s
+        # process was ramp-dominated)
+        deadline = time.time() + 600.0
+        target = buffers.num_slots
+        while (full_queue.size_approx() < target
+               and time.time() < deadline):
+            for i, pr in enumerate(procs):
+                if not pr.is_alive():
+                    raise RuntimeError(
+                        f"actor {i} died during ramp (exit {pr.exitcode})")
+            time.sleep(0.02)
+        log.info("ramp ready: %d/%d slots full", full_queue.size_approx(),
+                 target)
+        if world > 1:
+            torch.distributed.barrier()
+
     frames_per_batch = flags.unroll_length * flags.batch_size
     step_frames = resume_frames
     learner_steps = 0

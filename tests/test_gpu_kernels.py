@@ -298,6 +298,26 @@ def test_linear_mfma_plain_mode():
     torch.testing.assert_close(C, A @ B.t(), rtol=1e-4, atol=1e-4)
 
 
+@pytest.mark.parametrize("M,N,K,mode", [(640, 256, 1568, 1),
+                                        (640, 1568, 256, 0),
+                                        (333, 257, 1000, 1)])
+def test_linear_mfma_splitk_parity(M, N, K, mode):
+    """Shapes whose grids under-fill the 256-CU chip take the split-K
+    path (S > 1 via linear_splitk_factor): partials + deterministic
+    reduce must match the fused single-kernel math incl. bias/ReLU."""
+    from torched_impala_amd import ops
+    dev = _dev()
+    torch.manual_seed(2)
+    A = torch.randn(M, K, device=dev)
+    B = torch.randn(N, K, device=dev)
+    bias = torch.randn(N, device=dev)
+    C = ops.ext_hip().linear_mfma(A, B, bias, None, mode)
+    ref = A @ B.t() + bias
+    if mode == 1:
+        ref = torch.relu(ref)
+    torch.testing.assert_close(C, ref, rtol=2e-4, atol=2e-3)
+
+
 @pytest.mark.parametrize("clip_rho,clip_c,lam", [(0.8, 1.2, 0.95),
                                                  (2.0, 0.7, 0.5)])
 def test_vtrace_scan_parity_nondefault_clips(clip_rho, clip_c, lam):

@@ -37,6 +37,9 @@ def main():
     p.add_argument("--num-actions", type=int, default=6)
     p.add_argument("--compute-dtype", default="fp32", choices=["fp32", "bf16"])
     p.add_argument("--channels-last", action="store_true")
+    p.add_argument("--graphs", action="store_true",
+                   help="hipGraph-captured step (the product path) "
+                        "instead of eager learn_step")
     args = p.parse_args()
 
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
@@ -70,13 +73,40 @@ def main():
     if args.use_lstm:
         batch["initial_core_state"] = torch.zeros(2, B, 256, device=device)
 
+    if args.graphs and device.type == "cuda":
+        from torched_impala_amd.learner import GraphedStep
+
+        st = {}
+        for kk, v in batch.items():
+            if kk == "initial_core_state":       # (2,B,H) -> (B,2,1,H)
+                st[kk] = v.permute(1, 0, 2).unsqueeze(2).contiguous()
+            else:                                # (T+1,B,...) -> (B,T+1,1,...)
+                st[kk] = v.transpose(0, 1).unsqueeze(2).contiguous()
+
+        class _Stub:
+            def __init__(self, flags, st):
+                self.flags = flags
+                self.k = flags.batch_size
+                self.staging = {0: st}
+            from torched_impala_amd.learner import Batcher as _B
+            _to_time_major = _B._to_time_major
+            del _B
+
+        graphed = GraphedStep(flags, model, optimizer, _Stub(flags, st))
+
+        def one_step():
+            graphed.step(0, scheduler)
+    else:
+        def one_step():
+            learn_step(flags, model, dict(batch), optimizer, scheduler, None)
+
     for _ in range(args.warmup):
-        learn_step(flags, model, dict(batch), optimizer, scheduler, None)
+        one_step()
     if device.type == "cuda":
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        learn_step(flags, model, dict(batch), optimizer, scheduler, None)
+        one_step()
     if device.type == "cuda":
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
@@ -87,7 +117,7 @@ def main():
     print(json.dumps(dict(
         diagnostic="learner_capacity", model=args.model, lstm=args.use_lstm,
         hip=not args.no_hip_kernels, dtype=args.compute_dtype,
-        channels_last=args.channels_last, T=T, B=B,
+        channels_last=args.channels_last, graphs=bool(args.graphs), T=T, B=B,
         ms_per_step=round(ms, 3), frames_per_sec=round(fps, 1),
         peak_hbm_gib=round(mem_gb, 2))))
 

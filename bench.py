@@ -147,6 +147,7 @@ def main():
                 "seq_len": flags.unroll_length,
                 "parallelism": f"dp{n_gpus}",
                 "num_actors_per_gpu": flags.num_actors,
+                "host_cpus": os.cpu_count(),
                 "actor_jit_requested": flags.actor_jit,
                 "actor_jit": achieved_tier,
                 "graph_path_active": graph_active,

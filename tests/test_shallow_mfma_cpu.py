@@ -168,8 +168,13 @@ def test_mock_features_with_fc_grads(mock_ext):
     assert cos > 0.97, cos
 
 
-def test_flag_gate_defaults_off(monkeypatch):
+def test_flag_gate_defaults_on(monkeypatch):
+    """Default flipped ON in round 2 after hardware parity + the
+    875-vs-1218 us/step attribution (profiles/r4_mfma_attribution.md);
+    =0 is the A/B escape hatch."""
     monkeypatch.delenv("IMPALA_SHALLOW_MFMA", raising=False)
+    assert S.shallow_mfma_enabled()
+    monkeypatch.setenv("IMPALA_SHALLOW_MFMA", "0")
     assert not S.shallow_mfma_enabled()
     monkeypatch.setenv("IMPALA_SHALLOW_MFMA", "1")
     assert S.shallow_mfma_enabled()

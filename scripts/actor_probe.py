@@ -16,8 +16,14 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
-def _worker(args, idx, q):
+def _worker(args, idx, q, barrier=None):
     import torch
+    if args.pin:
+        try:
+            ncpu = os.cpu_count() or 1
+            os.sched_setaffinity(0, {idx % ncpu})
+        except OSError:
+            pass
     torch.set_num_threads(max(1, args.threads))
     from torched_impala_amd.config import Flags
     from torched_impala_amd.models import build_model
@@ -29,6 +35,8 @@ def _worker(args, idx, q):
     roll = Rollout(flags, idx, model)
     for _ in range(5):
         roll._advance()
+    if barrier is not None:
+        barrier.wait(timeout=300)  # synchronized windows: honest overlap
     n = 0
     t0 = time.perf_counter()
     while time.perf_counter() - t0 < args.seconds:
@@ -46,11 +54,15 @@ def main():
     p.add_argument("--threads", type=int, default=1)
     p.add_argument("--jit", default="native-bf16",
                    choices=["off", "safe", "frozen", "native", "native-bf16"])
+    p.add_argument("--pin", action="store_true",
+                   help="pin worker i to cpu i (distinct physical cores)")
+    p.add_argument("--no-barrier", action="store_true")
     args = p.parse_args()
     import torch.multiprocessing as mp
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(args, i, q))
+    barrier = None if args.no_barrier else ctx.Barrier(args.actors)
+    procs = [ctx.Process(target=_worker, args=(args, i, q, barrier))
              for i in range(args.actors)]
     for pr in procs:
         pr.start()

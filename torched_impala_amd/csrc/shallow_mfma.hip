@@ -65,14 +65,32 @@ __global__ __launch_bounds__(256, 2) void conv1_mfma(
     for (int i = tid; i < F1 * 256; i += 256) s_w[i] = ws[i];
   }
   {
+    // row-wise staging: one div per ROW (not per element), uchar4 global
+    // loads + bf16x4 LDS stores (84 % 4 == 0 keeps both aligned).  The
+    // original per-element i%/i/ staging was the kernel's hot phase
+    // (~89 idiv+store iterations per thread).
     const unsigned char* fr = frames + (size_t)img * C0 * H0 * W0;
-    for (int i = tid; i < C0 * PY * PX; i += 256) {
-      const int x = i % PX;
-      const int y = (i / PX) % PY;
-      const int c = i / (PX * PY);
+    for (int cy = tid; cy < C0 * PY; cy += 256) {
+      const int c = cy / PY, y = cy - (cy / PY) * PY;
       const int iy = iy0 + y;
-      s_p[i] = iy < H0 ? f2bf(fr[(c * H0 + iy) * W0 + x] * (1.f / 255.f))
-                       : f2bf(0.f);
+      bf16_t* dst = s_p + (size_t)cy * PX;
+      if (iy < H0) {
+        const unsigned char* src = fr + ((size_t)c * H0 + iy) * W0;
+        #pragma unroll
+        for (int x = 0; x < PX; x += 4) {
+          const uchar4 v = *reinterpret_cast<const uchar4*>(src + x);
+          bf16x4 o;
+          o[0] = f2bf(v.x * (1.f / 255.f));
+          o[1] = f2bf(v.y * (1.f / 255.f));
+          o[2] = f2bf(v.z * (1.f / 255.f));
+          o[3] = f2bf(v.w * (1.f / 255.f));
+          *reinterpret_cast<bf16x4*>(dst + x) = o;
+        }
+      } else {
+        #pragma unroll
+        for (int x = 0; x < PX; x += 4)
+          *reinterpret_cast<bf16x4*>(dst + x) = bf16x4{};
+      }
     }
   }
   __syncthreads();

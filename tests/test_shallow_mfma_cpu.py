@@ -69,11 +69,15 @@ class _MockExt:
         dw2 = torch.nn.grad.conv2d_weight(
             a1n, (32, 16, 4, 4), _q(da2), stride=2)
         dw3 = torch.nn.grad.conv2d_weight(a2n, (32, 32, 3, 3), dz3n)
-        return [da1.permute(0, 2, 3, 1).to(BF).contiguous(),
-                da2.permute(0, 2, 3, 1).to(BF).contiguous(),
-                dw1.permute(1, 2, 3, 0).reshape(256, 16),
-                dw2.permute(1, 2, 3, 0).reshape(256, 32),
-                dw3.permute(2, 3, 1, 0).reshape(9, 32, 32).reshape(288, 32)]
+        da1b = da1.permute(0, 2, 3, 1).to(BF).contiguous()
+        da2b = da2.permute(0, 2, 3, 1).to(BF).contiguous()
+        # round-2 contract: torch-layout dW + fused bias column sums
+        # (bf16-rounded da*, matching the device bwd_finalize inputs)
+        db1 = da1b.float().sum(dim=(0, 1, 2))
+        db2 = da2b.float().sum(dim=(0, 1, 2))
+        db3 = dz3n.sum(dim=(0, 2, 3))
+        return [da1b, da2b, dw1.contiguous(), dw2.contiguous(),
+                dw3.contiguous(), db1, db2, db3]
 
     @staticmethod
     def _w2_from_wr2(wr2):

@@ -39,23 +39,23 @@ constexpr int F3 = 32, K3 = 3, H3 = 7, W3 = 7;
 
 // ---------------------------------------------------------------------------
 // conv1: 4x84x84 u8 -> 16x20x20, K=8 S=4, fused /255 + bias + ReLU.
-// Grid.x = N * 2 (y-halves of 16/4 output rows). M-tile = 16x20 = 320 px
-// (20 m-frags, 5 per wave), N = 16 (one frag), K = 256 (8 steps).
-// Patch: planar LDS [4][68][84]; A-frag = patch[c][y0+ky][4*ox + 0..7],
-// an 8-half run at a 4-half-aligned offset -> two b64 reads.
+// ONE workgroup per image (round 2; was 2 y-halves): the full 4x84x84
+// image is staged once (55.1 KiB LDS, 2 blocks/CU) instead of two
+// 68-row halves (136 rows), and the weights once instead of twice —
+// PMC showed the kernel 70% wave-parked on staging.  M-tile = 400 px
+// (25 m-frags round-robin over 4 waves), N = 16 (one frag), K = 256
+// (8 steps).  A-frag = patch[c][4*oy+ky][4*ox + 0..7], an 8-half run
+// at a 4-half-aligned offset -> two b64 reads.
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256, 2) void conv1_mfma(
     const unsigned char* __restrict__ frames, const bf16_t* __restrict__ w,
     const float* __restrict__ b, bf16_t* __restrict__ a1, int N) {
-  constexpr int PY = 68, PX = 84;           // patch rows / cols (halfs)
-  __shared__ bf16_t s_p[C0 * PY * PX];      // 44.6 KiB
+  constexpr int PY = 84, PX = 84;           // full image (halfs)
+  __shared__ bf16_t s_p[C0 * PY * PX];      // 55.1 KiB
   __shared__ bf16_t s_w[F1 * 256];          // 8 KiB
 
-  const int img = blockIdx.x >> 1;
-  const int half = blockIdx.x & 1;          // 0: out rows 0..15, 1: 16..19
+  const int img = blockIdx.x;
   if (img >= N) return;
-  const int oy0 = half * 16;
-  const int iy0 = oy0 * S1;                 // first input row needed
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -65,39 +65,31 @@ __global__ __launch_bounds__(256, 2) void conv1_mfma(
     for (int i = tid; i < F1 * 256; i += 256) s_w[i] = ws[i];
   }
   {
-    // row-wise staging: one div per ROW (not per element), uchar4 global
-    // loads + bf16x4 LDS stores (84 % 4 == 0 keeps both aligned).  The
-    // original per-element i%/i/ staging was the kernel's hot phase
-    // (~89 idiv+store iterations per thread).
+    // row-wise staging: one div per ROW, uchar4 global loads + bf16x4
+    // LDS stores (84 % 4 == 0 keeps both aligned)
     const unsigned char* fr = frames + (size_t)img * C0 * H0 * W0;
     for (int cy = tid; cy < C0 * PY; cy += 256) {
-      const int c = cy / PY, y = cy - (cy / PY) * PY;
-      const int iy = iy0 + y;
       bf16_t* dst = s_p + (size_t)cy * PX;
-      if (iy < H0) {
-        const unsigned char* src = fr + ((size_t)c * H0 + iy) * W0;
-        #pragma unroll
-        for (int x = 0; x < PX; x += 4) {
-          const uchar4 v = *reinterpret_cast<const uchar4*>(src + x);
-          bf16x4 o;
-          o[0] = f2bf(v.x * (1.f / 255.f));
-          o[1] = f2bf(v.y * (1.f / 255.f));
-          o[2] = f2bf(v.z * (1.f / 255.f));
-          o[3] = f2bf(v.w * (1.f / 255.f));
-          *reinterpret_cast<bf16x4*>(dst + x) = o;
-        }
-      } else {
-        #pragma unroll
-        for (int x = 0; x < PX; x += 4)
-          *reinterpret_cast<bf16x4*>(dst + x) = bf16x4{};
+      const unsigned char* src = fr + (size_t)cy * W0;
+      #pragma unroll
+      for (int x = 0; x < PX; x += 4) {
+        const uchar4 v = *reinterpret_cast<const uchar4*>(src + x);
+        bf16x4 o;
+        o[0] = f2bf(v.x * (1.f / 255.f));
+        o[1] = f2bf(v.y * (1.f / 255.f));
+        o[2] = f2bf(v.z * (1.f / 255.f));
+        o[3] = f2bf(v.w * (1.f / 255.f));
+        *reinterpret_cast<bf16x4*>(dst + x) = o;
       }
     }
   }
   __syncthreads();
 
-  floatx4 acc[5];
+  constexpr int MFRAGS = 25;                 // 400 output px
+  constexpr int MPW = 7;                     // ceil(25/4) per wave
+  floatx4 acc[MPW];
   #pragma unroll
-  for (int m = 0; m < 5; ++m) acc[m] = floatx4{0.f, 0.f, 0.f, 0.f};
+  for (int m = 0; m < MPW; ++m) acc[m] = floatx4{0.f, 0.f, 0.f, 0.f};
   const int arow = lane & 15;
   const int q = lane >> 4;                   // k-quarter
 
@@ -109,9 +101,11 @@ __global__ __launch_bounds__(256, 2) void conv1_mfma(
     const bf16x8 bf = *reinterpret_cast<const bf16x8*>(
         s_w + (lane & 15) * 256 + s * 32 + q * 8);
     #pragma unroll
-    for (int m = 0; m < 5; ++m) {
-      const int p = (wave * 5 + m) * 16 + arow;   // px in 16x20 tile
-      const int oy = p / W1, ox = p % W1;         // oy 0..15
+    for (int m = 0; m < MPW; ++m) {
+      const int mf = wave + m * 4;
+      if (mf >= MFRAGS) break;
+      const int p = mf * 16 + arow;               // px in 20x20 out
+      const int oy = p / W1, ox = p % W1;
       const bf16_t* src =
           s_p + (c * PY + oy * S1 + ky) * PX + ox * S1;
       bf16x8 af;
@@ -127,17 +121,17 @@ __global__ __launch_bounds__(256, 2) void conv1_mfma(
   const int ccol = lane & 15;
   const int crow = (lane >> 4) * 4;
   #pragma unroll
-  for (int m = 0; m < 5; ++m) {
+  for (int m = 0; m < MPW; ++m) {
+    const int mf = wave + m * 4;
+    if (mf >= MFRAGS) break;
     const float bias = b[ccol];
     #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
-      const int p = (wave * 5 + m) * 16 + crow + reg;
-      const int oy = oy0 + p / W1, ox = p % W1;
-      if (oy < H1) {
-        float v = acc[m][reg] + bias;
-        v = v > 0.f ? v : 0.f;
-        a1[(((size_t)img * H1 + oy) * W1 + ox) * F1 + ccol] = f2bf(v);
-      }
+      const int p = mf * 16 + crow + reg;         // < 400 always
+      const int oy = p / W1, ox = p % W1;
+      float v = acc[m][reg] + bias;
+      v = v > 0.f ? v : 0.f;
+      a1[(((size_t)img * H1 + oy) * W1 + ox) * F1 + ccol] = f2bf(v);
     }
   }
 }
@@ -304,7 +298,7 @@ extern "C" void launch_shallow_mfma_fwd(const unsigned char* frames,
                                         const void* w3, const float* b3,
                                         void* a1, void* a2, void* a3, int N,
                                         hipStream_t s) {
-  hipLaunchKernelGGL(conv1_mfma, dim3(N * 2), dim3(256), 0, s, frames,
+  hipLaunchKernelGGL(conv1_mfma, dim3(N), dim3(256), 0, s, frames,
                      static_cast<const bf16_t*>(w1), b1,
                      static_cast<bf16_t*>(a1), N);
   hipLaunchKernelGGL(conv2_mfma, dim3(N), dim3(256), 0, s,

@@ -1,9 +1,9 @@
 """Shallow encoder on bf16 MFMA matrix cores (learner side).
 
-Autograd glue over csrc/shallow_mfma.hip + shallow_mfma_bwd.hip.  The
-fp32 LDS-fused encoder (ops/encoder.py) stays the default; this path is
-enabled with IMPALA_SHALLOW_MFMA=1 after the round-2 probe validation
-(deep_probe sections 8-9) and perf check — see ROADMAP.md.
+Autograd glue over csrc/shallow_mfma.hip + shallow_mfma_bwd.hip.
+DEFAULT path since round 2 (hardware parity + perf evidence in
+profiles/r4_mfma_attribution.md); IMPALA_SHALLOW_MFMA=0 restores the
+fp32 LDS-fused encoder (ops/encoder.py) for A/B.
 
 Weight packings (pure torch reshapes, done per forward; ~20 KiB total):
   w1p [16][256]  = conv1.weight.flatten(1)            k = c*64+ky*8+kx
@@ -13,8 +13,9 @@ Backward repacks:
   wr3 [32][288]  = conv3.weight.flip(2,3).permute(1,2,3,0).flatten(1)
   wr2 [4][16][128]: class (py,px): w2[:, :, py::2, px::2]
                     .permute(1,2,3,0).flatten(1)      k = (kyi*2+kxi)*32+f
-dW returns arrive k-major [M][F] and are reshaped back to the torch
-[F][C][K][K] layouts here.
+dW/db returns arrive already in the torch layouts: the fused
+bwd_finalize kernel does the slab sums, layout transposes and bias
+column sums in one launch (round 2).
 """
 from __future__ import annotations
 
@@ -63,17 +64,10 @@ class _ShallowMfmaEncode(torch.autograd.Function):
                 cls.append(w2.detach()[:, :, py::2, px::2]
                            .permute(1, 2, 3, 0).reshape(16, 128))
         wr2 = torch.stack(cls).to(bf).contiguous()
-        da1, da2, dw1k, dw2k, dw3k = ext.shallow_mfma_bwd(
+        (da1, da2, dw1, dw2, dw3, db1, db2, db3) = ext.shallow_mfma_bwd(
             frames.contiguous(), a1, a2, dz3, wr3, wr2)
-        # k-major -> torch layouts
-        dw1 = dw1k.view(4, 8, 8, 16).permute(3, 0, 1, 2).contiguous()
-        dw2 = dw2k.view(16, 4, 4, 32).permute(3, 0, 1, 2).contiguous()
-        dw3 = dw3k.view(9, 32, 32).permute(2, 1, 0).reshape(32, 32, 3, 3
-                                                            ).contiguous()
-        n = frames.shape[0]
-        db1 = da1.float().sum(dim=(0, 1, 2))
-        db2 = da2.float().sum(dim=(0, 1, 2))
-        db3 = dz3.float().view(n, 32, 49).sum(dim=(0, 2))
+        # weight grads arrive already in torch layouts and the bias
+        # column sums already reduced (fused bwd_finalize kernel)
         return None, dw1, db1, dw2, db2, dw3, db3
 
 

@@ -55,6 +55,13 @@ def main():
     if world != args.gpus and "RANK" in os.environ:
         world = int(os.environ["WORLD_SIZE"])
 
+    if world > 1 and args.num_actors == 32:
+        # measured node actor-supply curve (profiles/r4_mfma_attribution
+        # .md): aggregate peaks ~1.2M agent-f/s around 16-32 TOTAL actor
+        # processes on the 256-thread box and DEGRADES beyond ~64 — keep
+        # the job total near the peak instead of 32/rank x 8 = 256.
+        args.num_actors = max(4, 32 // world)
+
     flags = Flags(
         env=args.env,
         num_actors=args.num_actors,

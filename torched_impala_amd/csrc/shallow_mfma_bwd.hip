@@ -31,7 +31,10 @@ constexpr int C0 = 4, H0 = 84, W0 = 84;
 constexpr int F1 = 16, K1 = 8, S1 = 4, H1 = 20, W1 = 20;
 constexpr int F2 = 32, K2 = 4, S2 = 2, H2 = 9, W2 = 9;
 constexpr int F3 = 32, K3 = 3, H3 = 7, W3 = 7;
-constexpr int SLAB = 256;
+// 64 slabs suffice once the dW kernels grid-stride (<= DWG writers):
+// zero-fill + finalize traffic shrinks 4x vs the round-1 256.
+constexpr int SLAB = 64;
+constexpr int DWG = 512;    // dW resident-workgroup cap (2/CU x 256)
 
 // --------------------------------------------------------------------------
 // masked convert: d_z3 = bf16(d_a3_f32) .* (a3 > 0); both NCHW-flat.
@@ -208,13 +211,21 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
   __shared__ bf16_t s_d[F3 * 64];           // 4 KiB; rows padded to the
                                             // k-loop's 2x32 px range
 
-  const int img = blockIdx.x;
-  if (img >= N) return;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (288 * F3);
+  constexpr int MPW = 5;                    // ceil(18/4) m-frags per wave
+  floatx4 acc[MPW][2];
+  #pragma unroll
+  for (int m = 0; m < MPW; ++m) {
+    acc[m][0] = floatx4{0.f, 0.f, 0.f, 0.f};
+    acc[m][1] = floatx4{0.f, 0.f, 0.f, 0.f};
+  }
+  const int kbase = (lane >> 4) * 8;
 
+  for (int img = blockIdx.x; img < N; img += gridDim.x) {
+  __syncthreads();
   {
     const bf16_t* src = a2 + (size_t)img * H2 * W2 * F2;
     for (int i = tid; i < H2 * W2 * F2; i += 256) s_a[i] = src[i];
@@ -226,12 +237,11 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
   }
   __syncthreads();
 
-  const int kbase = (lane >> 4) * 8;
   // 18 m-frags (288/16), round-robin over waves
-  for (int mf = wave; mf < 18; mf += 4) {
-    floatx4 acc[2];
-    acc[0] = floatx4{0.f, 0.f, 0.f, 0.f};
-    acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+  for (int mi = 0; mi < MPW; ++mi) {
+    const int mf = wave + mi * 4;
+    if (mf >= 18) break;
     const int m = mf * 16 + (lane & 15);    // k_fwd = tap*32 + c
     const int tap = m >> 5, c = m & 31;
     const int ky = tap / 3, kx = tap % 3;
@@ -251,20 +261,31 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
           s_d + (lane & 15) * 64 + pxb);
       b1 = *reinterpret_cast<const bf16x8*>(
           s_d + (16 + (lane & 15)) * 64 + pxb);
-      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc[0], 0,
-                                                       0, 0);
-      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1, acc[1], 0,
-                                                       0, 0);
+      acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0,
+                                                           acc[mi][0], 0,
+                                                           0, 0);
+      acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1,
+                                                           acc[mi][1], 0,
+                                                           0, 0);
     }
+  }
+  }  // image loop
+
+  {
     const int dcol = lane & 15;
     const int drow = (lane >> 4) * 4;
     #pragma unroll
-    for (int n = 0; n < 2; ++n) {
-      const int f = n * 16 + dcol;
+    for (int mi = 0; mi < MPW; ++mi) {
+      const int mf = wave + mi * 4;
+      if (mf >= 18) break;
       #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int mm = mf * 16 + drow + reg;
-        atomicAdd(&dw[(size_t)mm * F3 + f], acc[n][reg]);
+      for (int n = 0; n < 2; ++n) {
+        const int f = n * 16 + dcol;
+        #pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int mm = mf * 16 + drow + reg;
+          atomicAdd(&dw[(size_t)mm * F3 + f], acc[mi][n][reg]);
+        }
       }
     }
   }
@@ -281,13 +302,21 @@ __global__ __launch_bounds__(256, 2) void conv2_dw(
   __shared__ bf16_t s_a[F1 * H1 * W1];      // 12.5 KiB planar
   __shared__ bf16_t s_d[F2 * 96];           // 6 KiB transposed
 
-  const int img = blockIdx.x;
-  if (img >= N) return;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (256 * F2);
+  constexpr int MPW = 4;                    // 16 m-frags over 4 waves
+  floatx4 acc[MPW][2];
+  #pragma unroll
+  for (int m = 0; m < MPW; ++m) {
+    acc[m][0] = floatx4{0.f, 0.f, 0.f, 0.f};
+    acc[m][1] = floatx4{0.f, 0.f, 0.f, 0.f};
+  }
+  const int kbase = (lane >> 4) * 8;
 
+  for (int img = blockIdx.x; img < N; img += gridDim.x) {
+  __syncthreads();
   {
     const bf16_t* src = a1 + (size_t)img * H1 * W1 * F1;
     for (int i = tid; i < F1 * H1 * W1; i += 256) {
@@ -304,11 +333,9 @@ __global__ __launch_bounds__(256, 2) void conv2_dw(
   }
   __syncthreads();
 
-  const int kbase = (lane >> 4) * 8;
-  for (int mf = wave; mf < 16; mf += 4) {   // 256/16 m-frags
-    floatx4 acc[2];
-    acc[0] = floatx4{0.f, 0.f, 0.f, 0.f};
-    acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+  for (int mi = 0; mi < MPW; ++mi) {        // 256/16 m-frags
+    const int mf = wave + mi * 4;
     const int m = mf * 16 + (lane & 15);    // c*16 + ky*4 + kx
     const int c = m >> 4;
     const int ky = (m & 15) >> 2, kx = m & 3;
@@ -328,20 +355,30 @@ __global__ __launch_bounds__(256, 2) void conv2_dw(
           s_d + (lane & 15) * 96 + pxb);
       const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
           s_d + (16 + (lane & 15)) * 96 + pxb);
-      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc[0], 0,
-                                                       0, 0);
-      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1, acc[1], 0,
-                                                       0, 0);
+      acc[mi][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0,
+                                                           acc[mi][0], 0,
+                                                           0, 0);
+      acc[mi][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1,
+                                                           acc[mi][1], 0,
+                                                           0, 0);
     }
+  }
+  }  // image loop
+
+  {
     const int dcol = lane & 15;
     const int drow = (lane >> 4) * 4;
     #pragma unroll
-    for (int n = 0; n < 2; ++n) {
-      const int f = n * 16 + dcol;
+    for (int mi = 0; mi < MPW; ++mi) {
+      const int mf = wave + mi * 4;
       #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int mm = mf * 16 + drow + reg;
-        atomicAdd(&dw[(size_t)mm * F2 + f], acc[n][reg]);
+      for (int n = 0; n < 2; ++n) {
+        const int f = n * 16 + dcol;
+        #pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int mm = mf * 16 + drow + reg;
+          atomicAdd(&dw[(size_t)mm * F2 + f], acc[mi][n][reg]);
+        }
       }
     }
   }
@@ -359,13 +396,18 @@ __global__ __launch_bounds__(256, 2) void conv1_dw(
   __shared__ bf16_t s_d[F1 * 416];          // 13 KiB transposed
   // 68.1 KiB total -> 2 blocks/CU in 160 KiB LDS
 
-  const int img = blockIdx.x;
-  if (img >= N) return;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (256 * F1);
+  constexpr int MPW = 4;                    // 16 m-frags over 4 waves
+  floatx4 acc[MPW];
+  #pragma unroll
+  for (int m = 0; m < MPW; ++m) acc[m] = floatx4{0.f, 0.f, 0.f, 0.f};
+  const int kbase = (lane >> 4) * 8;
 
+  for (int img = blockIdx.x; img < N; img += gridDim.x) {
+  __syncthreads();
   {
     // frames: contiguous u8 -> bf16, uchar4-vectorised (28224 % 4 == 0)
     const unsigned char* fr = frames + (size_t)img * C0 * H0 * W0;
@@ -389,9 +431,9 @@ __global__ __launch_bounds__(256, 2) void conv1_dw(
   }
   __syncthreads();
 
-  const int kbase = (lane >> 4) * 8;
-  for (int mf = wave; mf < 16; mf += 4) {
-    floatx4 acc = floatx4{0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+  for (int mi = 0; mi < MPW; ++mi) {
+    const int mf = wave + mi * 4;
     const int m = mf * 16 + (lane & 15);    // c*64 + ky*8 + kx
     const int c = m >> 6;
     const int ky = (m & 63) >> 3, kx = m & 7;
@@ -408,14 +450,23 @@ __global__ __launch_bounds__(256, 2) void conv1_dw(
       }
       const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
           s_d + (lane & 15) * 416 + s * 32 + kbase);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc, 0, 0, 0);
+      acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0, acc[mi],
+                                                        0, 0, 0);
     }
+  }
+  }  // image loop
+
+  {
     const int dcol = lane & 15;
     const int drow = (lane >> 4) * 4;
     #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      const int mm = mf * 16 + drow + reg;
-      atomicAdd(&dw[(size_t)mm * F1 + dcol], acc[reg]);
+    for (int mi = 0; mi < MPW; ++mi) {
+      const int mf = wave + mi * 4;
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int mm = mf * 16 + drow + reg;
+        atomicAdd(&dw[(size_t)mm * F1 + dcol], acc[mi][reg]);
+      }
     }
   }
 }
@@ -559,7 +610,7 @@ extern "C" void launch_shallow_mfma_bwd(
                      static_cast<const bf16_t*>(wr3),
                      static_cast<const bf16_t*>(a2),
                      static_cast<bf16_t*>(da2), N);
-  hipLaunchKernelGGL(conv3_dw, dim3(N), dim3(256), 0, s,
+  hipLaunchKernelGGL(conv3_dw, dim3(N < DWG ? N : DWG), dim3(256), 0, s,
                      static_cast<const bf16_t*>(a2),
                      static_cast<const bf16_t*>(dz3), dw3_slab, N);
   hipLaunchKernelGGL(conv2_dx, dim3(N), dim3(256), 0, s,
@@ -567,10 +618,10 @@ extern "C" void launch_shallow_mfma_bwd(
                      static_cast<const bf16_t*>(wr2),
                      static_cast<const bf16_t*>(a1),
                      static_cast<bf16_t*>(da1), N);
-  hipLaunchKernelGGL(conv2_dw, dim3(N), dim3(256), 0, s,
+  hipLaunchKernelGGL(conv2_dw, dim3(N < DWG ? N : DWG), dim3(256), 0, s,
                      static_cast<const bf16_t*>(a1),
                      static_cast<const bf16_t*>(da2), dw2_slab, N);
-  hipLaunchKernelGGL(conv1_dw, dim3(N), dim3(256), 0, s, frames,
+  hipLaunchKernelGGL(conv1_dw, dim3(N < DWG ? N : DWG), dim3(256), 0, s, frames,
                      static_cast<const bf16_t*>(da1), dw1_slab, N);
 }
 

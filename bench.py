@@ -146,7 +146,10 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": flags.compute_dtype,
+            # conservative label: the encoder conv GEMMs (bf16 MFMA) and
+            # the LSTM recurrent weights run bf16; V-trace/loss/optimizer
+            # and all master weights are fp32
+            "dtype": "bf16",
             "data": "synthetic",
             "config": {
                 "model": ("shallow+lstm" if flags.use_lstm else "shallow"),
@@ -164,6 +167,8 @@ def main():
                 "frame_convention": "agent_steps",
                 "env_frames_per_sec_x4_convention": value * 4,
                 "obs": "84x84x4 uint8 synthetic",
+                "precision": "mixed: bf16 conv/linear MFMA + LSTM "
+                             "weights, fp32 masters/V-trace/optimizer",
             },
         }), flush=True)
 

@@ -30,13 +30,14 @@ void launch_vtrace_scan(const float*, const float*, const float*, const float*,
 void launch_vtrace_loss_bwd(const float*, const float*, const long*,
                             const float*, const float*, const float*, float,
                             float, int, int, float*, float*, hipStream_t);
-void launch_lstm_seq_fwd(const float*, const float*, const float*,
-                         const float*, const float*, int, int, float*, float*,
-                         float*, float*, float*, float*, int, hipStream_t);
+void launch_lstm_seq_fwd(const float*, const void*, const float*,
+                         const float*, const float*, int, int, float*,
+                         float*, float*, float*, float*, float*, int, int,
+                         hipStream_t);
 void launch_lstm_seq_bwd(const float*, const float*, const float*,
                          const float*, const float*, const float*,
-                         const float*, const float*, int, int, float*, float*,
-                         float*, hipStream_t);
+                         const float*, const void*, int, int, float*,
+                         float*, float*, int, hipStream_t);
 void launch_grad_sqnorm(const float*, long, const float*, float*,
                         hipStream_t);
 void launch_rmsprop_update(float*, const float*, float*, long, const float*,
@@ -239,7 +240,10 @@ std::vector<torch::Tensor> lstm_seq_fwd(
     torch::Tensor Xg, torch::Tensor W_hhT, torch::Tensor notdone,
     torch::Tensor h0, torch::Tensor c0, bool save_for_bwd) {
   check_f32(Xg, "Xg");
-  check_f32(W_hhT, "W_hhT");
+  const bool wbf16 = W_hhT.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(wbf16 || W_hhT.scalar_type() == torch::kFloat32,
+              "W_hhT must be fp32 or bf16");
+  TORCH_CHECK(W_hhT.is_cuda() && W_hhT.is_contiguous(), "W_hhT layout");
   check_f32(notdone, "notdone");
   check_f32(h0, "h0");
   check_f32(c0, "c0");
@@ -263,13 +267,13 @@ std::vector<torch::Tensor> lstm_seq_fwd(
     C = torch::empty({1}, opts);
     Hp = torch::empty({1}, opts);
   }
-  launch_lstm_seq_fwd(Xg.data_ptr<float>(), W_hhT.data_ptr<float>(),
+  launch_lstm_seq_fwd(Xg.data_ptr<float>(), W_hhT.data_ptr(),
                       notdone.data_ptr<float>(), h0.data_ptr<float>(),
                       c0.data_ptr<float>(), T, B, H.data_ptr<float>(),
                       act.data_ptr<float>(), C.data_ptr<float>(),
                       Hp.data_ptr<float>(), hT.data_ptr<float>(),
                       cT.data_ptr<float>(), save_for_bwd ? 1 : 0,
-                      current_stream());
+                      wbf16 ? 1 : 0, current_stream());
   HIP_CHECK(hipGetLastError());
   return {H, act, C, Hp, hT, cT};
 }
@@ -280,7 +284,10 @@ std::vector<torch::Tensor> lstm_seq_bwd(
     torch::Tensor notdone, torch::Tensor W_hh) {
   check_f32(dH, "dH");
   check_f32(act, "act");
-  check_f32(W_hh, "W_hh");
+  const bool wbf16 = W_hh.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(wbf16 || W_hh.scalar_type() == torch::kFloat32,
+              "W_hh must be fp32 or bf16");
+  TORCH_CHECK(W_hh.is_cuda() && W_hh.is_contiguous(), "W_hh layout");
   const int T = dH.size(0);
   const int B = dH.size(1);
   TORCH_CHECK(dH.size(2) == 256 && W_hh.size(0) == 1024 && W_hh.size(1) == 256,
@@ -293,9 +300,10 @@ std::vector<torch::Tensor> lstm_seq_bwd(
   launch_lstm_seq_bwd(dH.data_ptr<float>(), dhT.data_ptr<float>(),
                       dcT.data_ptr<float>(), act.data_ptr<float>(),
                       C.data_ptr<float>(), c0.data_ptr<float>(),
-                      notdone.data_ptr<float>(), W_hh.data_ptr<float>(), T, B,
+                      notdone.data_ptr<float>(), W_hh.data_ptr(), T, B,
                       dG.data_ptr<float>(), dh0.data_ptr<float>(),
-                      dc0.data_ptr<float>(), current_stream());
+                      dc0.data_ptr<float>(), wbf16 ? 1 : 0,
+                      current_stream());
   HIP_CHECK(hipGetLastError());
   return {dG, dh0, dc0};
 }

@@ -54,6 +54,15 @@ __device__ __forceinline__ float fast_tanh(float x) {
   return (e - 1.0f) * fast_rcp(e + 1.0f);
 }
 
+// templated weight load: fp32 exact path or bf16 (half the L2 bytes —
+// the recurrence GEMV is L2-bandwidth-bound per CU: 1 MiB of W per WG
+// per timestep; r8 profile 189 us fwd at T=20 B=32).
+typedef __bf16 lstm_bf16_t;
+template <typename WT>
+__device__ __forceinline__ float loadw(const WT* p) {
+  return (float)*p;
+}
+
 // ---------------------------------------------------------------------------
 // forward: grid = (B) blocks x 512 threads
 //   Xg      (T, B, 1024)  x-part of gates incl. both biases
@@ -67,8 +76,9 @@ __device__ __forceinline__ float fast_tanh(float x) {
 //   Hp      (T, B, 256)   masked h input at step t (saved for bwd + dW_hh)
 //   hT, cT  (B, 256)
 // ---------------------------------------------------------------------------
+template <typename WT>
 __global__ void __launch_bounds__(THREADS)
-lstm_seq_fwd(const float* __restrict__ Xg, const float* __restrict__ W_hhT,
+lstm_seq_fwd(const float* __restrict__ Xg, const WT* __restrict__ W_hhT,
              const float* __restrict__ notdone, const float* __restrict__ h0,
              const float* __restrict__ c0, int T, int B,
              float* __restrict__ Hout, float* __restrict__ act,
@@ -103,14 +113,14 @@ lstm_seq_fwd(const float* __restrict__ Xg, const float* __restrict__ W_hhT,
     // ds_read_b128 broadcast of h, four batched W row loads in flight).
     {
       float acc = Xg[rowG + tid];
-      const float* wcol = W_hhT + tid;
+      const WT* wcol = W_hhT + tid;
       #pragma unroll 4
       for (int k0 = 0; k0 < H; k0 += 4) {
         const float4 hv = *reinterpret_cast<const float4*>(&s_h[k0]);
-        const float w0 = wcol[(size_t)(k0 + 0) * G4];
-        const float w1 = wcol[(size_t)(k0 + 1) * G4];
-        const float w2 = wcol[(size_t)(k0 + 2) * G4];
-        const float w3 = wcol[(size_t)(k0 + 3) * G4];
+        const float w0 = loadw(wcol + (size_t)(k0 + 0) * G4);
+        const float w1 = loadw(wcol + (size_t)(k0 + 1) * G4);
+        const float w2 = loadw(wcol + (size_t)(k0 + 2) * G4);
+        const float w3 = loadw(wcol + (size_t)(k0 + 3) * G4);
         acc = fmaf(hv.x, w0, acc);
         acc = fmaf(hv.y, w1, acc);
         acc = fmaf(hv.z, w2, acc);
@@ -149,12 +159,13 @@ lstm_seq_fwd(const float* __restrict__ Xg, const float* __restrict__ W_hhT,
 // backward: grid = (B) blocks x 512 threads, reverse time loop.
 // Emits dG (T,B,1024) and dh0/dc0; weight grads are host-side GEMMs.
 // ---------------------------------------------------------------------------
+template <typename WT>
 __global__ void __launch_bounds__(THREADS)
 lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
              const float* __restrict__ dcT, const float* __restrict__ act,
              const float* __restrict__ Cout, const float* __restrict__ c0,
              const float* __restrict__ notdone,
-             const float* __restrict__ W_hh,  // (1024, 256) native layout
+             const WT* __restrict__ W_hh,  // (1024, 256) native layout
              int T, int B, float* __restrict__ dG,
              float* __restrict__ dh0, float* __restrict__ dc0) {
   __shared__ __align__(16) float s_dh[H];
@@ -205,15 +216,15 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
     {
       const int k = tid & (H - 1);
       const int j0 = (tid >> 8) * (G4 / 4);
-      const float* wcol = W_hh + (size_t)j0 * H + k;
+      const WT* wcol = W_hh + (size_t)j0 * H + k;
       float acc = 0.f;
       #pragma unroll 4
       for (int j = 0; j < G4 / 4; j += 4) {
         const float4 dg = *reinterpret_cast<const float4*>(&s_dg[j0 + j]);
-        const float w0 = wcol[(size_t)(j + 0) * H];
-        const float w1 = wcol[(size_t)(j + 1) * H];
-        const float w2 = wcol[(size_t)(j + 2) * H];
-        const float w3 = wcol[(size_t)(j + 3) * H];
+        const float w0 = loadw(wcol + (size_t)(j + 0) * H);
+        const float w1 = loadw(wcol + (size_t)(j + 1) * H);
+        const float w2 = loadw(wcol + (size_t)(j + 2) * H);
+        const float w3 = loadw(wcol + (size_t)(j + 3) * H);
         acc = fmaf(dg.x, w0, acc);
         acc = fmaf(dg.y, w1, acc);
         acc = fmaf(dg.z, w2, acc);
@@ -238,24 +249,41 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
 
 extern "C" {
 
-void launch_lstm_seq_fwd(const float* Xg, const float* W_hhT,
+void launch_lstm_seq_fwd(const float* Xg, const void* W_hhT,
                          const float* notdone, const float* h0,
                          const float* c0, int T, int B, float* Hout,
                          float* act, float* Cout, float* Hp, float* hT,
-                         float* cT, int save_for_bwd, hipStream_t stream) {
-  hipLaunchKernelGGL(impala::lstm_seq_fwd, dim3(B), dim3(impala::THREADS), 0,
-                     stream, Xg, W_hhT, notdone, h0, c0, T, B, Hout, act,
-                     Cout, Hp, hT, cT, save_for_bwd);
+                         float* cT, int save_for_bwd, int wbf16,
+                         hipStream_t stream) {
+  if (wbf16)
+    hipLaunchKernelGGL(impala::lstm_seq_fwd<impala::lstm_bf16_t>, dim3(B),
+                       dim3(impala::THREADS), 0, stream, Xg,
+                       static_cast<const impala::lstm_bf16_t*>(W_hhT),
+                       notdone, h0, c0, T, B, Hout, act, Cout, Hp, hT, cT,
+                       save_for_bwd);
+  else
+    hipLaunchKernelGGL(impala::lstm_seq_fwd<float>, dim3(B),
+                       dim3(impala::THREADS), 0, stream, Xg,
+                       static_cast<const float*>(W_hhT), notdone, h0, c0, T,
+                       B, Hout, act, Cout, Hp, hT, cT, save_for_bwd);
 }
 
 void launch_lstm_seq_bwd(const float* dH, const float* dhT, const float* dcT,
                          const float* act, const float* Cout, const float* c0,
-                         const float* notdone, const float* W_hh, int T,
-                         int B, float* dG, float* dh0, float* dc0,
+                         const float* notdone, const void* W_hh, int T,
+                         int B, float* dG, float* dh0, float* dc0, int wbf16,
                          hipStream_t stream) {
-  hipLaunchKernelGGL(impala::lstm_seq_bwd, dim3(B), dim3(impala::THREADS), 0,
-                     stream, dH, dhT, dcT, act, Cout, c0, notdone, W_hh, T, B,
-                     dG, dh0, dc0);
+  if (wbf16)
+    hipLaunchKernelGGL(impala::lstm_seq_bwd<impala::lstm_bf16_t>, dim3(B),
+                       dim3(impala::THREADS), 0, stream, dH, dhT, dcT, act,
+                       Cout, c0, notdone,
+                       static_cast<const impala::lstm_bf16_t*>(W_hh), T, B,
+                       dG, dh0, dc0);
+  else
+    hipLaunchKernelGGL(impala::lstm_seq_bwd<float>, dim3(B),
+                       dim3(impala::THREADS), 0, stream, dH, dhT, dcT, act,
+                       Cout, c0, notdone, static_cast<const float*>(W_hh),
+                       T, B, dG, dh0, dc0);
 }
 
 }  // extern "C"

@@ -12,9 +12,19 @@ tests/test_gpu_kernels.py::test_lstm_seq_parity*).
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 from torched_impala_amd import ops
+
+
+def lstm_bf16_weights() -> bool:
+    """bf16 recurrent weights (fp32 masters + accumulation): the
+    recurrence GEMV is L2-bandwidth-bound on the 1 MiB W_hh per WG per
+    timestep (r8 profile: 189 us fwd), so halving the bytes ~halves the
+    sequential chain. IMPALA_LSTM_BF16=0 restores exact-fp32 weights."""
+    return os.environ.get("IMPALA_LSTM_BF16", "1") != "0"
 
 
 class _LstmSeq(torch.autograd.Function):
@@ -24,10 +34,14 @@ class _LstmSeq(torch.autograd.Function):
         # NB: grad mode is disabled inside Function.forward —
         # needs_input_grad is the correct signal here.
         need_bwd = ctx.needs_input_grad[0] or ctx.needs_input_grad[1]
+        bf16w = lstm_bf16_weights()
         W_hhT = W_hh.detach().t().contiguous()
+        if bf16w:
+            W_hhT = W_hhT.to(torch.bfloat16)
         H, act, C, Hp, hT, cT = ext.lstm_seq_fwd(
             Xg.contiguous(), W_hhT, notdone.contiguous(),
             h0.contiguous(), c0.contiguous(), need_bwd)
+        ctx.bf16w = bf16w
         if need_bwd:
             ctx.save_for_backward(act, C, Hp, c0, notdone, W_hh)
         ctx.need_bwd = need_bwd
@@ -42,10 +56,12 @@ class _LstmSeq(torch.autograd.Function):
             dhT = torch.zeros(B, 256, device=dH.device, dtype=dH.dtype)
         if dcT is None:
             dcT = torch.zeros(B, 256, device=dH.device, dtype=dH.dtype)
+        W = W_hh.detach().contiguous()
+        if getattr(ctx, "bf16w", False):
+            W = W.to(torch.bfloat16)
         dG, dh0, dc0 = ext.lstm_seq_bwd(
             dH.contiguous(), dhT.contiguous(), dcT.contiguous(),
-            act, C, c0.contiguous(), notdone.contiguous(),
-            W_hh.detach().contiguous())
+            act, C, c0.contiguous(), notdone.contiguous(), W)
         # dW_hh = sum_{t,b} dG[t,b,:]^T Hp[t,b,:]  — one (1024,TB)x(TB,256) GEMM
         dW_hh = dG.reshape(-1, 1024).t().mm(Hp.reshape(-1, 256))
         return dG, dW_hh, dh0, dc0, None

@@ -664,9 +664,9 @@ std::vector<torch::Tensor> shallow_mfma_bwd(torch::Tensor frames,
                           s3.data_ptr<float>(), N, current_stream());
   // fused finalize: slab sums -> torch weight layouts + bias column sums
   // (one launch; replaces ~12 eager ATen ops — profiles/r6 PMC)
-  auto dw1 = torch::empty({16, 4, 8, 8}, fopts);
-  auto dw2 = torch::empty({32, 16, 4, 4}, fopts);
-  auto dw3 = torch::empty({32, 32, 3, 3}, fopts);
+  auto dw1 = torch::zeros({16, 4, 8, 8}, fopts);
+  auto dw2 = torch::zeros({32, 16, 4, 4}, fopts);
+  auto dw3 = torch::zeros({32, 32, 3, 3}, fopts);
   auto db = torch::zeros({80}, fopts);
   float* dbp = db.data_ptr<float>();
   launch_shallow_bwd_finalize(

@@ -495,42 +495,49 @@ __global__ __launch_bounds__(256) void bwd_finalize(
     float* __restrict__ dw1, float* __restrict__ dw2,
     float* __restrict__ dw3, float* __restrict__ db1,
     float* __restrict__ db2, float* __restrict__ db3) {
-  constexpr int BW1 = 16, BW2 = 32, BW3 = 36;   // dw blocks (256 thr each)
-  constexpr int BB = 64;                        // db blocks per tensor
+  constexpr int SPLIT = 4;                      // slab-dim parallelism
+  constexpr int SLC = SLAB / SPLIT;             // slabs per block
+  constexpr int BW1 = 16 * SPLIT, BW2 = 32 * SPLIT, BW3 = 36 * SPLIT;
+  constexpr int BB = 160;                       // db blocks per tensor
   const int blk = blockIdx.x;
   const int tid = threadIdx.x;
 
   if (blk < BW1) {                               // dw1: 4096 outputs
-    const int o = blk * 256 + tid;
+    const int o = (blk / SPLIT) * 256 + tid;
+    const int sl0 = (blk % SPLIT) * SLC;
     const int kx = o & 7, ky = (o >> 3) & 7, c = (o >> 6) & 3;
     const int f = o >> 8;
     const int m = c * 64 + ky * 8 + kx;
     float v = 0.f;
-    for (int sl = 0; sl < SLAB; ++sl) v += s1[(size_t)sl * 4096 + m * 16 + f];
-    dw1[o] = v;
+    for (int sl = sl0; sl < sl0 + SLC; ++sl)
+      v += s1[(size_t)sl * 4096 + m * 16 + f];
+    atomicAdd(&dw1[o], v);                       // dw* zero-initialised
     return;
   }
   if (blk < BW1 + BW2) {                         // dw2: 8192 outputs
-    const int o = (blk - BW1) * 256 + tid;
+    const int o = ((blk - BW1) / SPLIT) * 256 + tid;
+    const int sl0 = ((blk - BW1) % SPLIT) * SLC;
     const int kx = o & 3, ky = (o >> 2) & 3, c = (o >> 4) & 15;
     const int f = o >> 8;
     const int m = c * 16 + ky * 4 + kx;
     float v = 0.f;
-    for (int sl = 0; sl < SLAB; ++sl) v += s2[(size_t)sl * 8192 + m * 32 + f];
-    dw2[o] = v;
+    for (int sl = sl0; sl < sl0 + SLC; ++sl)
+      v += s2[(size_t)sl * 8192 + m * 32 + f];
+    atomicAdd(&dw2[o], v);
     return;
   }
   if (blk < BW1 + BW2 + BW3) {                   // dw3: 9216 outputs
-    const int o = (blk - BW1 - BW2) * 256 + tid;
+    const int o = ((blk - BW1 - BW2) / SPLIT) * 256 + tid;
+    const int sl0 = ((blk - BW1 - BW2) % SPLIT) * SLC;
     if (o < 9216) {
       const int t = o % 9;                       // tap
       const int c = (o / 9) & 31;
       const int f = o / 288;
       const int m = t * 32 + c;
       float v = 0.f;
-      for (int sl = 0; sl < SLAB; ++sl)
+      for (int sl = sl0; sl < sl0 + SLC; ++sl)
         v += s3[(size_t)sl * 9216 + m * 32 + f];
-      dw3[o] = v;
+      atomicAdd(&dw3[o], v);
     }
     return;
   }

@@ -12,6 +12,23 @@ from __future__ import annotations
 
 import numpy as np
 
+# one frame bank per (shape) per PROCESS: the bank is seeded identically
+# for every env, so per-instance copies were pure waste — 16 envs/actor
+# x 1.8 MiB duplicated banks blew the actor working set past the EPYC's
+# 32 MiB/CCD L3 and capped the node's aggregate actor throughput
+# (profiles/r4_mfma_attribution.md supply curve).
+_BANK_CACHE: dict = {}
+
+
+def _shared_bank(shape):
+    b = _BANK_CACHE.get(shape)
+    if b is None:
+        rng = np.random.default_rng(12345)
+        b = rng.integers(0, 256, size=(SyntheticFrameEnv.FRAME_BANK, *shape),
+                         dtype=np.uint8)
+        _BANK_CACHE[shape] = b
+    return b
+
 
 class SyntheticFrameEnv:
     """Atari/DMLab-shaped synthetic environment.
@@ -31,9 +48,7 @@ class SyntheticFrameEnv:
         self.num_actions = num_actions
         self.episode_len = episode_len
         self.seed = seed
-        rng = np.random.default_rng(12345)  # bank shared across envs
-        self.bank = rng.integers(0, 256, size=(self.FRAME_BANK, *self.obs_shape),
-                                 dtype=np.uint8)
+        self.bank = _shared_bank(self.obs_shape)  # per-process, shared
         rng2 = np.random.default_rng(seed)
         self.reward_bank = (rng2.integers(0, 3, size=1024).astype(np.float32) - 1.0)
         # stagger the first episode boundary per env (same-seed envs stay

@@ -20,11 +20,11 @@ from torched_impala_amd import ops
 
 
 def lstm_bf16_weights() -> bool:
-    """bf16 recurrent weights (fp32 masters + accumulation): the
-    recurrence GEMV is L2-bandwidth-bound on the 1 MiB W_hh per WG per
-    timestep (r8 profile: 189 us fwd), so halving the bytes ~halves the
-    sequential chain. IMPALA_LSTM_BF16=0 restores exact-fp32 weights."""
-    return os.environ.get("IMPALA_LSTM_BF16", "1") != "0"
+    """bf16 recurrent weights (fp32 masters + accumulation). Measured
+    ~1% end-to-end (r9 A/B: 1.256 vs 1.267 ms/step) — the recurrence is
+    barrier/latency-bound, not W-bandwidth-bound — so the exact-fp32
+    path stays the DEFAULT; IMPALA_LSTM_BF16=1 enables the bf16 A/B."""
+    return os.environ.get("IMPALA_LSTM_BF16", "0") == "1"
 
 
 class _LstmSeq(torch.autograd.Function):

@@ -167,8 +167,8 @@ def main():
                 "frame_convention": "agent_steps",
                 "env_frames_per_sec_x4_convention": value * 4,
                 "obs": "84x84x4 uint8 synthetic",
-                "precision": "mixed: bf16 conv/linear MFMA + LSTM "
-                             "weights, fp32 masters/V-trace/optimizer",
+                "precision": "mixed: bf16 conv-encoder MFMA, fp32 "
+                             "masters/LSTM/V-trace/optimizer",
             },
         }), flush=True)
 

@@ -86,7 +86,6 @@ lstm_seq_fwd(const float* __restrict__ Xg, const WT* __restrict__ W_hhT,
              float* __restrict__ hT, float* __restrict__ cT,
              int save_for_bwd) {
   __shared__ __align__(16) float s_h[H];
-  __shared__ __align__(16) float s_cm[H];
   __shared__ __align__(16) float s_c[H];
   __shared__ __align__(16) float s_g[G4];
 
@@ -102,17 +101,15 @@ lstm_seq_fwd(const float* __restrict__ Xg, const WT* __restrict__ W_hhT,
     const float nd = notdone[t * B + b];
     const size_t rowH = ((size_t)t * B + b) * H;
     const size_t rowG = ((size_t)t * B + b) * G4;
-    if (tid < H) {
-      const float hm = s_h[tid] * nd;
-      s_h[tid] = hm;  // masked input h' for this step
-      s_cm[tid] = s_c[tid] * nd;
-      if (save_for_bwd) Hp[rowH + tid] = hm;
-    }
-    __syncthreads();
+    // the done-mask is a SCALAR per (t,b): gates = Xg + (nd*h).W
+    //                                            = Xg + nd*(h.W)
+    // so no masking phase (and barrier) is needed before the GEMV —
+    // the 16-wave barrier was ~1/3 of the sequential critical path.
+    if (save_for_bwd && tid < H) Hp[rowH + tid] = s_h[tid] * nd;
     // gate GEMV: thread j computes gate column j. k chunked by 4 (one
     // ds_read_b128 broadcast of h, four batched W row loads in flight).
     {
-      float acc = Xg[rowG + tid];
+      float acc = 0.f;
       const WT* wcol = W_hhT + tid;
       #pragma unroll 4
       for (int k0 = 0; k0 < H; k0 += 4) {
@@ -126,7 +123,7 @@ lstm_seq_fwd(const float* __restrict__ Xg, const WT* __restrict__ W_hhT,
         acc = fmaf(hv.z, w2, acc);
         acc = fmaf(hv.w, w3, acc);
       }
-      s_g[tid] = acc;
+      s_g[tid] = fmaf(nd, acc, Xg[rowG + tid]);
     }
     __syncthreads();
     if (tid < H) {
@@ -134,7 +131,7 @@ lstm_seq_fwd(const float* __restrict__ Xg, const WT* __restrict__ W_hhT,
       const float gf = sigmoidf(s_g[H + tid]);
       const float gg = fast_tanh(s_g[2 * H + tid]);
       const float go = sigmoidf(s_g[3 * H + tid]);
-      const float c = fmaf(gf, s_cm[tid], gi * gg);
+      const float c = fmaf(gf, s_c[tid] * nd, gi * gg);
       const float h = go * fast_tanh(c);
       if (save_for_bwd) {
         act[rowG + tid] = gi;
@@ -181,6 +178,12 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
   }
   __syncthreads();
 
+  // the per-step combine of the four GEMV quarter-partials is FOLDED
+  // into the next iteration's elementwise phase (reading s_part
+  // directly), cutting the 16-wave barriers per step from 3 to 2 —
+  // the recurrence is barrier-latency-bound, not bandwidth-bound.
+  float prev_nd = 1.f;
+  bool have_parts = false;
   for (int t = T - 1; t >= 0; --t) {
     const float nd = notdone[t * B + b];
     const size_t rowH = ((size_t)t * B + b) * H;
@@ -194,7 +197,11 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
       const float gf = act[rowG + H + tid];
       const float gg = act[rowG + 2 * H + tid];
       const float go = act[rowG + 3 * H + tid];
-      const float dh = dH[rowH + tid] + s_dh[tid];
+      const float carried = have_parts
+          ? (s_part[tid] + s_part[H + tid] + s_part[2 * H + tid] +
+             s_part[3 * H + tid]) * prev_nd
+          : s_dh[tid];
+      const float dh = dH[rowH + tid] + carried;
       const float dct = s_dc[tid] + dh * go * (1.f - tc * tc);
       const float dgi = (dct * gg) * gi * (1.f - gi);
       const float dgf = (dct * cprev) * gf * (1.f - gf);
@@ -233,14 +240,14 @@ lstm_seq_bwd(const float* __restrict__ dH, const float* __restrict__ dhT,
       s_part[tid] = acc;
     }
     __syncthreads();
-    if (tid < H) {
-      s_dh[tid] = (s_part[tid] + s_part[H + tid] + s_part[2 * H + tid] +
-                   s_part[3 * H + tid]) * nd;
-    }
-    __syncthreads();
+    prev_nd = nd;
+    have_parts = true;
   }
   if (tid < H) {
-    dh0[b * H + tid] = s_dh[tid];
+    dh0[b * H + tid] = have_parts
+        ? (s_part[tid] + s_part[H + tid] + s_part[2 * H + tid] +
+           s_part[3 * H + tid]) * prev_nd
+        : s_dh[tid];
     dc0[b * H + tid] = s_dc[tid];
   }
 }

@@ -35,6 +35,10 @@ constexpr int F3 = 32, K3 = 3, H3 = 7, W3 = 7;
 // zero-fill + finalize traffic shrinks 4x vs the round-1 256.
 constexpr int SLAB = 64;
 constexpr int DWG = 512;    // dW resident-workgroup cap (2/CU x 256)
+// bwd_finalize geometry — grid and kernel MUST agree (single source):
+constexpr int FIN_SPLIT = 4;    // slab-dim parallelism per dW chunk
+constexpr int FIN_BB = 160;     // db column-sum blocks per tensor
+constexpr int FIN_GRID = (16 + 32 + 36) * FIN_SPLIT + 3 * FIN_BB;
 
 // --------------------------------------------------------------------------
 // masked convert: d_z3 = bf16(d_a3_f32) .* (a3 > 0); both NCHW-flat.
@@ -495,10 +499,10 @@ __global__ __launch_bounds__(256) void bwd_finalize(
     float* __restrict__ dw1, float* __restrict__ dw2,
     float* __restrict__ dw3, float* __restrict__ db1,
     float* __restrict__ db2, float* __restrict__ db3) {
-  constexpr int SPLIT = 4;                      // slab-dim parallelism
-  constexpr int SLC = SLAB / SPLIT;             // slabs per block
+  constexpr int SLC = SLAB / FIN_SPLIT;         // slabs per block
+  constexpr int SPLIT = FIN_SPLIT;
   constexpr int BW1 = 16 * SPLIT, BW2 = 32 * SPLIT, BW3 = 36 * SPLIT;
-  constexpr int BB = 160;                       // db blocks per tensor
+  constexpr int BB = FIN_BB;
   const int blk = blockIdx.x;
   const int tid = threadIdx.x;
 
@@ -642,8 +646,12 @@ extern "C" void launch_shallow_bwd_finalize(
     const void* da2, const void* dz3, int N, float* dw1, float* dw2,
     float* dw3, float* db1, float* db2, float* db3, hipStream_t s) {
   using impala_shallow_mfma_bwd::bf16_t;
+  // FIN_GRID is the kernel's section geometry — single source of truth
+  // (r9/r10 ran a stale 276-block grid: dw3 58% written, db sections
+  // never launched — caught by test_shallow_encoder_parity_mfma).
   hipLaunchKernelGGL(impala_shallow_mfma_bwd::bwd_finalize,
-                     dim3(16 + 32 + 36 + 3 * 64), dim3(256), 0, s, s1, s2,
+                     dim3(impala_shallow_mfma_bwd::FIN_GRID), dim3(256), 0,
+                     s, s1, s2,
                      s3, static_cast<const bf16_t*>(da1),
                      static_cast<const bf16_t*>(da2),
                      static_cast<const bf16_t*>(dz3), N, dw1, dw2, dw3, db1,

@@ -147,6 +147,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--synthetic-episode-len", type=int, default=f.synthetic_episode_len)
     p.add_argument("--seed", type=int, default=f.seed)
     p.add_argument("--metrics-jsonl", default=None)
+    p.add_argument("--checkpoint-interval-s", type=float,
+                   default=f.checkpoint_interval_s)
     p.add_argument("--weight-sync-interval", type=int, default=f.weight_sync_interval)
     return p
 

@@ -1,0 +1,82 @@
+"""Find the first non-finite learner step under real training (the r12
+GPU soak showed loss=nan; short tests stay finite). Reports the step at
+which total_loss/pg/baseline/entropy first go non-finite, plus the last
+good stats, for A/B-ing the suspected gates:
+
+  python scripts/nan_probe.py --steps 3000                 # default path
+  IMPALA_SHALLOW_MFMA=0 python scripts/nan_probe.py ...    # fp32 encoder
+  python scripts/nan_probe.py --actor-jit frozen ...       # no int8 actor
+  python scripts/nan_probe.py --no-graphs ...
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import math
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=2000)
+    p.add_argument("--num-actors", type=int, default=16)
+    p.add_argument("--envs-per-actor", type=int, default=16)
+    p.add_argument("--actor-jit", default="native-bf16")
+    p.add_argument("--use-lstm", action="store_true")
+    p.add_argument("--no-graphs", action="store_true")
+    p.add_argument("--learning-rate", type=float, default=0.0006)
+    p.add_argument("--seed", type=int, default=77)
+    args = p.parse_args()
+
+    from torched_impala_amd.config import Flags
+    from torched_impala_amd import learner
+
+    flags = Flags(env="synthetic-atari", num_actors=args.num_actors,
+                  envs_per_actor=args.envs_per_actor, batch_size=32,
+                  unroll_length=20, total_steps=1 << 60,
+                  actor_jit=args.actor_jit, use_lstm=args.use_lstm,
+                  graph_capture=not args.no_graphs,
+                  learning_rate=args.learning_rate,
+                  disable_checkpoint=True,
+                  device="cuda" if torch.cuda.is_available() else "cpu",
+                  seed=args.seed)
+
+    state = {"first_bad": None, "bad_key": None, "last_good": {}}
+
+    def on_step(ls, frames, stats):
+        vals = {k: float(v) for k, v in stats.items()}
+        bad = [k for k, v in vals.items() if not math.isfinite(v)]
+        if bad and state["first_bad"] is None:
+            state["first_bad"] = ls
+            state["bad_key"] = bad
+            print(json.dumps({"first_nonfinite_step": ls, "bad": bad,
+                              "stats": vals,
+                              "last_good": state["last_good"]}), flush=True)
+        if not bad:
+            state["last_good"] = vals
+        if ls % 200 == 0:
+            print(f"[probe] step {ls}: " +
+                  " ".join(f"{k}={v:.4g}" for k, v in vals.items()),
+                  flush=True)
+
+    out = learner.train(flags, max_learner_steps=args.steps, on_step=on_step)
+    print(json.dumps({
+        "probe": "nan", "steps": args.steps,
+        "first_nonfinite_step": state["first_bad"],
+        "bad_keys": state["bad_key"],
+        "final": {k: v for k, v in out.items()
+                  if isinstance(v, (int, float))},
+        "config": {"actor_jit": args.actor_jit, "lstm": args.use_lstm,
+                   "graphs": not args.no_graphs,
+                   "mfma": os.environ.get("IMPALA_SHALLOW_MFMA", "default"),
+                   "lr": args.learning_rate},
+    }), flush=True)
+
+
+if __name__ == "__main__":
+    main()

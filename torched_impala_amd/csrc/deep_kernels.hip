@@ -369,7 +369,7 @@ __global__ __launch_bounds__(THREADS, 4) void conv3x3_dx_nhwc(
 // the kernel is GEMM-bound.  dw_slab: [DW_NSLAB][cout*9*cin] zeroed
 // fp32; workgroup i accumulates into slab i % DW_NSLAB (caller sums).
 // ---------------------------------------------------------------------------
-constexpr int DW_NSLAB = 256;
+constexpr int DW_NSLAB = 64;   // DW_GRID writers / 64 slabs = low contention
 constexpr int DW_GRID = 1024;          // resident-WG target (256 CU x 2 x2)
 
 template <int CIN, int COUT, bool RELU_IN, int TTY = TY, int TTX = TX>
@@ -510,59 +510,94 @@ __global__ __launch_bounds__(THREADS, 2) void conv3x3_dw_nhwc(
 __global__ __launch_bounds__(256) void maxpool3x3s2_nhwc(
     const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
     unsigned char* __restrict__ argmax, int N, int H, int W, int C) {
+  // 8 channels per thread (NHWC is c-contiguous): bf16x8 window loads,
+  // one u64 argmax store — the scalar per-element version was
+  // index-idiv and latency bound (r12: 29 ms/step fwd, 79 ms bwd).
   const int HO = (H + 1) / 2, WO = (W + 1) / 2;
-  const size_t total = (size_t)N * HO * WO * C;
+  const int CG = C >> 3;
+  const size_t total = (size_t)N * HO * WO * CG;
   for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (size_t)gridDim.x * blockDim.x) {
-    const int c = i % C;
-    const size_t p = i / C;
-    const int ox = p % WO;
-    const int oy = (p / WO) % HO;
-    const int img = p / ((size_t)WO * HO);
+    const int cg = (int)(i % CG);
+    const size_t p = i / CG;
+    const int ox = (int)(p % WO);
+    const int oy = (int)((p / WO) % HO);
+    const int img = (int)(p / ((size_t)WO * HO));
     const int iy0 = oy * 2 - 1, ix0 = ox * 2 - 1;
-    float best = -3.4e38f;
-    int bidx = 0;
+    float best[8];
+    int bidx[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) { best[j] = -3.4e38f; bidx[j] = 0; }
+    #pragma unroll
     for (int dy = 0; dy < 3; ++dy) {
       const int iy = iy0 + dy;
       if (iy < 0 || iy >= H) continue;
+      #pragma unroll
       for (int dx = 0; dx < 3; ++dx) {
         const int ix = ix0 + dx;
         if (ix < 0 || ix >= W) continue;
-        const float v =
-            bf2f(x[(((size_t)img * H + iy) * W + ix) * C + c]);
-        if (v > best) { best = v; bidx = dy * 3 + dx; }
+        const bf16x8 v8 = *reinterpret_cast<const bf16x8*>(
+            x + (((size_t)img * H + iy) * W + ix) * C + cg * 8);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float v = bf2f(v8[j]);
+          if (v > best[j]) { best[j] = v; bidx[j] = dy * 3 + dx; }
+        }
       }
     }
-    y[i] = f2bf(best);
-    if (argmax != nullptr) argmax[i] = (unsigned char)bidx;
+    const size_t o = (((size_t)img * HO + oy) * WO + ox) * C + cg * 8;
+    bf16x8 yv;
+    unsigned long long am = 0;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      yv[j] = f2bf(best[j]);
+      am |= (unsigned long long)(unsigned char)bidx[j] << (8 * j);
+    }
+    *reinterpret_cast<bf16x8*>(y + o) = yv;
+    if (argmax != nullptr)
+      *reinterpret_cast<unsigned long long*>(argmax + o) = am;
   }
 }
 
 // Pool backward — deterministic gather: each input px scans the <=4
 // output windows that cover it and claims dY where argmax points here.
+// Same 8-channel vectorisation; argmax read as one u64 per window.
 __global__ __launch_bounds__(256) void maxpool3x3s2_bwd_nhwc(
     const bf16_t* __restrict__ dy, const unsigned char* __restrict__ argmax,
     bf16_t* __restrict__ dx, int N, int H, int W, int C) {
   const int HO = (H + 1) / 2, WO = (W + 1) / 2;
-  const size_t total = (size_t)N * H * W * C;
+  const int CG = C >> 3;
+  const size_t total = (size_t)N * H * W * CG;
   for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (size_t)gridDim.x * blockDim.x) {
-    const int c = i % C;
-    const size_t p = i / C;
-    const int ix = p % W;
-    const int iy = (p / W) % H;
-    const int img = p / ((size_t)W * H);
-    float g = 0.f;
-    // windows: oy*2-1 <= iy <= oy*2+1  =>  oy in [(iy-1+1)/2, (iy+1)/2]
+    const int cg = (int)(i % CG);
+    const size_t p = i / CG;
+    const int ix = (int)(p % W);
+    const int iy = (int)((p / W) % H);
+    const int img = (int)(p / ((size_t)W * H));
+    float g[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) g[j] = 0.f;
+    // windows: oy*2-1 <= iy <= oy*2+1  =>  oy in [iy/2, (iy+1)/2]
     const int oy_lo = (iy - 1 + 1) / 2, oy_hi = (iy + 1) / 2;
     const int ox_lo = (ix - 1 + 1) / 2, ox_hi = (ix + 1) / 2;
     for (int oy = oy_lo; oy <= oy_hi && oy < HO; ++oy)
       for (int ox = ox_lo; ox <= ox_hi && ox < WO; ++ox) {
-        const size_t o = (((size_t)img * HO + oy) * WO + ox) * C + c;
+        const size_t o = (((size_t)img * HO + oy) * WO + ox) * C + cg * 8;
         const int want = (iy - (oy * 2 - 1)) * 3 + (ix - (ox * 2 - 1));
-        if (argmax[o] == (unsigned char)want) g += bf2f(dy[o]);
+        const unsigned long long am =
+            *reinterpret_cast<const unsigned long long*>(argmax + o);
+        const bf16x8 d8 = *reinterpret_cast<const bf16x8*>(dy + o);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (((am >> (8 * j)) & 0xffull) == (unsigned)want)
+            g[j] += bf2f(d8[j]);
       }
-    dx[i] = f2bf(g);
+    bf16x8 out;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = f2bf(g[j]);
+    *reinterpret_cast<bf16x8*>(dx + (((size_t)img * H + iy) * W + ix) * C +
+                               cg * 8) = out;
   }
 }
 
@@ -732,7 +767,7 @@ extern "C" int launch_deep_conv3x3(const void* x, const void* w,
 extern "C" void launch_deep_maxpool(const void* x, void* y, void* argmax,
                                     int N, int H, int W, int C,
                                     hipStream_t s) {
-  const size_t total = (size_t)N * ((H + 1) / 2) * ((W + 1) / 2) * C;
+  const size_t total = (size_t)N * ((H + 1) / 2) * ((W + 1) / 2) * (C / 8);
   int blocks = (int)((total + 255) / 256);
   if (blocks > 16384) blocks = 16384;
   hipLaunchKernelGGL(maxpool3x3s2_nhwc, dim3(blocks), dim3(256), 0, s,
@@ -743,7 +778,7 @@ extern "C" void launch_deep_maxpool(const void* x, void* y, void* argmax,
 extern "C" void launch_deep_maxpool_bwd(const void* dy, const void* argmax,
                                         void* dx, int N, int H, int W, int C,
                                         hipStream_t s) {
-  const size_t total = (size_t)N * H * W * C;
+  const size_t total = (size_t)N * H * W * (C / 8);
   int blocks = (int)((total + 255) / 256);
   if (blocks > 16384) blocks = 16384;
   hipLaunchKernelGGL(maxpool3x3s2_bwd_nhwc, dim3(blocks), dim3(256), 0, s,
@@ -792,6 +827,20 @@ extern "C" int launch_deep_conv3x3_dx(const void* dy, const void* wt,
   else
     return -1;
   return 0;
+}
+
+// sum the DW_NSLAB slab copies into the final fp32 dW (identical
+// layout) — replaces an ATen column-reduce measured at ~300 us/call
+// (r12 deep profile: 31 ms/step of reduce_kernel over 13 conv layers).
+__global__ __launch_bounds__(256) void dw_slab_reduce(
+    const float* __restrict__ slab, float* __restrict__ out, int numel) {
+  const int o = blockIdx.x * 256 + threadIdx.x;
+  if (o >= numel) return;
+  float v = 0.f;
+  #pragma unroll 4
+  for (int sl = 0; sl < DW_NSLAB; ++sl)
+    v += slab[(size_t)sl * numel + o];
+  out[o] = v;
 }
 
 extern "C" int deep_dw_nslab() { return DW_NSLAB; }
@@ -872,3 +921,10 @@ extern "C" void launch_mfma_probe(const void* A, const void* Wt, float* D,
 }
 
 }  // namespace impala_deep
+
+extern "C" void launch_deep_dw_reduce(const float* slab, float* out,
+                                      int numel, hipStream_t s) {
+  hipLaunchKernelGGL(impala_deep::dw_slab_reduce,
+                     dim3((numel + 255) / 256), dim3(256), 0, s, slab, out,
+                     numel);
+}

@@ -67,6 +67,7 @@ int launch_deep_conv3x3_dx(const void*, const void*, const void*, void*,
 int launch_deep_conv3x3_dw(const void*, const void*, float*, int, int, int,
                            int, int, int, hipStream_t);
 int deep_dw_nslab();
+void launch_deep_dw_reduce(const float*, float*, int, hipStream_t);
 void launch_deep_u8_to_nhwc4(const void*, void*, int, int, int, hipStream_t);
 void launch_nchw_f32_to_nhwc_bf16(const void*, void*, int, int, int, int,
                                   hipStream_t);
@@ -475,6 +476,7 @@ torch::Tensor deep_conv3x3(torch::Tensor x, torch::Tensor w, torch::Tensor b,
 
 std::vector<torch::Tensor> deep_maxpool(torch::Tensor x, bool want_argmax) {
   check_bf16_nhwc(x, "x");
+  TORCH_CHECK(x.size(3) % 8 == 0, "maxpool kernels are 8-channel vectorised");
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   DeviceScope guard(x.get_device());
   auto y = torch::empty({N, (H + 1) / 2, (W + 1) / 2, C}, x.options());
@@ -547,8 +549,12 @@ torch::Tensor deep_conv3x3_dw(torch::Tensor x, torch::Tensor dy,
                                   slab.data_ptr<float>(), N, H, W, cin,
                                   cout, relu_in ? 1 : 0, current_stream());
   TORCH_CHECK(rc == 0, "unsupported dw cin/cout ", cin, "/", cout);
+  auto out = torch::empty({cout, 3, 3, cin},
+                          x.options().dtype(torch::kFloat32));
+  launch_deep_dw_reduce(slab.data_ptr<float>(), out.data_ptr<float>(),
+                        cout * 9 * cin, current_stream());
   HIP_CHECK(hipGetLastError());
-  return slab.sum(0);
+  return out;
 }
 
 // u8 [N][3][H][W] -> bf16 [N][H][W][4] with /255 and a zero pad channel

@@ -113,3 +113,73 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def stress(args):
+    """--stress mode: graphed-only replay with FRESH RANDOM batch data
+    written into the fixed staging buffers before every replay (the real
+    Batcher's pattern), lr=0 so weights must stay bitwise-frozen; any
+    non-finite loss/param is memory corruption in the replayed step."""
+    device = torch.device("cuda", 0)
+    obs, A = (4, 84, 84), 6
+    flags = Flags(env="synthetic-atari", batch_size=32, unroll_length=20,
+                  use_lstm=args.use_lstm, total_steps=1 << 50,
+                  learning_rate=0.0)
+    torch.manual_seed(42)
+    model = build_model(flags, obs, A).to(device)
+    opt, sched = L.build_optimizer(flags, model)
+    p0 = opt.flat_param.clone()
+    st = staging(flags, obs, A, device, seed=7)
+    stub = Stub(flags, st)
+    graphed = L.GraphedStep(flags, model, opt, stub)
+    g = torch.Generator(device="cuda").manual_seed(123)
+    T = flags.unroll_length
+    for step in range(args.steps):
+        # refresh the staging buffers in place (what Batcher's H2D does)
+        st["frame"].random_(0, 256, generator=g)
+        st["reward"].normal_(generator=g)
+        st["done"].bernoulli_(0.05, generator=g)
+        st["action"].random_(0, A, generator=g)
+        st["last_action"].random_(0, A, generator=g)
+        st["behavior_logits"].normal_(generator=g)
+        stats = graphed.step(0, sched)
+        if step % 100 == 0 or step == args.steps - 1:
+            torch.cuda.synchronize()
+            loss = float(stats["total_loss"])
+            pfin = bool(torch.isfinite(opt.flat_param).all())
+            gfin = bool(torch.isfinite(opt.flat_grad).all())
+            pmax = float((opt.flat_param - p0).abs().max())
+            print(f"[stress] step {step}: loss={loss:.4g} param_fin={pfin} "
+                  f"grad_fin={gfin} param_drift={pmax:.3e}", flush=True)
+            if not math.isfinite(loss) or not pfin or pmax > 0:
+                # drill: which grad slices are bad
+                per = []
+                for n_, p_ in model.named_parameters():
+                    off, num = opt.slices[p_]
+                    gsl = opt.flat_grad[off:off + num]
+                    psl = opt.flat_param[off:off + num]
+                    per.append(dict(
+                        name=n_, g_fin=bool(torch.isfinite(gsl).all()),
+                        p_fin=bool(torch.isfinite(psl).all()),
+                        g_absmax=float(gsl.abs().max()),
+                        p_drift=float((psl - p0[off:off + num]).abs().max())))
+                print(json.dumps({"corrupt_at": step, "per_param": per}),
+                      flush=True)
+                sys.exit(2)
+    print(json.dumps({"stress": "clean", "steps": args.steps}))
+
+
+_orig_main = main
+
+
+def main():  # noqa: F811
+    import argparse as _ap
+    pre = _ap.ArgumentParser(add_help=False)
+    pre.add_argument("--stress", action="store_true")
+    pre.add_argument("--steps", type=int, default=400)
+    pre.add_argument("--use-lstm", action="store_true")
+    known, _ = pre.parse_known_args()
+    if known.stress:
+        stress(known)
+    else:
+        _orig_main()

@@ -111,10 +111,6 @@ def main():
     print(json.dumps({"drift_probe": "clean", "steps": args.steps}))
 
 
-if __name__ == "__main__":
-    main()
-
-
 def stress(args):
     """--stress mode: graphed-only replay with FRESH RANDOM batch data
     written into the fixed staging buffers before every replay (the real
@@ -183,3 +179,7 @@ def main():  # noqa: F811
         stress(known)
     else:
         _orig_main()
+
+
+if __name__ == "__main__":
+    main()

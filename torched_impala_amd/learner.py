@@ -81,7 +81,7 @@ class Batcher:
             self.staging = []
             self.consumed_ev = []
             for _ in range(2):
-                st = {key: torch.empty((self.k, *a.shape[1:]), dtype=a.dtype,
+                st = {key: torch.zeros((self.k, *a.shape[1:]), dtype=a.dtype,
                                        device=device)
                       for key, a in buffers.arenas.items()}
                 self.staging.append(st)
@@ -328,10 +328,31 @@ class GraphedStep:
         self.outputs.clear()
         self.pool = None
 
+    def ensure_captured(self):
+        """Capture graphs for ALL staging buffers up-front, BEFORE any
+        replay: a capture that shares a mempool with an already-replayed
+        graph may recycle blocks that graph's replays still read/write
+        (the torch pool-sharing contract is capture-everything-first).
+        The lazy per-buffer capture interleaved replays between captures
+        and corrupted pool memory data-dependently — NaN losses at
+        ~1-2k steps in the r12/r13 soaks, MFMA+graphs only (more live
+        pool tensors), and plausibly round-1's flaky async
+        hipErrorInvalidValue. IMPALA_GRAPH_LAZY_CAPTURE=1 restores the
+        old behavior for A/B."""
+        st = self.batcher.staging
+        idxs = sorted(st.keys() if isinstance(st, dict) else
+                      range(len(st)))
+        for j in idxs:
+            if j not in self.graphs:
+                self._capture(j)
+
     def step(self, i: int, scheduler) -> Dict[str, torch.Tensor]:
         self.optimizer.sync_lr()
         if i not in self.graphs:
-            self._capture(i)
+            if os.environ.get("IMPALA_GRAPH_LAZY_CAPTURE", "0") == "1":
+                self._capture(i)
+            else:
+                self.ensure_captured()
         self.graphs[i].replay()
         if self.world > 1:
             # SUM; the 1/world mean is the optimizer's device-side scale
@@ -523,7 +544,7 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
                 # the job. One MIN-reduce of the capture outcome keeps
                 # every rank on the same path.
                 try:
-                    graphed._capture(si)
+                    graphed.ensure_captured()
                     ok = 1.0
                 except Exception:
                     log.exception("graph capture failed on this rank")

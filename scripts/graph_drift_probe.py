@@ -115,7 +115,10 @@ def stress(args):
     """--stress mode: graphed-only replay with FRESH RANDOM batch data
     written into the fixed staging buffers before every replay (the real
     Batcher's pattern), lr=0 so weights must stay bitwise-frozen; any
-    non-finite loss/param is memory corruption in the replayed step."""
+    non-finite loss/param is memory corruption in the replayed step.
+    --buffers 2 reproduces train()'s double-buffer alternation (capture
+    g0 -> replay -> capture g1 under IMPALA_GRAPH_LAZY_CAPTURE=1 — the
+    shared-pool contract violation)."""
     device = torch.device("cuda", 0)
     obs, A = (4, 84, 84), 6
     flags = Flags(env="synthetic-atari", batch_size=32, unroll_length=20,
@@ -125,12 +128,15 @@ def stress(args):
     model = build_model(flags, obs, A).to(device)
     opt, sched = L.build_optimizer(flags, model)
     p0 = opt.flat_param.clone()
-    st = staging(flags, obs, A, device, seed=7)
-    stub = Stub(flags, st)
+    st0 = staging(flags, obs, A, device, seed=7)
+    stub = Stub(flags, st0)
+    if args.buffers == 2:
+        stub.staging[1] = staging(flags, obs, A, device, seed=8)
     graphed = L.GraphedStep(flags, model, opt, stub)
     g = torch.Generator(device="cuda").manual_seed(123)
     T = flags.unroll_length
     for step in range(args.steps):
+        st = stub.staging[step % args.buffers]
         # refresh the staging buffers in place (what Batcher's H2D does)
         st["frame"].random_(0, 256, generator=g)
         st["reward"].normal_(generator=g)
@@ -138,7 +144,7 @@ def stress(args):
         st["action"].random_(0, A, generator=g)
         st["last_action"].random_(0, A, generator=g)
         st["behavior_logits"].normal_(generator=g)
-        stats = graphed.step(0, sched)
+        stats = graphed.step(step % args.buffers, sched)
         if step % 100 == 0 or step == args.steps - 1:
             torch.cuda.synchronize()
             loss = float(stats["total_loss"])
@@ -174,6 +180,7 @@ def main():  # noqa: F811
     pre.add_argument("--stress", action="store_true")
     pre.add_argument("--steps", type=int, default=400)
     pre.add_argument("--use-lstm", action="store_true")
+    pre.add_argument("--buffers", type=int, default=1, choices=[1, 2])
     known, _ = pre.parse_known_args()
     if known.stress:
         stress(known)

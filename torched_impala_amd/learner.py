@@ -521,6 +521,12 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
         if world > 1:
             torch.distributed.barrier()
 
+    # debug: IMPALA_DUMP_NONFINITE=<path> saves the batch + optimizer
+    # state the first time total_loss goes non-finite (costs one host
+    # sync per step; leave unset in production)
+    _dump_nonfinite = os.environ.get("IMPALA_DUMP_NONFINITE")
+    _dumped = False
+
     frames_per_batch = flags.unroll_length * flags.batch_size
     step_frames = resume_frames
     learner_steps = 0
@@ -577,6 +583,31 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
             batcher.mark_consumed(batch)
             learner_steps += 1
             step_frames += frames_per_batch * world
+
+            if _dump_nonfinite and not _dumped:
+                tl = float(stats["total_loss"])
+                if tl != tl or tl == float("inf") or tl == float("-inf"):
+                    _dumped = True
+                    blob = dict(step=learner_steps,
+                                flat_param=optimizer.flat_param.cpu()
+                                if isinstance(optimizer, FlatRMSprop)
+                                else None,
+                                flat_grad=optimizer.flat_grad.cpu()
+                                if isinstance(optimizer, FlatRMSprop)
+                                else None,
+                                flat_sqavg=optimizer.flat_sqavg.cpu()
+                                if isinstance(optimizer, FlatRMSprop)
+                                else None,
+                                stats={k: float(v)
+                                       for k, v in stats.items()})
+                    if si is not None and batcher.is_cuda:
+                        blob["staging"] = {
+                            k: v.cpu() for k, v in
+                            batcher.staging[si].items()}
+                    torch.save(blob, _dump_nonfinite)
+                    log.error("non-finite loss at learner step %d — "
+                              "dumped state to %s", learner_steps,
+                              _dump_nonfinite)
 
             if learner_steps % flags.weight_sync_interval == 0:
                 timers.start("weight_publish")

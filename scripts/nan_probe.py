@@ -31,6 +31,7 @@ def main():
     p.add_argument("--no-graphs", action="store_true")
     p.add_argument("--learning-rate", type=float, default=0.0006)
     p.add_argument("--seed", type=int, default=77)
+    p.add_argument("--episode-len", type=int, default=1000)
     args = p.parse_args()
 
     from torched_impala_amd.config import Flags
@@ -43,6 +44,7 @@ def main():
                   graph_capture=not args.no_graphs,
                   learning_rate=args.learning_rate,
                   disable_checkpoint=True,
+                  synthetic_episode_len=args.episode_len,
                   device="cuda" if torch.cuda.is_available() else "cpu",
                   seed=args.seed)
 

@@ -32,6 +32,11 @@ def main():
     p.add_argument("--learning-rate", type=float, default=0.0006)
     p.add_argument("--seed", type=int, default=77)
     p.add_argument("--episode-len", type=int, default=1000)
+    p.add_argument("--hunt", default=None,
+                   help="path prefix: device-side loss ring + rolling "
+                        "staging-clone window, sparse finiteness checks "
+                        "(no per-step host sync), dump the poisoned "
+                        "batch on detection")
     args = p.parse_args()
 
     from torched_impala_amd.config import Flags
@@ -49,24 +54,76 @@ def main():
                   seed=args.seed)
 
     state = {"first_bad": None, "bad_key": None, "last_good": {}}
+    ctx = None
 
-    def on_step(ls, frames, stats):
-        vals = {k: float(v) for k, v in stats.items()}
-        bad = [k for k, v in vals.items() if not math.isfinite(v)]
-        if bad and state["first_bad"] is None:
-            state["first_bad"] = ls
-            state["bad_key"] = bad
-            print(json.dumps({"first_nonfinite_step": ls, "bad": bad,
-                              "stats": vals,
-                              "last_good": state["last_good"]}), flush=True)
-        if not bad:
-            state["last_good"] = vals
-        if ls % 200 == 0:
-            print(f"[probe] step {ls}: " +
-                  " ".join(f"{k}={v:.4g}" for k, v in vals.items()),
-                  flush=True)
+    if args.hunt:
+        import types
+        ctx = types.SimpleNamespace()
+        RING = 8192
+        WINDOW = 12
+        hunt = {"ring": None, "clones": [], "done": False}
 
-    out = learner.train(flags, max_learner_steps=args.steps, on_step=on_step)
+        def on_step(ls, frames, stats):
+            # NO host sync in the hot path: device-side records only
+            if hunt["ring"] is None:
+                dev = stats["total_loss"].device
+                hunt["ring"] = torch.zeros(RING, device=dev)
+                st0 = ctx.batcher.staging[0]
+                for _ in range(WINDOW):
+                    hunt["clones"].append(
+                        {k: torch.empty_like(v) for k, v in st0.items()})
+            hunt["ring"][ls % RING].copy_(stats["total_loss"].reshape(()))
+            si = (ls - 1) % 2
+            cl = hunt["clones"][ls % WINDOW]
+            for k, v in ctx.batcher.staging[si].items():
+                cl[k].copy_(v)
+            if ls % 49 == 0 and not hunt["done"]:
+                lo = max(1, ls - 48)
+                seg = hunt["ring"][lo:ls + 1]  # RING > max steps: no wrap
+                if not bool(torch.isfinite(seg).all()):
+                    torch.cuda.synchronize()
+                    vals = hunt["ring"][:ls + 1].cpu()
+                    bad_steps = [i for i in range(lo, ls + 1)
+                                 if not math.isfinite(float(vals[i]))]
+                    first = bad_steps[0]
+                    blob = dict(first_bad=first, checked_at=ls,
+                                ring=vals,
+                                flat_param=ctx.optimizer.flat_param.cpu(),
+                                flat_grad=ctx.optimizer.flat_grad.cpu(),
+                                flat_sqavg=ctx.optimizer.flat_sqavg.cpu(),
+                                window={})
+                    for w in range(max(first - 3, lo), min(first + 2, ls + 1)):
+                        blob["window"][w] = {
+                            k: v.cpu() for k, v in
+                            hunt["clones"][w % WINDOW].items()}
+                    torch.save(blob, args.hunt)
+                    print(json.dumps({"hunt_first_bad": first,
+                                      "checked_at": ls,
+                                      "dump": args.hunt}), flush=True)
+                    hunt["done"] = True
+                    state["first_bad"] = first
+            if ls % 500 == 0:
+                print(f"[hunt] step {ls}", flush=True)
+    else:
+        def on_step(ls, frames, stats):
+            vals = {k: float(v) for k, v in stats.items()}
+            bad = [k for k, v in vals.items() if not math.isfinite(v)]
+            if bad and state["first_bad"] is None:
+                state["first_bad"] = ls
+                state["bad_key"] = bad
+                print(json.dumps({"first_nonfinite_step": ls, "bad": bad,
+                                  "stats": vals,
+                                  "last_good": state["last_good"]}),
+                      flush=True)
+            if not bad:
+                state["last_good"] = vals
+            if ls % 200 == 0:
+                print(f"[probe] step {ls}: " +
+                      " ".join(f"{k}={v:.4g}" for k, v in vals.items()),
+                      flush=True)
+
+    out = learner.train(flags, max_learner_steps=args.steps,
+                        on_step=on_step, debug_ctx=ctx)
     print(json.dumps({
         "probe": "nan", "steps": args.steps,
         "first_nonfinite_step": state["first_bad"],

@@ -82,3 +82,27 @@ def test_flat_rmsprop_grad_views():
     assert opt.flat_grad.abs().sum() == 0
     for p in m.parameters():
         assert p.grad.abs().sum() == 0
+
+
+def test_nonfinite_grad_skips_update():
+    """One inf/NaN gradient element must SKIP the update entirely
+    (params and sqavg untouched, skip counter bumped) — a poisoned
+    gradient otherwise NaNs the parameters permanently (0*inf=NaN even
+    at lr=0)."""
+    import torch
+    from torched_impala_amd.ops.optim import FlatRMSprop
+    p = torch.nn.Parameter(torch.randn(64))
+    opt = FlatRMSprop([p], lr=0.01, max_grad_norm=40.0)
+    p0 = opt.flat_param.clone()
+    sq0 = opt.flat_sqavg.clone()
+    opt.flat_grad.normal_()
+    opt.flat_grad[7] = float("inf")
+    opt.step()
+    assert float(opt.skip_t) == 1.0
+    torch.testing.assert_close(opt.flat_param, p0, rtol=0, atol=0)
+    torch.testing.assert_close(opt.flat_sqavg, sq0, rtol=0, atol=0)
+    # a clean grad afterwards applies normally
+    opt.flat_grad.normal_()
+    opt.step()
+    assert float(opt.skip_t) == 1.0
+    assert not torch.equal(opt.flat_param, p0)

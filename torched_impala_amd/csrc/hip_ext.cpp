@@ -42,7 +42,7 @@ void launch_grad_sqnorm(const float*, long, const float*, float*,
                         hipStream_t);
 void launch_rmsprop_update(float*, const float*, float*, long, const float*,
                            const float*, const float*, float, float, float,
-                           hipStream_t);
+                           float*, hipStream_t);
 void launch_shallow_encoder_fwd(const unsigned char*, const float*,
                                 const float*, const float*, const float*,
                                 const float*, const float*, int, float*,
@@ -317,7 +317,13 @@ std::vector<torch::Tensor> lstm_seq_bwd(
 void fused_clip_rmsprop(torch::Tensor flat_p, torch::Tensor flat_g,
                         torch::Tensor flat_sqavg, torch::Tensor lr,
                         torch::Tensor norm_sq, torch::Tensor scale,
-                        double max_norm, double alpha, double eps) {
+                        double max_norm, double alpha, double eps,
+                        c10::optional<torch::Tensor> skip_count) {
+  float* skip_p = nullptr;
+  if (skip_count.has_value()) {
+    check_f32(*skip_count, "skip_count");
+    skip_p = skip_count->data_ptr<float>();
+  }
   check_f32(flat_p, "flat_p");
   check_f32(flat_g, "flat_g");
   check_f32(flat_sqavg, "flat_sqavg");
@@ -335,7 +341,7 @@ void fused_clip_rmsprop(torch::Tensor flat_p, torch::Tensor flat_g,
                         flat_sqavg.data_ptr<float>(), n,
                         lr.data_ptr<float>(), norm_sq.data_ptr<float>(),
                         scale.data_ptr<float>(), (float)max_norm,
-                        (float)alpha, (float)eps, stream);
+                        (float)alpha, (float)eps, skip_p, stream);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -696,7 +702,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vtrace_loss_bwd", &vtrace_loss_bwd);
   m.def("lstm_seq_fwd", &lstm_seq_fwd);
   m.def("lstm_seq_bwd", &lstm_seq_bwd);
-  m.def("fused_clip_rmsprop", &fused_clip_rmsprop);
+  m.def("fused_clip_rmsprop", &fused_clip_rmsprop, py::arg("flat_p"),
+        py::arg("flat_g"), py::arg("flat_sqavg"), py::arg("lr"),
+        py::arg("norm_sq"), py::arg("scale"), py::arg("max_norm"),
+        py::arg("alpha"), py::arg("eps"),
+        py::arg("skip_count") = py::none());
   m.def("shallow_encoder_fwd", &shallow_encoder_fwd);
   m.def("shallow_encoder_bwd", &shallow_encoder_bwd);
   m.def("linear_mfma", &linear_mfma, py::arg("A"), py::arg("B"),

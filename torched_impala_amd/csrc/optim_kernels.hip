@@ -45,8 +45,19 @@ __global__ void rmsprop_update_kernel(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ sqavg, long n, const float* __restrict__ lr,
     const float* __restrict__ norm_sq, const float* __restrict__ scale,
-    float max_norm, float alpha, float eps) {
+    float max_norm, float alpha, float eps,
+    float* __restrict__ skip_count) {
   const float s = scale[0];
+  // non-finite global norm => at least one grad element is inf/NaN:
+  // SKIP the whole update (params and sqavg untouched) instead of
+  // poisoning the parameters permanently (0 * inf = NaN even at lr=0).
+  // Standard overflow-skip semantics (GradScaler-style); skips are
+  // counted in skip_count[0] for observability.
+  if (!isfinite(norm_sq[0])) {
+    if (blockIdx.x == 0 && threadIdx.x == 0 && skip_count != nullptr)
+      skip_count[0] += 1.f;
+    return;
+  }
   float coef = 1.f;
   if (max_norm > 0.f) {
     const float norm = sqrtf(norm_sq[0]);
@@ -79,13 +90,14 @@ void launch_grad_sqnorm(const float* g, long n, const float* scale,
 void launch_rmsprop_update(float* p, const float* g, float* sqavg, long n,
                            const float* lr, const float* norm_sq,
                            const float* scale, float max_norm, float alpha,
-                           float eps, hipStream_t stream) {
+                           float eps, float* skip_count,
+                           hipStream_t stream) {
   const int block = 256;
   long grid = (n + block - 1) / block;
   if (grid > 2048) grid = 2048;
   hipLaunchKernelGGL(impala::rmsprop_update_kernel, dim3((int)grid),
                      dim3(block), 0, stream, p, g, sqavg, n, lr, norm_sq,
-                     scale, max_norm, alpha, eps);
+                     scale, max_norm, alpha, eps, skip_count);
 }
 
 }  // extern "C"

@@ -399,7 +399,7 @@ def start_actors(flags: Flags, ctx, buffers, free_queue, full_queue, mirror,
 
 
 def train(flags: Flags, max_learner_steps: Optional[int] = None,
-          on_step=None) -> Dict[str, float]:
+          on_step=None, debug_ctx=None) -> Dict[str, float]:
     """Full actor-learner training. Returns final stats."""
     device = torch.device(flags.run_device)
     rank, world = init_distributed(device.type)
@@ -461,6 +461,12 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
                    else GradReducer(model))
     metrics = MetricsLogger(flags.metrics_jsonl)
     timers = StageTimers()
+    if debug_ctx is not None:
+        # diagnostic harnesses (scripts/nan_probe.py --hunt) get direct
+        # handles; never used in production paths
+        debug_ctx.batcher = batcher
+        debug_ctx.optimizer = optimizer
+        debug_ctx.model = model
 
     run_dir = None
     ckpt_path = None
@@ -692,6 +698,8 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
     # fallback must never be reported as the graph-path number)
     out["graph_path_active"] = bool(graphed is not None) if use_graphs \
         else False
+    if isinstance(optimizer, FlatRMSprop):
+        out["grad_skip_count"] = float(optimizer.skip_t.item())
     out.update(step_frames=step_frames, learner_steps=learner_steps,
                mean_return=metrics.mean_return or float("nan"),
                fps=step_frames / max(time.time() - t_start, 1e-9),

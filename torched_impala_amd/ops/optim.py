@@ -56,6 +56,11 @@ class FlatRMSprop(torch.optim.Optimizer):
                 off += n
         self.lr_t = torch.zeros(1, device=device)
         self.norm_sq = torch.zeros(1, device=device)
+        # non-finite-gradient skip counter (device; graph-capturable):
+        # the fused kernel refuses to apply an update whose global norm
+        # is inf/NaN — one poisoned gradient otherwise NaNs the params
+        # permanently (0*inf=NaN even at lr=0)
+        self.skip_t = torch.zeros(1, device=device)
         self.scale_t = torch.full((1,), 1.0 / world_size, device=device)
         self._steps = 0
 
@@ -97,12 +102,15 @@ class FlatRMSprop(torch.optim.Optimizer):
             ops.ext_hip().fused_clip_rmsprop(
                 self.flat_param, self.flat_grad, self.flat_sqavg,
                 self.lr_t, self.norm_sq, self.scale_t,
-                self.max_grad_norm, alpha, eps)
+                self.max_grad_norm, alpha, eps, self.skip_t)
             return
         # eager reference path (CPU tests / escape hatch) — same math
         gr = self.flat_grad * self.scale_t
         if self.max_grad_norm > 0:
             norm = gr.norm()
+            if not torch.isfinite(norm):
+                self.skip_t += 1  # skip the poisoned update (see above)
+                return
             coef = torch.clamp(self.max_grad_norm / (norm + 1e-6), max=1.0)
             gr = gr * coef
         self.flat_sqavg.mul_(alpha).addcmul_(gr, gr, value=1.0 - alpha)

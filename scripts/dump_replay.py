@@ -40,7 +40,9 @@ def finite_report(name, t):
 def main():
     path = sys.argv[1]
     blob = torch.load(path, map_location="cpu", weights_only=False)
-    print("dump step:", blob["step"], "stats:", blob["stats"])
+    print("dump step:", blob.get("step"), "stats:", blob.get("stats"))
+    if "per_param" in blob:
+        print("bad params:", blob["per_param"])
     print("== dumped tensors ==")
     ok_batch = True
     for k, v in (blob.get("staging") or {}).items():

@@ -32,6 +32,10 @@ def main():
     p.add_argument("--learning-rate", type=float, default=0.0006)
     p.add_argument("--seed", type=int, default=77)
     p.add_argument("--episode-len", type=int, default=1000)
+    p.add_argument("--gradwatch", default=None,
+                   help="path: on the FIRST skipped update (non-finite "
+                        "grad), report per-parameter grad finiteness and "
+                        "dump the batch + grads for dump_replay")
     p.add_argument("--hunt", default=None,
                    help="path prefix: device-side loss ring + rolling "
                         "staging-clone window, sparse finiteness checks "
@@ -56,7 +60,43 @@ def main():
     state = {"first_bad": None, "bad_key": None, "last_good": {}}
     ctx = None
 
-    if args.hunt:
+    if args.gradwatch:
+        import types
+        ctx = types.SimpleNamespace()
+        gw = {"last_skips": 0.0, "done": False}
+
+        def on_step(ls, frames, stats):
+            if gw["done"]:
+                return
+            skips = float(ctx.optimizer.skip_t.item())
+            if skips > gw["last_skips"]:
+                torch.cuda.synchronize()
+                fg = ctx.optimizer.flat_grad
+                per = []
+                for n_, p_ in ctx.model.named_parameters():
+                    off, num = ctx.optimizer.slices[p_]
+                    sl = fg[off:off + num]
+                    nf = int((~torch.isfinite(sl)).sum())
+                    if nf:
+                        per.append(dict(name=n_, nonfinite=nf,
+                                        numel=num))
+                si = (ls - 1) % 2
+                blob = dict(step=ls, skips=skips,
+                            per_param=per,
+                            flat_param=ctx.optimizer.flat_param.cpu(),
+                            flat_grad=fg.cpu(),
+                            staging={k: v.cpu() for k, v in
+                                     ctx.batcher.staging[si].items()},
+                            stats={k: float(v) for k, v in stats.items()})
+                torch.save(blob, args.gradwatch)
+                print(json.dumps({"gradwatch_step": ls, "skips": skips,
+                                  "bad_params": per}), flush=True)
+                gw["done"] = True
+                state["first_bad"] = ls
+            gw["last_skips"] = skips
+            if ls % 500 == 0:
+                print(f"[gradwatch] step {ls} skips={skips}", flush=True)
+    elif args.hunt:
         import types
         ctx = types.SimpleNamespace()
         RING = 8192

@@ -72,6 +72,16 @@ def main():
             if skips > gw["last_skips"]:
                 torch.cuda.synchronize()
                 fg = ctx.optimizer.flat_grad
+                try:
+                    from torched_impala_amd.ops.shallow_mfma import STASH
+                    chain = {k: [int((~torch.isfinite(v.float())).sum()),
+                                 float(v.float()[torch.isfinite(v.float())]
+                                       .abs().max())]
+                             for k, v in STASH.items()}
+                    print(json.dumps({"chain_nonfinite": chain}),
+                          flush=True)
+                except Exception as e:  # pragma: no cover
+                    print("stash failed:", e, flush=True)
                 per = []
                 for n_, p_ in ctx.model.named_parameters():
                     off, num = ctx.optimizer.slices[p_]

@@ -25,6 +25,16 @@ import torch
 
 from . import ext_hip
 
+# IMPALA_MFMA_STASH=1: keep references to the last call's intermediates
+# for post-hoc finiteness forensics (scripts/nan_probe.py --gradwatch).
+# Refs only — on the graph path these alias pool tensors, valid until
+# the next replay.
+STASH: dict = {}
+
+
+def _stash_enabled() -> bool:
+    return os.environ.get("IMPALA_MFMA_STASH", "0") == "1"
+
 
 def shallow_mfma_enabled() -> bool:
     """Default ON since round 2: hardware parity 9/9 PASS (r2 probe) and
@@ -48,6 +58,9 @@ class _ShallowMfmaEncode(torch.autograd.Function):
             w2p, b2.detach().float().contiguous(), w3p,
             b3.detach().float().contiguous())
         ctx.save_for_backward(frames, a1, a2, a3, w2, w3)
+        if _stash_enabled():
+            STASH.update(frames=frames, a1=a1, a2=a2, a3=a3,
+                         w1p=w1p, w2p=w2p, w3p=w3p)
         return a3
 
     @staticmethod
@@ -66,6 +79,10 @@ class _ShallowMfmaEncode(torch.autograd.Function):
         wr2 = torch.stack(cls).to(bf).contiguous()
         (da1, da2, dw1, dw2, dw3, db1, db2, db3) = ext.shallow_mfma_bwd(
             frames.contiguous(), a1, a2, dz3, wr3, wr2)
+        if _stash_enabled():
+            STASH.update(d_a3=d_a3, dz3=dz3, wr3=wr3, wr2=wr2, da1=da1,
+                         da2=da2, dw1=dw1, dw2=dw2, dw3=dw3, db1=db1,
+                         db2=db2, db3=db3)
         # weight grads arrive already in torch layouts and the bias
         # column sums already reduced (fused bwd_finalize kernel)
         return None, dw1, db1, dw2, db2, dw3, db3

@@ -129,15 +129,25 @@ class Batcher:
         self.idx = (self.idx + 1) % 2
         slots = self._pop_slots()
         st = self.staging[i]
-        with torch.cuda.stream(self.copy_stream):
-            self.copy_stream.wait_event(self.consumed_ev[i])
+        if os.environ.get("IMPALA_NO_COPY_OVERLAP", "0") == "1":
+            # diagnostic: H2D on the compute stream (no side-stream
+            # overlap) — race-isolation A/B for the r12-r19 forensics
             for key, arena in self.buffers.arenas.items():
                 dst = st[key]
                 for j, m in enumerate(slots):
                     dst[j].copy_(arena[m], non_blocking=True)
             copy_done = torch.cuda.Event()
-            copy_done.record(self.copy_stream)
-        torch.cuda.current_stream().wait_event(copy_done)
+            copy_done.record()
+        else:
+            with torch.cuda.stream(self.copy_stream):
+                self.copy_stream.wait_event(self.consumed_ev[i])
+                for key, arena in self.buffers.arenas.items():
+                    dst = st[key]
+                    for j, m in enumerate(slots):
+                        dst[j].copy_(arena[m], non_blocking=True)
+                copy_done = torch.cuda.Event()
+                copy_done.record(self.copy_stream)
+            torch.cuda.current_stream().wait_event(copy_done)
         self._pending.append((copy_done, slots))
         batch = self._to_time_major(st) if materialize else {}
         batch["_staging_index"] = i
@@ -617,7 +627,8 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
 
             if learner_steps % flags.weight_sync_interval == 0:
                 timers.start("weight_publish")
-                if use_flat_publish:
+                if use_flat_publish and os.environ.get(
+                        "IMPALA_NO_ASYNC_PUBLISH", "0") != "1":
                     # one D2D snapshot + one async D2H; seqlock flips on
                     # completion (poll) — host cost is an event query
                     mirror.publish_flat_async(optimizer.flat_param)

@@ -99,12 +99,19 @@ def shallow_mfma_encode(net, frames_u8: torch.Tensor) -> torch.Tensor:
 
 def shallow_mfma_features(net, frames_u8: torch.Tensor) -> torch.Tensor:
     """Full encode incl. the fc: the conv stack on the MFMA kernels, the
-    fc as a rocBLAS bf16 GEMM through plain autograd (weight grads land
-    on the fp32 masters via the tracked .to(bf16) casts)."""
-    bf = torch.bfloat16
+    fc on the custom fp32 split-K MFMA GEMM (ops/linear.py).
+
+    IMPALA_MFMA_FC=rocblas switches the fc to a rocBLAS bf16 GEMM via
+    tracked .to(bf16) casts — that was the round-2 default until the
+    r19/r21 forensics traced transient garbage fc-grads under hipGraph
+    replay to the hipBLASLt-bf16 path (see ROADMAP: NaN investigation)."""
     a3 = _ShallowMfmaEncode.apply(
         frames_u8, net.conv1.weight, net.conv1.bias, net.conv2.weight,
         net.conv2.bias, net.conv3.weight, net.conv3.bias)
-    y = torch.nn.functional.linear(a3, net.fc.weight.to(bf),
-                                   net.fc.bias.to(bf))
-    return torch.relu(y).float()
+    if os.environ.get("IMPALA_MFMA_FC", "mfma") == "rocblas":
+        bf = torch.bfloat16
+        y = torch.nn.functional.linear(a3, net.fc.weight.to(bf),
+                                       net.fc.bias.to(bf))
+        return torch.relu(y).float()
+    from torched_impala_amd.ops.linear import linear_relu
+    return linear_relu(net.fc, a3.float())

@@ -411,6 +411,28 @@ def test_linear_mfma_splitk_parity(M, N, K, mode):
     torch.testing.assert_close(C, ref, rtol=2e-4, atol=2e-3)
 
 
+@pytest.mark.parametrize("M,N,K,mode,outb", [(640, 256, 1568, 1, False),
+                                             (640, 1568, 256, 0, True),
+                                             (256, 1568, 640, 0, False),
+                                             (333, 257, 100, 1, False)])
+def test_linear_bf16_parity(M, N, K, mode, outb):
+    """In-tree bf16 GEMM (fwd/dX/dW shapes of the fc) vs fp32 reference
+    at bf16 tolerances; covers split-K and bf16-out paths."""
+    from torched_impala_amd import ops
+    dev = _dev()
+    torch.manual_seed(4)
+    A = torch.randn(M, K, device=dev).to(torch.bfloat16)
+    B = torch.randn(N, K, device=dev).to(torch.bfloat16)
+    bias = torch.randn(N, device=dev)
+    C = ops.ext_hip().linear_bf16(A, B, bias, mode, outb)
+    assert C.dtype == (torch.bfloat16 if outb else torch.float32)
+    ref = A.float() @ B.float().t() + bias
+    if mode == 1:
+        ref = torch.relu(ref)
+    torch.testing.assert_close(C.float(), ref, rtol=5e-2,
+                               atol=0.05 * ref.abs().max().item() + 1e-3)
+
+
 @pytest.mark.parametrize("clip_rho,clip_c,lam", [(0.8, 1.2, 0.95),
                                                  (2.0, 0.7, 0.5)])
 def test_vtrace_scan_parity_nondefault_clips(clip_rho, clip_c, lam):

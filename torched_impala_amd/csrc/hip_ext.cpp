@@ -53,6 +53,10 @@ void launch_shallow_encoder_bwd(const unsigned char*, const float*,
                                 float*, float*, hipStream_t);
 int shallow_encoder_slab_elems();
 int linear_splitk_factor(int, int, int);
+int linear_splitk_factor_bf16(int, int, int);
+void launch_linear_mfma_bf16(const void*, const void*, const float*, int,
+                             int, int, int, int, void*, float*, int,
+                             hipStream_t);
 void launch_linear_mfma(const float*, const float*, const float*,
                         const float*, int, int, int, int, float*, float*,
                         int, hipStream_t);
@@ -442,6 +446,43 @@ torch::Tensor linear_mfma(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
+// bf16-input MFMA GEMM (in-tree; see profiles/r2_nan_rootcause.md):
+// C = act(A @ B^T + bias), A (M,K) bf16, B (N,K) bf16, C fp32 or bf16.
+torch::Tensor linear_bf16(torch::Tensor A, torch::Tensor B,
+                          c10::optional<torch::Tensor> bias, int64_t mode,
+                          bool out_bf16) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16 &&
+                  A.is_contiguous(), "A must be cuda bf16 contiguous");
+  TORCH_CHECK(B.is_cuda() && B.scalar_type() == torch::kBFloat16 &&
+                  B.is_contiguous(), "B must be cuda bf16 contiguous");
+  const int M = A.size(0);
+  const int K = A.size(1);
+  const int N = B.size(0);
+  TORCH_CHECK(B.size(1) == K, "K mismatch");
+  DeviceScope guard(A.get_device());
+  auto C = torch::empty({M, N}, A.options().dtype(
+                                    out_bf16 ? torch::kBFloat16
+                                             : torch::kFloat32));
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    check_f32(*bias, "bias");
+    bias_p = bias->data_ptr<float>();
+  }
+  const int S = linear_splitk_factor_bf16(M, N, K);
+  torch::Tensor ws;
+  float* ws_p = nullptr;
+  if (S > 1) {
+    ws = torch::empty({(int64_t)S, M, N},
+                      A.options().dtype(torch::kFloat32));
+    ws_p = ws.data_ptr<float>();
+  }
+  launch_linear_mfma_bf16(A.data_ptr(), B.data_ptr(), bias_p, M, N, K,
+                          (int)mode, out_bf16 ? 1 : 0, C.data_ptr(), ws_p,
+                          S, current_stream());
+  HIP_CHECK(hipGetLastError());
+  return C;
+}
+
 // ---------------------------------------------------------------------------
 // deep ResNet NHWC bf16 kernels (deep_kernels.hip; validated by deep_probe)
 // ---------------------------------------------------------------------------
@@ -712,6 +753,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_mfma", &linear_mfma, py::arg("A"), py::arg("B"),
         py::arg("bias") = py::none(), py::arg("ymask") = py::none(),
         py::arg("mode") = 0);
+  m.def("linear_bf16", &linear_bf16, py::arg("A"), py::arg("B"),
+        py::arg("bias") = py::none(), py::arg("mode") = 0,
+        py::arg("out_bf16") = false);
   m.def("deep_conv3x3", &deep_conv3x3, py::arg("x"), py::arg("w"),
         py::arg("b"), py::arg("res") = py::none(),
         py::arg("relu_in") = false);

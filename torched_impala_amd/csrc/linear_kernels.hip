@@ -175,6 +175,173 @@ linear_splitk_reduce(const float* __restrict__ P,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// bf16-input GEMM on v_mfma_f32_16x16x32_bf16 (fp32 accumulate): the
+// same 64x64 block / 32x32-per-wave / split-K structure as linear_mfma,
+// with bf16 LDS staging (BK=64, 2 MFMA k-steps per stage).  Written for
+// the fc behind the bf16 MFMA encoder after the hipBLASLt bf16 GEMM
+// path was root-caused corrupting gradients under hipGraph replay
+// (profiles/r2_nan_rootcause.md) — in-tree bf16 GEMMs are clean.
+//   C[i][j] = act( sum_t A[i][t]*B[j][t] + bias[j] )
+// mode: 0 plain, 1 bias+relu.  out_bf16: emit C as bf16 (else fp32).
+// ---------------------------------------------------------------------------
+typedef __bf16 lin_bf16;
+typedef __attribute__((ext_vector_type(8))) __bf16 lin_bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 lin_bf16x4;
+
+constexpr int BBK = 64;                 // bf16 K tile
+constexpr int BSTR = BBK + 8;           // halfs; 144 B rows (16B aligned)
+
+__global__ void __launch_bounds__(THREADS, 2)
+linear_mfma_bf16(const lin_bf16* __restrict__ A,
+                 const lin_bf16* __restrict__ B,
+                 const float* __restrict__ bias, int M, int N, int K,
+                 int mode, int kchunk, int out_bf16,
+                 void* __restrict__ C) {
+  __shared__ __align__(16) lin_bf16 s_a[2][BM * BSTR];
+  __shared__ __align__(16) lin_bf16 s_b[2][BN * BSTR];
+
+  const int bm = blockIdx.x * BM;
+  const int bn = blockIdx.y * BN;
+  const int kb0 = blockIdx.z * kchunk;
+  const int kend = min(K, kb0 + kchunk);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = (wave >> 1) * 32;
+  const int wn = (wave & 1) * 32;
+
+  f32x4 acc[2][2] = {};
+
+  // staging: 256 threads x 2 halves; thread covers row r = tid/4, 16
+  // halfs at k-offset (tid%4)*16 -> one 16B load per half-row chunk
+  const int srow = tid >> 2;            // 0..63
+  const int skoff = (tid & 3) * 16;     // 0,16,32,48
+
+  auto stage = [&](int buf, int k0) {
+    {
+      const int gr = bm + srow;
+      lin_bf16x8 v0 = {}, v1 = {};
+      if (gr < M) {
+        const lin_bf16* src = A + (size_t)gr * K + k0 + skoff;
+        const int krem = kend - (k0 + skoff);
+        if (krem >= 16) {
+          v0 = *reinterpret_cast<const lin_bf16x8*>(src);
+          v1 = *reinterpret_cast<const lin_bf16x8*>(src + 8);
+        } else {
+          for (int i = 0; i < 16 && i < krem; ++i) {
+            if (i < 8) v0[i] = src[i];
+            else v1[i - 8] = src[i];
+          }
+        }
+      }
+      lin_bf16* dst = s_a[buf] + srow * BSTR + skoff;
+      *reinterpret_cast<lin_bf16x8*>(dst) = v0;
+      *reinterpret_cast<lin_bf16x8*>(dst + 8) = v1;
+    }
+    {
+      const int gr = bn + srow;
+      lin_bf16x8 v0 = {}, v1 = {};
+      if (gr < N) {
+        const lin_bf16* src = B + (size_t)gr * K + k0 + skoff;
+        const int krem = kend - (k0 + skoff);
+        if (krem >= 16) {
+          v0 = *reinterpret_cast<const lin_bf16x8*>(src);
+          v1 = *reinterpret_cast<const lin_bf16x8*>(src + 8);
+        } else {
+          for (int i = 0; i < 16 && i < krem; ++i) {
+            if (i < 8) v0[i] = src[i];
+            else v1[i - 8] = src[i];
+          }
+        }
+      }
+      lin_bf16* dst = s_b[buf] + srow * BSTR + skoff;
+      *reinterpret_cast<lin_bf16x8*>(dst) = v0;
+      *reinterpret_cast<lin_bf16x8*>(dst + 8) = v1;
+    }
+  };
+
+  stage(0, kb0);
+  __syncthreads();
+
+  const int nk = (kend - kb0 + BBK - 1) / BBK;
+  const int ar = lane & 15;
+  const int ak = (lane >> 4) * 8;       // this lane's k sub-offset
+  for (int kt = 0; kt < nk; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nk) stage(cur ^ 1, kb0 + (kt + 1) * BBK);
+    #pragma unroll
+    for (int kk = 0; kk < BBK; kk += 32) {
+      const lin_bf16x8 a0 = *reinterpret_cast<const lin_bf16x8*>(
+          s_a[cur] + (wm + ar) * BSTR + kk + ak);
+      const lin_bf16x8 a1 = *reinterpret_cast<const lin_bf16x8*>(
+          s_a[cur] + (wm + 16 + ar) * BSTR + kk + ak);
+      const lin_bf16x8 b0 = *reinterpret_cast<const lin_bf16x8*>(
+          s_b[cur] + (wn + ar) * BSTR + kk + ak);
+      const lin_bf16x8 b1 = *reinterpret_cast<const lin_bf16x8*>(
+          s_b[cur] + (wn + 16 + ar) * BSTR + kk + ak);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0],
+                                                          0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1],
+                                                          0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0],
+                                                          0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1],
+                                                          0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int ccol = lane & 15;
+  const int crow = (lane >> 4) * 4;
+  const bool split = gridDim.z > 1;
+  float* outf = reinterpret_cast<float*>(C) + (size_t)blockIdx.z * M * N;
+  lin_bf16* outh = reinterpret_cast<lin_bf16*>(C);
+  #pragma unroll
+  for (int r = 0; r < 2; ++r) {
+    #pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      #pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = bm + wm + r * 16 + crow + reg;
+        const int col = bn + wn + c * 16 + ccol;
+        if (row < M && col < N) {
+          float v = acc[r][c][reg];
+          if (!split) {
+            if (bias != nullptr) v += bias[col];
+            if (mode == 1) v = v > 0.f ? v : 0.f;
+            if (out_bf16) {
+              outh[(size_t)row * N + col] = (lin_bf16)v;
+              continue;
+            }
+          }
+          outf[(size_t)row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
+// S-way partial sum + bias + activation, bf16 or fp32 output
+__global__ void __launch_bounds__(256)
+linear_splitk_reduce_bf16out(const float* __restrict__ P,
+                             const float* __restrict__ bias, int S, int N,
+                             size_t MN, int mode, int out_bf16,
+                             void* __restrict__ C) {
+  float* cf = reinterpret_cast<float*>(C);
+  lin_bf16* ch = reinterpret_cast<lin_bf16*>(C);
+  for (size_t e = (size_t)blockIdx.x * blockDim.x + threadIdx.x; e < MN;
+       e += (size_t)gridDim.x * blockDim.x) {
+    float v = 0.f;
+    for (int s = 0; s < S; ++s) v += P[(size_t)s * MN + e];
+    if (bias != nullptr) v += bias[(int)(e % N)];
+    if (mode == 1) v = v > 0.f ? v : 0.f;
+    if (out_bf16) ch[e] = (lin_bf16)v;
+    else cf[e] = v;
+  }
+}
+
 }  // namespace impala
 
 extern "C" {
@@ -212,4 +379,36 @@ void launch_linear_mfma(const float* A, const float* B, const float* bias,
   }
 }
 
+
+// bf16-input variant; ws sized S*M*N floats when S>1.
+int linear_splitk_factor_bf16(int M, int N, int K) {
+  const int gm = (M + impala::BM - 1) / impala::BM;
+  const int gn = (N + impala::BN - 1) / impala::BN;
+  int s = (512 + gm * gn - 1) / (gm * gn);
+  const int smax = K / (2 * impala::BBK) > 0 ? K / (2 * impala::BBK) : 1;
+  if (s > smax) s = smax;
+  if (s > 16) s = 16;
+  if (s < 1) s = 1;
+  return s;
+}
+
+void launch_linear_mfma_bf16(const void* A, const void* B, const float* bias,
+                             int M, int N, int K, int mode, int out_bf16,
+                             void* C, float* ws, int S, hipStream_t stream) {
+  dim3 grid((M + impala::BM - 1) / impala::BM,
+            (N + impala::BN - 1) / impala::BN, S);
+  const int kchunk = S > 1 ? (K + S - 1) / S : K;
+  hipLaunchKernelGGL(impala::linear_mfma_bf16, grid, dim3(impala::THREADS),
+                     0, stream, static_cast<const impala::lin_bf16*>(A),
+                     static_cast<const impala::lin_bf16*>(B), bias, M, N, K,
+                     mode, kchunk, out_bf16, S > 1 ? (void*)ws : C);
+  if (S > 1) {
+    const size_t MN = (size_t)M * N;
+    int blocks = (int)((MN + 255) / 256);
+    if (blocks > 2048) blocks = 2048;
+    hipLaunchKernelGGL(impala::linear_splitk_reduce_bf16out, dim3(blocks),
+                       dim3(256), 0, stream, ws, bias, S, N, MN, mode,
+                       out_bf16, C);
+  }
+}
 }  // extern "C"

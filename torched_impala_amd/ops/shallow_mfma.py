@@ -99,19 +99,24 @@ def shallow_mfma_encode(net, frames_u8: torch.Tensor) -> torch.Tensor:
 
 def shallow_mfma_features(net, frames_u8: torch.Tensor) -> torch.Tensor:
     """Full encode incl. the fc: the conv stack on the MFMA kernels, the
-    fc on the custom fp32 split-K MFMA GEMM (ops/linear.py).
+    fc on the in-tree bf16 split-K MFMA GEMM (ops/linear.py,
+    v_mfma_f32_16x16x32_bf16, fp32 masters).
 
-    IMPALA_MFMA_FC=rocblas switches the fc to a rocBLAS bf16 GEMM via
-    tracked .to(bf16) casts — that was the round-2 default until the
-    r19/r21 forensics traced transient garbage fc-grads under hipGraph
-    replay to the hipBLASLt-bf16 path (see ROADMAP: NaN investigation)."""
+    IMPALA_MFMA_FC: "bf16" (default) | "fp32" (the fp32 split-K GEMM) |
+    "rocblas" (hipBLASLt bf16 via tracked casts — the round-2 default
+    until r19/r21 forensics traced transient garbage fc-grads under
+    hipGraph replay to that path; see profiles/r2_nan_rootcause.md)."""
     a3 = _ShallowMfmaEncode.apply(
         frames_u8, net.conv1.weight, net.conv1.bias, net.conv2.weight,
         net.conv2.bias, net.conv3.weight, net.conv3.bias)
-    if os.environ.get("IMPALA_MFMA_FC", "mfma") == "rocblas":
+    mode = os.environ.get("IMPALA_MFMA_FC", "bf16")
+    if mode == "rocblas":
         bf = torch.bfloat16
         y = torch.nn.functional.linear(a3, net.fc.weight.to(bf),
                                        net.fc.bias.to(bf))
         return torch.relu(y).float()
-    from torched_impala_amd.ops.linear import linear_relu
-    return linear_relu(net.fc, a3.float())
+    if mode == "fp32":
+        from torched_impala_amd.ops.linear import linear_relu
+        return linear_relu(net.fc, a3.float())
+    from torched_impala_amd.ops.linear import linear_relu_bf16
+    return linear_relu_bf16(net.fc, a3)

@@ -31,14 +31,28 @@ constexpr int C0 = 4, H0 = 84, W0 = 84;
 constexpr int F1 = 16, K1 = 8, S1 = 4, H1 = 20, W1 = 20;
 constexpr int F2 = 32, K2 = 4, S2 = 2, H2 = 9, W2 = 9;
 constexpr int F3 = 32, K3 = 3, H3 = 7, W3 = 7;
-// 64 slabs suffice once the dW kernels grid-stride (<= DWG writers):
-// zero-fill + finalize traffic shrinks 4x vs the round-1 256.
-constexpr int SLAB = 64;
+// slab count is a HOST-side runtime knob (IMPALA_SHALLOW_SLAB,
+// default 128): more slabs = less dW atomic contention (conv1_dw has
+// 640 concurrent writers) but more zero-fill + finalize traffic.
+// r7 (256 slabs): conv1_dw 48 us; r24 (64): 99 us; 128 = the measured
+// balance point candidate.
+static inline int shallow_slab_count() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("IMPALA_SHALLOW_SLAB");
+    v = e ? atoi(e) : 128;
+    if (v < 4) v = 4;
+    if (v > 1024) v = 1024;
+    v &= ~3;                      // finalize splits the slab dim by 4
+  }
+  return v;
+}
 constexpr int DWG = 512;    // dW resident-workgroup cap (2/CU x 256)
 // bwd_finalize geometry — grid and kernel MUST agree (single source):
 constexpr int FIN_SPLIT = 4;    // slab-dim parallelism per dW chunk
 constexpr int FIN_BB = 160;     // db column-sum blocks per tensor
 constexpr int FIN_GRID = (16 + 32 + 36) * FIN_SPLIT + 3 * FIN_BB;
+// (kernels take the runtime slab count as an argument)
 
 // --------------------------------------------------------------------------
 // masked convert: d_z3 = bf16(d_a3_f32) .* (a3 > 0); both NCHW-flat.
@@ -210,7 +224,7 @@ __global__ __launch_bounds__(256, 2) void conv2_dx(
 // --------------------------------------------------------------------------
 __global__ __launch_bounds__(256, 2) void conv3_dw(
     const bf16_t* __restrict__ a2, const bf16_t* __restrict__ dz3,
-    float* __restrict__ dw_slab, int N) {
+    float* __restrict__ dw_slab, int N, int nslab) {
   __shared__ bf16_t s_a[H2 * W2 * F2];      // 5.1 KiB NHWC
   __shared__ bf16_t s_d[F3 * 64];           // 4 KiB; rows padded to the
                                             // k-loop's 2x32 px range
@@ -218,7 +232,7 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (288 * F3);
+  float* dw = dw_slab + (size_t)(blockIdx.x % nslab) * (288 * F3);
   constexpr int MPW = 5;                    // ceil(18/4) m-frags per wave
   floatx4 acc[MPW][2];
   #pragma unroll
@@ -302,14 +316,14 @@ __global__ __launch_bounds__(256, 2) void conv3_dw(
 // --------------------------------------------------------------------------
 __global__ __launch_bounds__(256, 2) void conv2_dw(
     const bf16_t* __restrict__ a1, const bf16_t* __restrict__ dz2,
-    float* __restrict__ dw_slab, int N) {
+    float* __restrict__ dw_slab, int N, int nslab) {
   __shared__ bf16_t s_a[F1 * H1 * W1];      // 12.5 KiB planar
   __shared__ bf16_t s_d[F2 * 96];           // 6 KiB transposed
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (256 * F2);
+  float* dw = dw_slab + (size_t)(blockIdx.x % nslab) * (256 * F2);
   constexpr int MPW = 4;                    // 16 m-frags over 4 waves
   floatx4 acc[MPW][2];
   #pragma unroll
@@ -395,7 +409,8 @@ __global__ __launch_bounds__(256, 2) void conv2_dw(
 // --------------------------------------------------------------------------
 __global__ __launch_bounds__(256, 2) void conv1_dw(
     const unsigned char* __restrict__ frames,
-    const bf16_t* __restrict__ dz1, float* __restrict__ dw_slab, int N) {
+    const bf16_t* __restrict__ dz1, float* __restrict__ dw_slab, int N,
+    int nslab) {
   __shared__ bf16_t s_a[C0 * H0 * W0];      // 55.1 KiB planar
   __shared__ bf16_t s_d[F1 * 416];          // 13 KiB transposed
   // 68.1 KiB total -> 2 blocks/CU in 160 KiB LDS
@@ -403,7 +418,7 @@ __global__ __launch_bounds__(256, 2) void conv1_dw(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  float* dw = dw_slab + (size_t)(blockIdx.x % SLAB) * (256 * F1);
+  float* dw = dw_slab + (size_t)(blockIdx.x % nslab) * (256 * F1);
   constexpr int MPW = 4;                    // 16 m-frags over 4 waves
   floatx4 acc[MPW];
   #pragma unroll
@@ -496,10 +511,11 @@ __global__ __launch_bounds__(256) void bwd_finalize(
     const float* __restrict__ s1, const float* __restrict__ s2,
     const float* __restrict__ s3, const bf16_t* __restrict__ da1,
     const bf16_t* __restrict__ da2, const bf16_t* __restrict__ dz3, int N,
+    int nslab,
     float* __restrict__ dw1, float* __restrict__ dw2,
     float* __restrict__ dw3, float* __restrict__ db1,
     float* __restrict__ db2, float* __restrict__ db3) {
-  constexpr int SLC = SLAB / FIN_SPLIT;         // slabs per block
+  const int SLC = nslab / FIN_SPLIT;            // slabs per block
   constexpr int SPLIT = FIN_SPLIT;
   constexpr int BW1 = 16 * SPLIT, BW2 = 32 * SPLIT, BW3 = 36 * SPLIT;
   constexpr int BB = FIN_BB;
@@ -594,7 +610,7 @@ __global__ __launch_bounds__(256) void bwd_finalize(
   }
 }
 
-extern "C" int shallow_mfma_slab() { return SLAB; }
+extern "C" int shallow_mfma_slab() { return shallow_slab_count(); }
 
 extern "C" void launch_shallow_mask_convert(const float* d, const void* act,
                                             void* o, long n,
@@ -621,9 +637,10 @@ extern "C" void launch_shallow_mfma_bwd(
                      static_cast<const bf16_t*>(wr3),
                      static_cast<const bf16_t*>(a2),
                      static_cast<bf16_t*>(da2), N);
+  const int nslab = shallow_slab_count();
   hipLaunchKernelGGL(conv3_dw, dim3(N < DWG ? N : DWG), dim3(256), 0, s,
                      static_cast<const bf16_t*>(a2),
-                     static_cast<const bf16_t*>(dz3), dw3_slab, N);
+                     static_cast<const bf16_t*>(dz3), dw3_slab, N, nslab);
   hipLaunchKernelGGL(conv2_dx, dim3(N), dim3(256), 0, s,
                      static_cast<const bf16_t*>(da2),
                      static_cast<const bf16_t*>(wr2),
@@ -631,12 +648,12 @@ extern "C" void launch_shallow_mfma_bwd(
                      static_cast<bf16_t*>(da1), N);
   hipLaunchKernelGGL(conv2_dw, dim3(N < DWG ? N : DWG), dim3(256), 0, s,
                      static_cast<const bf16_t*>(a1),
-                     static_cast<const bf16_t*>(da2), dw2_slab, N);
+                     static_cast<const bf16_t*>(da2), dw2_slab, N, nslab);
   // conv1_dw keeps one-image-per-WG (grid N): its 68 KiB staging phase
   // overlaps across queued blocks; the capped grid-stride variant lost
   // that overlap (measured 48 -> 91 us, r8 CSV).
   hipLaunchKernelGGL(conv1_dw, dim3(N), dim3(256), 0, s, frames,
-                     static_cast<const bf16_t*>(da1), dw1_slab, N);
+                     static_cast<const bf16_t*>(da1), dw1_slab, N, nslab);
 }
 
 }  // namespace impala_shallow_mfma_bwd
@@ -654,6 +671,7 @@ extern "C" void launch_shallow_bwd_finalize(
                      s, s1, s2,
                      s3, static_cast<const bf16_t*>(da1),
                      static_cast<const bf16_t*>(da2),
-                     static_cast<const bf16_t*>(dz3), N, dw1, dw2, dw3, db1,
-                     db2, db3);
+                     static_cast<const bf16_t*>(dz3), N,
+                     impala_shallow_mfma_bwd::shallow_slab_count(), dw1,
+                     dw2, dw3, db1, db2, db3);
 }

@@ -76,8 +76,14 @@ __device__ __forceinline__ float loadw(const WT* p) {
 //   Hp      (T, B, 256)   masked h input at step t (saved for bwd + dW_hh)
 //   hT, cT  (B, 256)
 // ---------------------------------------------------------------------------
+// forward runs 512 threads (8 waves), TWO gate columns per thread:
+// the per-step cost is barrier-dominated (2 x full-block barriers x T)
+// and a narrower block halves the barrier width while the dual-column
+// GEMV keeps the same loads in flight per lane.
+constexpr int FWD_THREADS = 512;
+
 template <typename WT>
-__global__ void __launch_bounds__(THREADS)
+__global__ void __launch_bounds__(FWD_THREADS)
 lstm_seq_fwd(const float* __restrict__ Xg, const WT* __restrict__ W_hhT,
              const float* __restrict__ notdone, const float* __restrict__ h0,
              const float* __restrict__ c0, int T, int B,
@@ -103,27 +109,36 @@ lstm_seq_fwd(const float* __restrict__ Xg, const WT* __restrict__ W_hhT,
     const size_t rowG = ((size_t)t * B + b) * G4;
     // the done-mask is a SCALAR per (t,b): gates = Xg + (nd*h).W
     //                                            = Xg + nd*(h.W)
-    // so no masking phase (and barrier) is needed before the GEMV —
-    // the 16-wave barrier was ~1/3 of the sequential critical path.
+    // so no masking phase (and barrier) is needed before the GEMV.
     if (save_for_bwd && tid < H) Hp[rowH + tid] = s_h[tid] * nd;
-    // gate GEMV: thread j computes gate column j. k chunked by 4 (one
-    // ds_read_b128 broadcast of h, four batched W row loads in flight).
+    // gate GEMV: thread j computes gate columns j and j+512. k chunked
+    // by 4 (one ds_read_b128 broadcast of h, eight W loads in flight).
     {
-      float acc = 0.f;
-      const WT* wcol = W_hhT + tid;
+      float acc0 = 0.f, acc1 = 0.f;
+      const WT* wc0 = W_hhT + tid;
+      const WT* wc1 = W_hhT + tid + 512;
       #pragma unroll 4
       for (int k0 = 0; k0 < H; k0 += 4) {
         const float4 hv = *reinterpret_cast<const float4*>(&s_h[k0]);
-        const float w0 = loadw(wcol + (size_t)(k0 + 0) * G4);
-        const float w1 = loadw(wcol + (size_t)(k0 + 1) * G4);
-        const float w2 = loadw(wcol + (size_t)(k0 + 2) * G4);
-        const float w3 = loadw(wcol + (size_t)(k0 + 3) * G4);
-        acc = fmaf(hv.x, w0, acc);
-        acc = fmaf(hv.y, w1, acc);
-        acc = fmaf(hv.z, w2, acc);
-        acc = fmaf(hv.w, w3, acc);
+        const float a0 = loadw(wc0 + (size_t)(k0 + 0) * G4);
+        const float a1 = loadw(wc0 + (size_t)(k0 + 1) * G4);
+        const float a2 = loadw(wc0 + (size_t)(k0 + 2) * G4);
+        const float a3 = loadw(wc0 + (size_t)(k0 + 3) * G4);
+        const float b0 = loadw(wc1 + (size_t)(k0 + 0) * G4);
+        const float b1 = loadw(wc1 + (size_t)(k0 + 1) * G4);
+        const float b2 = loadw(wc1 + (size_t)(k0 + 2) * G4);
+        const float b3 = loadw(wc1 + (size_t)(k0 + 3) * G4);
+        acc0 = fmaf(hv.x, a0, acc0);
+        acc0 = fmaf(hv.y, a1, acc0);
+        acc0 = fmaf(hv.z, a2, acc0);
+        acc0 = fmaf(hv.w, a3, acc0);
+        acc1 = fmaf(hv.x, b0, acc1);
+        acc1 = fmaf(hv.y, b1, acc1);
+        acc1 = fmaf(hv.z, b2, acc1);
+        acc1 = fmaf(hv.w, b3, acc1);
       }
-      s_g[tid] = fmaf(nd, acc, Xg[rowG + tid]);
+      s_g[tid] = fmaf(nd, acc0, Xg[rowG + tid]);
+      s_g[tid + 512] = fmaf(nd, acc1, Xg[rowG + tid + 512]);
     }
     __syncthreads();
     if (tid < H) {
@@ -264,13 +279,13 @@ void launch_lstm_seq_fwd(const float* Xg, const void* W_hhT,
                          hipStream_t stream) {
   if (wbf16)
     hipLaunchKernelGGL(impala::lstm_seq_fwd<impala::lstm_bf16_t>, dim3(B),
-                       dim3(impala::THREADS), 0, stream, Xg,
+                       dim3(impala::FWD_THREADS), 0, stream, Xg,
                        static_cast<const impala::lstm_bf16_t*>(W_hhT),
                        notdone, h0, c0, T, B, Hout, act, Cout, Hp, hT, cT,
                        save_for_bwd);
   else
     hipLaunchKernelGGL(impala::lstm_seq_fwd<float>, dim3(B),
-                       dim3(impala::THREADS), 0, stream, Xg,
+                       dim3(impala::FWD_THREADS), 0, stream, Xg,
                        static_cast<const float*>(W_hhT), notdone, h0, c0, T,
                        B, Hout, act, Cout, Hp, hT, cT, save_for_bwd);
 }

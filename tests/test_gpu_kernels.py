@@ -455,3 +455,27 @@ def test_vtrace_scan_parity_nondefault_clips(clip_rho, clip_c, lam):
     torch.testing.assert_close(vs.cpu(), ref.vs, rtol=1e-5, atol=1e-5)
     torch.testing.assert_close(adv.cpu(), ref.pg_advantages, rtol=1e-5,
                                atol=1e-5)
+
+
+def test_nonfinite_grad_skip_fused_kernel():
+    """GPU fused clip+RMSProp: an inf grad element must skip the whole
+    update (device skip counter) — the defense the r12-r21 forensics
+    landed (profiles/r2_nan_rootcause.md)."""
+    from torched_impala_amd.ops.optim import FlatRMSprop
+    dev = _dev()
+    p = torch.nn.Parameter(torch.randn(4096, device=dev))
+    opt = FlatRMSprop([p], lr=0.01, max_grad_norm=40.0)
+    p0 = opt.flat_param.clone()
+    sq0 = opt.flat_sqavg.clone()
+    opt.flat_grad.normal_()
+    opt.flat_grad[123] = float("inf")
+    opt.step()
+    torch.cuda.synchronize()
+    assert float(opt.skip_t) == 1.0
+    torch.testing.assert_close(opt.flat_param, p0, rtol=0, atol=0)
+    torch.testing.assert_close(opt.flat_sqavg, sq0, rtol=0, atol=0)
+    opt.flat_grad.normal_()
+    opt.step()
+    torch.cuda.synchronize()
+    assert float(opt.skip_t) == 1.0
+    assert not torch.equal(opt.flat_param, p0)

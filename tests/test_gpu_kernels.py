@@ -479,3 +479,32 @@ def test_nonfinite_grad_skip_fused_kernel():
     torch.cuda.synchronize()
     assert float(opt.skip_t) == 1.0
     assert not torch.equal(opt.flat_param, p0)
+
+
+def test_unpack_time_major_parity():
+    """The rollout-unpack kernel == the ATen permute+reshape transform
+    for every staging key shape (incl. the LSTM core-state layout and
+    non-16B-aligned inners)."""
+    from torched_impala_amd import ops
+    dev = _dev()
+    ext = ops.ext_hip()
+    torch.manual_seed(0)
+    cases = [
+        torch.randint(0, 256, (2, 21, 16, 4, 84, 84), dtype=torch.uint8),
+        torch.randn(2, 21, 16),
+        torch.rand(2, 21, 16) < 0.5,
+        torch.randint(0, 6, (2, 21, 16)),
+        torch.randn(2, 21, 16, 6),
+        torch.randint(0, 99, (2, 21, 16), dtype=torch.int32),
+        torch.randn(4, 2, 16, 256),        # initial_core_state
+        torch.rand(32, 11, 1) < 0.5,       # n=1 (1-byte inner tail)
+        torch.randn(32, 11, 1, 6),
+    ]
+    for src in cases:
+        src = src.to(dev)
+        got = ext.unpack_time_major(src.contiguous(), 1)
+        k, t1 = src.shape[0], src.shape[1]
+        ref = src.transpose(0, 1).reshape(
+            t1, k * src.shape[2], *src.shape[3:])
+        assert got.shape == ref.shape, (got.shape, ref.shape)
+        assert torch.equal(got, ref), f"mismatch for {tuple(src.shape)}"

@@ -77,7 +77,8 @@ def build_actor(force: bool = False) -> str:
 
 HIP_SOURCES = ["hip_ext.cpp", "vtrace_kernels.hip", "lstm_kernels.hip",
                "optim_kernels.hip", "conv_kernels.hip", "linear_kernels.hip",
-               "deep_kernels.hip", "shallow_mfma.hip", "shallow_mfma_bwd.hip"]
+               "deep_kernels.hip", "shallow_mfma.hip", "shallow_mfma_bwd.hip",
+               "pack_kernels.hip"]
 
 
 def build_probe(force: bool = False) -> str:

@@ -53,6 +53,8 @@ void launch_shallow_encoder_bwd(const unsigned char*, const float*,
                                 float*, float*, hipStream_t);
 int shallow_encoder_slab_elems();
 int linear_splitk_factor(int, int, int);
+void launch_unpack_time_major(const void*, void*, int, int, long,
+                              hipStream_t);
 int linear_splitk_factor_bf16(int, int, int);
 void launch_linear_mfma_bf16(const void*, const void*, const float*, int,
                              int, int, int, int, void*, float*, int,
@@ -446,6 +448,35 @@ torch::Tensor linear_mfma(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
+// slot-major staging (K, T1, ...) -> time-major (T1, K*..., ...) as one
+// block-permutation kernel (SURVEY.md §2c rollout_pack/unpack).
+torch::Tensor unpack_time_major(torch::Tensor src, int64_t lead_keep) {
+  // src (K, T1, *rest) contiguous -> out (T1, K*rest[0], *rest[1:])
+  // when lead_keep == 1 the first rest dim is merged with K (the
+  // envs-per-slot dim n), matching Batcher._to_time_major.
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous(), "src layout");
+  const int K = src.size(0);
+  const int T1 = src.size(1);
+  DeviceScope guard(src.get_device());
+  std::vector<int64_t> oshape;
+  oshape.push_back(T1);
+  int64_t inner_elems = 1;
+  for (int i = 2; i < src.dim(); ++i) inner_elems *= src.size(i);
+  if (lead_keep == 1 && src.dim() >= 3) {
+    oshape.push_back((int64_t)K * src.size(2));
+    for (int i = 3; i < src.dim(); ++i) oshape.push_back(src.size(i));
+  } else {
+    oshape.push_back((int64_t)K);
+    for (int i = 2; i < src.dim(); ++i) oshape.push_back(src.size(i));
+  }
+  auto out = torch::empty(oshape, src.options());
+  const long inner_bytes = inner_elems * src.element_size();
+  launch_unpack_time_major(src.data_ptr(), out.data_ptr(), K, T1,
+                           inner_bytes, current_stream());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 // bf16-input MFMA GEMM (in-tree; see profiles/r2_nan_rootcause.md):
 // C = act(A @ B^T + bias), A (M,K) bf16, B (N,K) bf16, C fp32 or bf16.
 torch::Tensor linear_bf16(torch::Tensor A, torch::Tensor B,
@@ -753,6 +784,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_mfma", &linear_mfma, py::arg("A"), py::arg("B"),
         py::arg("bias") = py::none(), py::arg("ymask") = py::none(),
         py::arg("mode") = 0);
+  m.def("unpack_time_major", &unpack_time_major, py::arg("src"),
+        py::arg("lead_keep") = 1);
   m.def("linear_bf16", &linear_bf16, py::arg("A"), py::arg("B"),
         py::arg("bias") = py::none(), py::arg("mode") = 0,
         py::arg("out_bf16") = false);

@@ -163,7 +163,18 @@ class Batcher:
     def _to_time_major(self, st: Dict[str, torch.Tensor]):
         out = {}
         k, n = self.k, self.flags.envs_per_actor
+        use_kernel = False
+        for v in st.values():
+            use_kernel = v.is_cuda and ops.hip_available()
+            break
+        ext = ops.ext_hip() if use_kernel else None
         for key, v in st.items():
+            if use_kernel:
+                # one block-permutation kernel per tensor instead of an
+                # ATen permute+reshape copy (SURVEY §2c rollout_unpack;
+                # ~80 -> ~25 us/step inside the captured graph)
+                out[key] = ext.unpack_time_major(v.contiguous(), 1)
+                continue
             if key == "initial_core_state":
                 # (k, 2, n, H) -> (2, k*n, H)
                 out[key] = v.permute(1, 0, 2, 3).reshape(2, k * n, -1)

@@ -274,3 +274,38 @@ def test_multirank_resume_lockstep(tmp_path):
     assert res[0][0] == res[1][0] == base_frames + 3 * 4 * 10 * 2, res
     # and their models stayed in lockstep (bitwise-equal param digests)
     assert res[0][1] == res[1][1], res
+
+
+def _ramp_worker(rank, world, port, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from torched_impala_amd import learner
+    from torched_impala_amd.config import Flags
+    flags = Flags(env="cartpole", num_actors=1, batch_size=4,
+                  unroll_length=10, total_steps=10**9, device="cpu",
+                  disable_checkpoint=True, ramp_ready=True, seed=5 + rank)
+    stats = learner.train(flags, max_learner_steps=2)
+    q.put((rank, stats["learner_steps"]))
+
+
+@pytest.mark.timeout(300)
+def test_ramp_barrier_world2_no_deadlock():
+    """bench.py sets ramp_ready on every rank; the ring-full wait + the
+    cross-rank barrier must not deadlock under world 2 (the driver's
+    SCALE run takes exactly this path)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ramp_worker, args=(r, 2, 29631, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(2):
+        rank, steps = q.get(timeout=240)
+        res[rank] = steps
+    for p in procs:
+        p.join(timeout=60)
+    assert res == {0: 2, 1: 2}, res

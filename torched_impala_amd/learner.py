@@ -547,36 +547,6 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
                  target)
         if world > 1:
             torch.distributed.barrier()
-        if graphed is not None and device.type == "cuda":
-            # settle: run a FIXED number of real steps (same count on
-            # every rank — the flat all-reduce stays collective-
-            # consistent) and roll the optimizer state back after.
-            # Purpose: the timed window of a short driver-chosen bench
-            # (e.g. 20 steps / 5 warmup) otherwise lands in the
-            # post-ramp recovery phase — actors waking from the
-            # ring-full stall, first jit/weight re-packs, GPU clocks
-            # ramping — and under-reads steady state by ~35%
-            # (gpurun_out/r2_rehearsal.log: 427k vs 660-690k). The K
-            # timed steps still do full work; training semantics are
-            # unchanged (params/sqavg restored bitwise).
-            N_SETTLE = 1500
-            snap = (optimizer.flat_param.clone(),
-                    optimizer.flat_sqavg.clone())
-            for _ in range(N_SETTLE):
-                b = batcher.next_batch(materialize=False)
-                s_i = b.get("_staging_index")
-                graphed.step(s_i, scheduler)
-                batcher.mark_consumed(b)
-                if use_flat_publish:
-                    mirror.publish_flat_async(optimizer.flat_param)
-            with torch.no_grad():
-                optimizer.flat_param.copy_(snap[0])
-                optimizer.flat_sqavg.copy_(snap[1])
-            if use_flat_publish:
-                mirror._finish_pending(wait=True)
-                mirror.publish_flat_async(optimizer.flat_param)
-            torch.cuda.synchronize()
-            log.info("ramp settle: %d steps", N_SETTLE)
 
     # debug: IMPALA_DUMP_NONFINITE=<path> saves the batch + optimizer
     # state the first time total_loss goes non-finite (costs one host

@@ -535,10 +535,7 @@ def train(flags: Flags, max_learner_steps: Optional[int] = None,
         # over short windows (73 ms timed region on a 7 s
         # process was ramp-dominated)
         deadline = time.time() + 600.0
-        # leave one batch's worth of slots unfilled: with the ring
-        # completely full every actor sleeps at t0 and the first timed
-        # steps measure the staggered wake-up instead of steady state
-        target = max(1, buffers.num_slots - batcher.k)
+        target = buffers.num_slots
         while (full_queue.size_approx() < target
                and time.time() < deadline):
             for i, pr in enumerate(procs):
